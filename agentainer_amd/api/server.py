@@ -1,0 +1,323 @@
+"""REST API + agent proxy — the reference's HTTP surface, engine-backed.
+
+Route table mirrors `internal/api/server.go:68-107`:
+
+  public:      GET  /health
+               ANY  /agent/{id}/{path...}         (unauthenticated proxy)
+  authed:      POST /agents                        deploy
+               GET  /agents                        list
+               GET  /agents/{id}
+               POST /agents/{id}/start|stop|restart|pause|resume
+               DELETE /agents/{id}                 remove
+               GET  /agents/{id}/logs
+               POST /agents/{id}/invoke            (real dispatch — the
+                                                    reference's stub,
+                                                    server.go:407-430, is
+                                                    implemented here per
+                                                    SURVEY.md §7.4)
+               GET  /agents/{id}/metrics, /agents/{id}/metrics/history
+               GET  /agents/{id}/requests, /agents/{id}/requests/{reqId}
+               POST /agents/{id}/requests/{reqId}/replay
+               GET  /agents/{id}/health, /health/agents
+  extra (unifying CLI-direct paths per SURVEY.md §1 note):
+               POST/GET /backups, POST /backups/{id}/restore,
+               DELETE /backups/{id}, GET /audit, GET /logs
+
+Responses use the reference's envelope {success, message, data}
+(server.go:50-54). Auth: bearer token or ?token= (server.go:449-478);
+per-agent tokens are honored in addition to the server token (fixing the
+stored-but-unused Token field). Input caps mirror deploy
+validation: name<=64, model<=256, env<=50 entries (server.go:157-179).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Any, Dict, Optional
+
+from fastapi import Depends, FastAPI, Header, Query, Request as HttpRequest
+from fastapi.responses import JSONResponse
+
+from ..registry import AgentNotFound
+from ..engine.base import ModelNotFound
+from ..service import Runtime
+
+
+def envelope(success: bool, message: str = "", data: Any = None) -> Dict[str, Any]:
+    return {"success": success, "message": message, "data": data}
+
+
+def create_app(rt: Runtime) -> FastAPI:
+    app = FastAPI(title="agentainer-amd", version="0.1.0")
+    app.state.runtime = rt
+    server_token = rt.config.get("security", "api_token")
+
+    # ---------- auth ----------
+
+    def _client(request: HttpRequest) -> Dict[str, str]:
+        return {
+            "ip": request.client.host if request.client else "",
+            "user_agent": request.headers.get("user-agent", ""),
+        }
+
+    async def require_auth(request: HttpRequest,
+                           authorization: Optional[str] = Header(None),
+                           token: Optional[str] = Query(None)) -> str:
+        presented = token
+        if authorization and authorization.lower().startswith("bearer "):
+            presented = authorization[7:]
+        valid = {server_token}
+        # per-agent tokens may manage that agent's own routes
+        agent_id = request.path_params.get("agent_id")
+        if agent_id:
+            agent = rt.agents.try_get(agent_id)
+            if agent and agent.token:
+                valid.add(agent.token)
+        if presented not in valid:
+            raise AuthError()
+        return presented or ""
+
+    class AuthError(Exception):
+        pass
+
+    @app.exception_handler(AuthError)
+    async def _auth_err(request, exc):
+        return JSONResponse(status_code=401,
+                            content=envelope(False, "unauthorized"))
+
+    @app.exception_handler(AgentNotFound)
+    async def _nf_err(request, exc):
+        return JSONResponse(status_code=404, content=envelope(False, str(exc)))
+
+    @app.exception_handler(ModelNotFound)
+    async def _model_err(request, exc):
+        return JSONResponse(status_code=400, content=envelope(False, str(exc)))
+
+    # ---------- public ----------
+
+    @app.get("/health")
+    async def health():
+        return envelope(True, "ok", {"status": "healthy", "ts": time.time()})
+
+    @app.api_route("/agent/{agent_id}/{agent_path:path}",
+                   methods=["GET", "POST", "PUT", "DELETE", "PATCH"])
+    async def proxy(agent_id: str, agent_path: str, request: HttpRequest):
+        body: Any = None
+        if request.method in ("POST", "PUT", "PATCH"):
+            try:
+                body = await request.json()
+            except Exception:
+                raw = await request.body()
+                body = {"message": raw.decode("utf-8", "replace")} if raw else {}
+        replay = request.headers.get("x-agentainer-replay", "").lower() == "true"
+        req_id = request.headers.get("x-agentainer-request-id") or None
+        status, payload = rt.agent_request(
+            agent_id, request.method, "/" + agent_path, body=body,
+            headers={"content-type": request.headers.get("content-type", "")},
+            replay=replay, req_id=req_id)
+        return JSONResponse(status_code=status, content=payload)
+
+    # ---------- management ----------
+
+    @app.post("/agents")
+    async def deploy(request: HttpRequest, _tok: str = Depends(require_auth)):
+        body = await request.json()
+        name = str(body.get("name", ""))
+        model = str(body.get("model", body.get("image", "")))
+        env = body.get("env_vars", body.get("env", {})) or {}
+        # input caps (server.go:157-179)
+        if not name or len(name) > 64:
+            return JSONResponse(status_code=422,
+                                content=envelope(False, "name required, <=64 chars"))
+        if not model or len(model) > 256:
+            return JSONResponse(status_code=422,
+                                content=envelope(False, "model required, <=256 chars"))
+        if len(env) > 50:
+            return JSONResponse(status_code=422,
+                                content=envelope(False, "too many env vars (max 50)"))
+        agent = rt.agents.deploy(
+            name=name, model=model,
+            dtype=body.get("dtype", "bf16"),
+            tp_degree=int(body.get("tp_degree", 1)),
+            kv_budget=int(body.get("kv_budget", 0)),
+            max_context=int(body.get("max_context", 8192)),
+            env=env,
+            auto_restart=bool(body.get("auto_restart", False)),
+            token=body.get("token"),
+            health_check=body.get("health_check"),
+            system_prompt=body.get("system_prompt", ""),
+            sampling=body.get("sampling") or {},
+        )
+        cl = _client(request)
+        rt.logger.audit("api", "deploy", agent.id, "success", **cl)
+        return envelope(True, f"agent {agent.name} deployed", agent.to_dict())
+
+    @app.get("/agents")
+    async def list_agents(_tok: str = Depends(require_auth)):
+        rt.reconciler.sync_all()  # sync-before-list (agent.go:393-398)
+        return envelope(True, "", [a.to_dict() for a in rt.agents.list()])
+
+    @app.get("/agents/{agent_id}")
+    async def get_agent(agent_id: str, _tok: str = Depends(require_auth)):
+        rt.reconciler.sync_agent(agent_id)
+        return envelope(True, "", rt.agents.get(agent_id).to_dict())
+
+    def _lifecycle(op: str):
+        async def handler(agent_id: str, request: HttpRequest,
+                          _tok: str = Depends(require_auth)):
+            fn = getattr(rt.agents, op)
+            try:
+                agent = fn(agent_id)
+            except AgentNotFound:
+                raise
+            except Exception as exc:  # noqa: BLE001
+                rt.logger.audit("api", op, agent_id, "failure", details=str(exc),
+                                **_client(request))
+                return JSONResponse(status_code=409, content=envelope(False, str(exc)))
+            rt.reconciler.sync_agent(agent_id)  # quick-sync after op (agent.go:174-178)
+            rt.logger.audit("api", op, agent_id, "success", **_client(request))
+            return envelope(True, f"agent {op}", agent.to_dict() if agent else None)
+        return handler
+
+    app.post("/agents/{agent_id}/start")(_lifecycle("start"))
+    app.post("/agents/{agent_id}/stop")(_lifecycle("stop"))
+    app.post("/agents/{agent_id}/restart")(_lifecycle("restart"))
+    app.post("/agents/{agent_id}/pause")(_lifecycle("pause"))
+    app.post("/agents/{agent_id}/resume")(_lifecycle("resume"))
+
+    @app.delete("/agents/{agent_id}")
+    async def remove(agent_id: str, request: HttpRequest,
+                     _tok: str = Depends(require_auth)):
+        rt.agents.remove(agent_id, request_manager=rt.requests)
+        rt.logger.audit("api", "remove", agent_id, "success", **_client(request))
+        return envelope(True, "agent removed")
+
+    @app.get("/agents/{agent_id}/logs")
+    async def logs(agent_id: str, limit: int = 200, _tok: str = Depends(require_auth)):
+        return envelope(True, "", rt.agents.get_logs(agent_id, limit=limit))
+
+    @app.post("/agents/{agent_id}/invoke")
+    async def invoke(agent_id: str, request: HttpRequest,
+                     _tok: str = Depends(require_auth)):
+        """Authenticated dispatch — implemented for real (ref stub, §7.4)."""
+        body = await request.json()
+        path = body.get("path", "/chat")
+        method = body.get("method", "POST")
+        status, payload = rt.agent_request(agent_id, method, path,
+                                           body=body.get("body", body))
+        return JSONResponse(status_code=status,
+                            content=envelope(status < 400, "", payload))
+
+    # ---------- requests (WAL) ----------
+
+    @app.get("/agents/{agent_id}/requests")
+    async def list_requests(agent_id: str, status: str = "",
+                            _tok: str = Depends(require_auth)):
+        rt.agents.get(agent_id)
+        allr = rt.requests.all_requests(agent_id)
+        if status:
+            allr = {status: allr.get(status, [])}
+        return envelope(True, "", {k: [r.to_dict() for r in v] for k, v in allr.items()})
+
+    @app.get("/agents/{agent_id}/requests/{req_id}")
+    async def get_request(agent_id: str, req_id: str, _tok: str = Depends(require_auth)):
+        r = rt.requests.get(agent_id, req_id)
+        if r is None:
+            return JSONResponse(status_code=404,
+                                content=envelope(False, f"request {req_id} not found"))
+        return envelope(True, "", r.to_dict())
+
+    @app.post("/agents/{agent_id}/requests/{req_id}/replay")
+    async def replay_request(agent_id: str, req_id: str,
+                             _tok: str = Depends(require_auth)):
+        r = rt.requests.get(agent_id, req_id)
+        if r is None:
+            return JSONResponse(status_code=404,
+                                content=envelope(False, f"request {req_id} not found"))
+        status, payload = rt.agent_request(agent_id, r.method, r.path, body=r.body,
+                                           replay=True, req_id=req_id)
+        if status == 200:
+            rt.requests.store_response(agent_id, req_id, payload)
+        return JSONResponse(status_code=status, content=envelope(status < 400, "", payload))
+
+    # ---------- health / metrics ----------
+
+    @app.get("/agents/{agent_id}/health")
+    async def agent_health(agent_id: str, _tok: str = Depends(require_auth)):
+        rt.agents.get(agent_id)
+        st = rt.health.get_status(agent_id) or rt.health.check_one(agent_id)
+        return envelope(True, "", st)
+
+    @app.get("/health/agents")
+    async def all_health(_tok: str = Depends(require_auth)):
+        return envelope(True, "", rt.health.get_all_statuses())
+
+    @app.get("/agents/{agent_id}/metrics")
+    async def agent_metrics(agent_id: str, _tok: str = Depends(require_auth)):
+        rt.agents.get(agent_id)
+        return envelope(True, "", rt.metrics.get_metrics(agent_id) or {})
+
+    @app.get("/agents/{agent_id}/metrics/history")
+    async def metrics_history(agent_id: str, duration_s: float = 3600.0,
+                              _tok: str = Depends(require_auth)):
+        rt.agents.get(agent_id)
+        return envelope(True, "", rt.metrics.get_metrics_history(agent_id, duration_s))
+
+    @app.get("/metrics/engine")
+    async def engine_metrics(_tok: str = Depends(require_auth)):
+        return envelope(True, "", rt.engine.stats())
+
+    # ---------- backups / audit / logs ----------
+
+    @app.post("/backups")
+    async def create_backup(request: HttpRequest, _tok: str = Depends(require_auth)):
+        body = await request.json()
+        b = rt.backups.create(body.get("name", "backup"), body.get("description", ""),
+                              body.get("agent_ids"))
+        rt.logger.audit("api", "backup.create", b["id"], "success", **_client(request))
+        return envelope(True, "", {"id": b["id"], "n_agents": len(b["agents"])})
+
+    @app.get("/backups")
+    async def list_backups(_tok: str = Depends(require_auth)):
+        return envelope(True, "", rt.backups.list())
+
+    @app.post("/backups/{backup_id}/restore")
+    async def restore_backup(backup_id: str, request: HttpRequest,
+                             _tok: str = Depends(require_auth)):
+        agents = rt.backups.restore(backup_id)
+        rt.logger.audit("api", "backup.restore", backup_id, "success", **_client(request))
+        return envelope(True, "", [a.to_dict() for a in agents])
+
+    @app.delete("/backups/{backup_id}")
+    async def delete_backup(backup_id: str, _tok: str = Depends(require_auth)):
+        rt.backups.delete(backup_id)
+        return envelope(True, "backup deleted")
+
+    @app.get("/audit")
+    async def audit_logs(user: str = "", action: str = "", resource: str = "",
+                         limit: int = 200, _tok: str = Depends(require_auth)):
+        return envelope(True, "", rt.logger.get_audit_logs(
+            user=user, action=action, resource=resource, limit=limit))
+
+    @app.get("/logs")
+    async def server_logs(level: str = "", component: str = "", agent_id: str = "",
+                          limit: int = 200, _tok: str = Depends(require_auth)):
+        return envelope(True, "", rt.logger.get_logs(
+            level=level, component=component, agent_id=agent_id, limit=limit))
+
+    return app
+
+
+def run_server(rt: Runtime, host: Optional[str] = None, port: Optional[int] = None):
+    import uvicorn
+
+    app = create_app(rt)
+    rt.start_workers()
+    rt.recover()
+    try:
+        uvicorn.run(app,
+                    host=host or rt.config.get("server", "host"),
+                    port=int(port or rt.config.get("server", "port")),
+                    log_level="warning")
+    finally:
+        rt.shutdown()

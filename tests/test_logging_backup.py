@@ -1,0 +1,79 @@
+"""Structured/audit logging + backup/restore."""
+
+import json
+import os
+
+
+def test_log_sinks_and_query(runtime):
+    runtime.logger.info("hello", component="test", agent_id="a-1")
+    runtime.logger.error("bad thing", component="test")
+    logs = runtime.logger.get_logs(component="test")
+    assert len(logs) == 2
+    only_err = runtime.logger.get_logs(level="error")
+    assert len(only_err) == 1 and only_err[0]["message"] == "bad thing"
+    # file sink exists and is JSONL
+    log_file = os.path.join(runtime.logger.log_dir, "agentainer.log")
+    with open(log_file) as f:
+        lines = [json.loads(line) for line in f]
+    assert len(lines) == 2
+
+
+def test_log_level_filtering(runtime):
+    runtime.logger.level = "warn"
+    assert runtime.logger.info("suppressed") == {}
+    assert runtime.logger.warn("kept")["message"] == "kept"
+    runtime.logger.level = "info"
+
+
+def test_audit_trail(runtime):
+    runtime.logger.audit("cli", "deploy", "agent-1", "success", ip="1.2.3.4")
+    runtime.logger.audit("cli", "remove", "agent-1", "failure")
+    entries = runtime.logger.get_audit_logs(action="deploy")
+    assert len(entries) == 1
+    assert entries[0]["user"] == "cli" and entries[0]["ip"] == "1.2.3.4"
+    assert len(runtime.logger.get_audit_logs()) == 2
+
+
+def test_tail_stream(runtime):
+    seen = []
+    unsub = runtime.logger.tail(seen.append)
+    runtime.logger.info("streamed")
+    assert len(seen) == 1 and json.loads(seen[0])["message"] == "streamed"
+    unsub()
+
+
+def test_per_agent_log(runtime):
+    a = runtime.agents.deploy(name="la", model="echo")
+    runtime.logger.info("agent event", agent_id=a.id)
+    entries = runtime.agents.get_logs(a.id)
+    assert len(entries) == 1 and entries[0]["message"] == "agent event"
+
+
+def test_backup_restore_roundtrip(runtime):
+    a = runtime.agents.deploy(name="ba", model="echo", auto_restart=True,
+                              system_prompt="be nice")
+    runtime.agents.start(a.id)
+    runtime.agent_request(a.id, "POST", "/chat", body={"message": "remember me"})
+    b = runtime.backups.create("snap1", "test backup")
+    assert len(b["agents"]) == 1
+    listed = runtime.backups.list()
+    assert listed and listed[0]["id"] == b["id"]
+    restored = runtime.backups.restore(b["id"])
+    assert len(restored) == 1
+    r = restored[0]
+    assert r.name == "ba-restored"  # {name}-restored (manager.go:132-191)
+    assert r.system_prompt == "be nice"
+    hist = runtime.store.lrange(f"agent:{r.id}:conversations")
+    assert len(hist) == 1 and hist[0]["user"] == "remember me"
+
+
+def test_backup_export_import(runtime, tmp_path):
+    runtime.agents.deploy(name="ex", model="echo")
+    b = runtime.backups.create("exported")
+    out = str(tmp_path / "bundle.tar.gz")
+    runtime.backups.export(b["id"], out)
+    runtime.backups.delete(b["id"])
+    assert runtime.backups.list() == []
+    ids = runtime.backups.import_(out)
+    assert ids == [b["id"]]
+    assert runtime.backups.load(b["id"])["name"] == "exported"

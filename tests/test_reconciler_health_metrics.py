@@ -1,0 +1,117 @@
+"""Reconciler convergence, health monitor auto-restart, metrics history."""
+
+import time
+
+from agentainer_amd.registry import FAILED, PAUSED, RUNNING, STOPPED
+
+
+def _started(rt, name="r1", **kw):
+    a = rt.agents.deploy(name=name, model="echo", **kw)
+    rt.agents.start(a.id)
+    return a
+
+
+# ---------- reconciler ----------
+
+def test_reconciler_adopts_engine_truth(runtime):
+    a = _started(runtime)
+    # engine-side pause without going through the manager
+    runtime.engine.pause(a.id)
+    runtime.reconciler.sync_agent(a.id)
+    assert runtime.agents.get(a.id).status == PAUSED
+    runtime.engine.unpause(a.id)
+    runtime.reconciler.sync_agent(a.id)
+    assert runtime.agents.get(a.id).status == RUNNING
+
+
+def test_reconciler_marks_vanished_stopped(runtime):
+    a = _started(runtime)
+    runtime.engine.detach(a.id, offload_kv=False)  # attachment vanishes
+    runtime.reconciler.sync_agent(a.id)
+    assert runtime.agents.get(a.id).status == STOPPED
+
+
+def test_reconciler_auto_restarts(runtime):
+    a = _started(runtime, auto_restart=True)
+    runtime.engine.detach(a.id, offload_kv=False)
+    runtime.reconciler.sync_agent(a.id)
+    assert runtime.agents.get(a.id).status == RUNNING  # restart-policy analog
+    assert runtime.engine.is_attached(a.id)
+
+
+def test_reconciler_detaches_orphans(runtime):
+    a = _started(runtime)
+    # delete the registry record behind the manager's back
+    runtime.store.delete(f"agent:{a.id}")
+    runtime.store.srem("agents:list", a.id)
+    runtime.reconciler.sync_all()
+    assert not runtime.engine.is_attached(a.id)
+
+
+# ---------- health ----------
+
+def test_health_probe_and_status(runtime):
+    a = _started(runtime)
+    runtime.health.start_monitoring(a.id, {"interval": 0.01, "retries": 3})
+    st = runtime.health.check_one(a.id)
+    assert st["healthy"] is True
+    stored = runtime.health.get_status(a.id)
+    assert stored["healthy"] is True
+    assert a.id in runtime.health.get_all_statuses()
+
+
+def test_health_auto_restart_after_retries(runtime):
+    a = _started(runtime, auto_restart=True)
+    runtime.health.start_monitoring(a.id, {"interval": 0.0, "retries": 3})
+    runtime.engine.crash()
+    # first 2 failures: no restart yet
+    runtime.health.check_one(a.id)
+    runtime.health.check_one(a.id)
+    assert runtime.health.get_status(a.id)["consecutive_failures"] == 2
+    runtime.engine.recover()  # allow the restart's attach to succeed
+    runtime.health.check_one(a.id)  # 3rd failure => restart
+    assert runtime.engine.is_attached(a.id)
+    assert runtime.agents.get(a.id).status == RUNNING
+
+
+def test_health_event_driven_registration(runtime):
+    """Status events (pattern pub/sub) register/deregister monitors — the
+    fixed version of the reference's broken Subscribe."""
+    runtime.health.start()
+    try:
+        a = _started(runtime, name="ev")
+        assert a.id in runtime.health.monitored_ids()
+        runtime.agents.stop(a.id)
+        assert a.id not in runtime.health.monitored_ids()
+    finally:
+        runtime.health.stop()
+
+
+# ---------- metrics ----------
+
+def test_metrics_sample_and_history(runtime):
+    a = _started(runtime)
+    for i in range(3):
+        runtime.agent_request(a.id, "POST", "/chat", body={"message": f"m{i}"})
+    now = time.time()
+    s1 = runtime.metrics.sample_agent(
+        a.id, runtime.engine.stats()["agents"][a.id], now=now)
+    assert s1["requests_total"] == 3
+    assert s1["e2e_p50_s"] >= 0
+    cur = runtime.metrics.get_metrics(a.id)
+    assert cur["requests_total"] == 3
+    runtime.agent_request(a.id, "POST", "/chat", body={"message": "m3"})
+    s2 = runtime.metrics.sample_agent(
+        a.id, runtime.engine.stats()["agents"][a.id], now=now + 10)
+    assert s2["req_per_s"] > 0
+    hist = runtime.metrics.get_metrics_history(a.id, duration_s=3600, now=now + 11)
+    assert len(hist) == 2
+
+
+def test_metrics_history_trim(runtime):
+    a = _started(runtime)
+    old = time.time() - 25 * 3600
+    runtime.metrics.sample_agent(a.id, {"tokens": 1, "requests": 1}, now=old)
+    runtime.metrics.sample_agent(a.id, {"tokens": 2, "requests": 2}, now=time.time())
+    hist = runtime.metrics.get_metrics_history(a.id, duration_s=48 * 3600)
+    assert len(hist) == 1  # 24h retention trim
