@@ -1,0 +1,225 @@
+"""Paged KV-cache manager — HBM3E page pool + pinned-host checkpoints.
+
+The MI355X replacement for the reference's "container keeps its
+filesystem + Redis state" durability (SURVEY.md §2.3 "KV-cache manager"
+row, §5 Checkpoint/resume):
+
+  * One page pool per model layer in HBM (layouts match the attention
+    kernels: k [P, n_kv, D/8, PS, 8], v [P, n_kv, PS, D]).
+  * Sequences (= agent conversations) own page lists via a page table.
+  * stop/offload gathers a sequence's pages into a contiguous GPU staging
+    buffer (ops.gather_kv_pages) and streams it to PINNED host memory with
+    a non-blocking copy (hipMemcpyAsync under torch); resume streams it
+    back and rebuilds the page table. Checkpoints can also be serialized
+    to disk for backup/export.
+
+Sized for 288 GB HBM3E: by default the pool takes the configured fraction
+of free device memory after weights.
+"""
+
+from __future__ import annotations
+
+import threading
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .. import ops
+
+
+class OutOfPages(RuntimeError):
+    pass
+
+
+@dataclass
+class Sequence:
+    seq_id: str
+    pages: List[int] = field(default_factory=list)
+    length: int = 0  # tokens written
+
+
+@dataclass
+class KVCheckpoint:
+    """Host-side image of one sequence's KV state."""
+
+    length: int
+    n_pages: int
+    data: torch.Tensor  # pinned host [n_layers, n_pages, page_shorts] bf16
+
+
+class KVCacheManager:
+    def __init__(self, n_layers: int, n_kv: int, head_dim: int, page_size: int,
+                 n_pages: int, device="cpu", dtype=torch.bfloat16):
+        assert head_dim % 8 == 0
+        self.n_layers = n_layers
+        self.n_kv = n_kv
+        self.head_dim = head_dim
+        self.page_size = page_size
+        self.n_pages = n_pages
+        self.device = device
+        self.dtype = dtype
+        self.k_caches: List[torch.Tensor] = []
+        self.v_caches: List[torch.Tensor] = []
+        for _ in range(n_layers):
+            self.k_caches.append(torch.zeros(
+                n_pages, n_kv, head_dim // 8, page_size, 8, dtype=dtype, device=device))
+            self.v_caches.append(torch.zeros(
+                n_pages, n_kv, page_size, head_dim, dtype=dtype, device=device))
+        self._free: List[int] = list(range(n_pages - 1, -1, -1))
+        self._seqs: Dict[str, Sequence] = {}
+        self._lock = threading.RLock()
+        # shorts per page per layer: K plane + V plane
+        self.page_shorts = 2 * n_kv * head_dim * page_size
+
+    # ---------- queries ----------
+
+    @property
+    def free_pages(self) -> int:
+        with self._lock:
+            return len(self._free)
+
+    @property
+    def used_pages(self) -> int:
+        return self.n_pages - self.free_pages
+
+    def kv_caches(self) -> List[Tuple[torch.Tensor, torch.Tensor]]:
+        return list(zip(self.k_caches, self.v_caches))
+
+    def has_seq(self, seq_id: str) -> bool:
+        with self._lock:
+            return seq_id in self._seqs
+
+    def seq_len(self, seq_id: str) -> int:
+        with self._lock:
+            s = self._seqs.get(seq_id)
+            return s.length if s else 0
+
+    def seq_bytes(self, seq_id: str) -> int:
+        with self._lock:
+            s = self._seqs.get(seq_id)
+            if not s:
+                return 0
+            return len(s.pages) * self.page_shorts * self.n_layers * 2
+
+    # ---------- allocation ----------
+
+    def create_seq(self, seq_id: str) -> Sequence:
+        with self._lock:
+            if seq_id in self._seqs:
+                raise ValueError(f"sequence {seq_id} exists")
+            s = Sequence(seq_id)
+            self._seqs[seq_id] = s
+            return s
+
+    def free_seq(self, seq_id: str) -> None:
+        with self._lock:
+            s = self._seqs.pop(seq_id, None)
+            if s:
+                self._free.extend(reversed(s.pages))
+
+    def reset_seq(self, seq_id: str) -> None:
+        """Drop a sequence's KV but keep it registered (context truncation)."""
+        with self._lock:
+            s = self._seqs[seq_id]
+            self._free.extend(reversed(s.pages))
+            s.pages = []
+            s.length = 0
+
+    def can_append(self, seq_id: str, n_tokens: int) -> bool:
+        with self._lock:
+            s = self._seqs[seq_id]
+            have = len(s.pages) * self.page_size - s.length
+            need_pages = max(0, -(-(n_tokens - have) // self.page_size))
+            return need_pages <= len(self._free)
+
+    def append_slots(self, seq_id: str, n_tokens: int) -> List[int]:
+        """Reserve slots for n_tokens new tokens; allocates pages as needed.
+        Returns global slot ids (page * page_size + offset)."""
+        with self._lock:
+            s = self._seqs[seq_id]
+            slots = []
+            for _ in range(n_tokens):
+                if s.length == len(s.pages) * self.page_size:
+                    if not self._free:
+                        raise OutOfPages(
+                            f"KV pool exhausted ({self.n_pages} pages)")
+                    s.pages.append(self._free.pop())
+                page = s.pages[s.length // self.page_size]
+                off = s.length % self.page_size
+                slots.append(page * self.page_size + off)
+                s.length += 1
+            return slots
+
+    def page_table(self, seq_ids: List[str], device=None) -> torch.Tensor:
+        """int32 [B, max_pages_in_batch] (padded with 0)."""
+        with self._lock:
+            rows = [self._seqs[sid].pages for sid in seq_ids]
+        width = max((len(r) for r in rows), default=1) or 1
+        t = torch.zeros(len(rows), width, dtype=torch.int32)
+        for i, r in enumerate(rows):
+            if r:
+                t[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
+        return t.to(device or self.device)
+
+    def seq_lens(self, seq_ids: List[str], device=None) -> torch.Tensor:
+        with self._lock:
+            lens = [self._seqs[sid].length for sid in seq_ids]
+        return torch.tensor(lens, dtype=torch.int32, device=device or self.device)
+
+    # ---------- checkpoint (stop/resume, crash recovery) ----------
+
+    def offload(self, seq_id: str, free: bool = True) -> Optional[KVCheckpoint]:
+        """Gather the sequence's pages and stream them to pinned host memory."""
+        with self._lock:
+            s = self._seqs.get(seq_id)
+            if s is None or not s.pages:
+                if free and s is not None:
+                    self.free_seq(seq_id)
+                return None
+            pages = list(s.pages)
+            length = s.length
+        n = len(pages)
+        page_ids = torch.tensor(pages, dtype=torch.int32, device=self.device)
+        is_gpu = str(self.device) not in ("cpu",)
+        host = torch.empty(self.n_layers, n, self.page_shorts, dtype=self.dtype,
+                           pin_memory=is_gpu)
+        if is_gpu:
+            staging = torch.empty(n, self.page_shorts, dtype=self.dtype,
+                                  device=self.device)
+            for li in range(self.n_layers):
+                ops.gather_kv_pages(staging, self.k_caches[li], self.v_caches[li],
+                                    page_ids)
+                host[li].copy_(staging, non_blocking=True)
+            torch.cuda.synchronize()
+        else:
+            staging = torch.empty(n, self.page_shorts, dtype=self.dtype)
+            for li in range(self.n_layers):
+                ops.gather_kv_pages(staging, self.k_caches[li], self.v_caches[li],
+                                    page_ids)
+                host[li].copy_(staging)
+        if free:
+            self.free_seq(seq_id)
+        return KVCheckpoint(length=length, n_pages=n, data=host)
+
+    def restore(self, seq_id: str, ckpt: Optional[KVCheckpoint]) -> None:
+        """Upload a checkpoint into freshly allocated pages."""
+        with self._lock:
+            if seq_id in self._seqs:
+                self.free_seq(seq_id)
+            s = self.create_seq(seq_id)
+            if ckpt is None or ckpt.n_pages == 0:
+                return
+            if len(self._free) < ckpt.n_pages:
+                raise OutOfPages("not enough pages to restore checkpoint")
+            s.pages = [self._free.pop() for _ in range(ckpt.n_pages)]
+            s.length = ckpt.length
+            pages = list(s.pages)
+        page_ids = torch.tensor(pages, dtype=torch.int32, device=self.device)
+        is_gpu = str(self.device) not in ("cpu",)
+        for li in range(self.n_layers):
+            staging = ckpt.data[li].to(self.device, non_blocking=is_gpu)
+            ops.scatter_kv_pages(self.k_caches[li], self.v_caches[li],
+                                 staging, page_ids)
+        if is_gpu:
+            torch.cuda.synchronize()

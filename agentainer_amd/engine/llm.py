@@ -1,0 +1,652 @@
+"""LLMEngine — the in-process multi-tenant inference engine.
+
+This is what replaces the reference's Docker data plane (SURVEY.md north
+star): each attached agent is a (shared) model binding plus a private
+conversation KV sequence in the paged HBM pool. The continuous-batching
+scheduler below is the analog of the reference's request queue + replay
+worker feeding containers (replay_worker.go:57-117): admission comes from
+the per-model request queue, prefill is batched under a token budget,
+decode runs one batched step for every running sequence.
+
+Structure per model (agents SHARE one weight set — BASELINE config 3):
+  ModelInstance: weights + KVCacheManager + waiting/running queues +
+  one engine thread running steps.
+
+Lifecycle mapping:
+  attach   = bind shard (refcounted load) + create/restore KV sequence
+  pause    = close admission; KV stays resident
+  detach   = drain + offload KV to pinned host (+ disk for restarts)
+  chat     = enqueue -> scheduler -> response (greedy => deterministic, so
+             at-least-once crash replay regenerates identical output)
+"""
+
+from __future__ import annotations
+
+import os
+import queue
+import threading
+import time
+import traceback
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from .. import ops
+from ..store import Store
+from .base import ModelNotFound
+from .kvcache import KVCacheManager, KVCheckpoint, OutOfPages
+from .tokenizer import ByteTokenizer
+from ..models.llama import (AttnMetadata, LLAMA_CONFIGS, LlamaConfig,
+                            LlamaForCausalLM)
+
+DEFAULT_MAX_NEW = 64
+
+
+class EngineDead(RuntimeError):
+    pass
+
+
+@dataclass
+class GenRequest:
+    agent_id: str
+    prompt_tokens: List[int]
+    max_new: int
+    temperature: float
+    top_p: float
+    seed: int
+    done: threading.Event = field(default_factory=threading.Event)
+    generated: List[int] = field(default_factory=list)
+    error: Optional[str] = None
+    enq_t: float = 0.0
+    first_token_t: float = 0.0
+    fin_t: float = 0.0
+
+
+@dataclass
+class AgentBinding:
+    agent: Any
+    model_name: str
+    seq_id: str
+    paused: bool = False
+    requests: int = 0
+    tokens: int = 0
+    queue: "queue.Queue[GenRequest]" = field(default_factory=queue.Queue)
+    active: Optional[GenRequest] = None
+
+
+class ModelInstance:
+    """One loaded model + KV pool + scheduler thread."""
+
+    def __init__(self, name: str, cfg: LlamaConfig, device: str,
+                 engine_cfg: Dict[str, Any], weights_path: Optional[str] = None):
+        self.name = name
+        self.cfg = cfg
+        self.device = device
+        self.tokenizer = ByteTokenizer(cfg.vocab_size)
+        self.model = LlamaForCausalLM(cfg, device=device)
+        if weights_path:
+            self.model.load_safetensors(weights_path)
+        page_size = int(engine_cfg.get("kv_page_size", 16))
+        n_pages = self._pool_pages(cfg, page_size, device, engine_cfg)
+        self.kvm = KVCacheManager(cfg.n_layers, cfg.n_kv_heads, cfg.head_dim,
+                                  page_size, n_pages, device=device)
+        self.max_batch_tokens = int(engine_cfg.get("max_batch_tokens", 8192))
+        self.max_decode_batch = int(engine_cfg.get("max_decode_batch", 256))
+        self.refcount = 0
+        self.waiting: "queue.Queue[GenRequest]" = queue.Queue()
+        self.running: List[GenRequest] = []
+        self._bindings: Dict[str, AgentBinding] = {}
+        self._lock = threading.RLock()
+        self._stop = threading.Event()
+        self._wake = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self.steps = 0
+        self.decode_tokens = 0
+        self.prefill_tokens = 0
+        self.occupancy_acc = 0.0
+
+    @staticmethod
+    def _pool_pages(cfg: LlamaConfig, page_size: int, device: str,
+                    engine_cfg: Dict[str, Any]) -> int:
+        bytes_per_page = cfg.kv_bytes_per_token() * page_size
+        pool_gb = float(engine_cfg.get("kv_pool_gb", 0.0))
+        if device.startswith("cuda") and pool_gb <= 0:
+            free, _total = torch.cuda.mem_get_info()
+            pool_bytes = int(free * 0.80)  # leave headroom for activations
+        elif pool_gb > 0:
+            pool_bytes = int(pool_gb * (1 << 30))
+        else:
+            pool_bytes = 64 << 20  # CPU tests: 64 MB
+        return max(8, pool_bytes // bytes_per_page)
+
+    # ---------- scheduler thread ----------
+
+    def start(self):
+        if self._thread is None:
+            self._stop.clear()
+            self._thread = threading.Thread(target=self._loop,
+                                            name=f"engine-{self.name}", daemon=True)
+            self._thread.start()
+
+    def stop(self):
+        self._stop.set()
+        self._wake.set()
+        if self._thread is not None:
+            self._thread.join(timeout=10.0)
+            self._thread = None
+
+    def alive(self) -> bool:
+        return self._thread is not None and self._thread.is_alive()
+
+    def _loop(self):
+        while not self._stop.is_set():
+            try:
+                did = self.step()
+            except Exception:
+                traceback.print_exc()
+                did = False
+            if not did:
+                self._wake.wait(timeout=0.005)
+                self._wake.clear()
+
+    # ---------- agent-side API ----------
+
+    def submit(self, req: GenRequest):
+        req.enq_t = time.time()
+        self.waiting.put(req)
+        self._wake.set()
+
+    # ---------- one engine step ----------
+
+    def step(self) -> bool:
+        """Admit + prefill, then one decode step. Returns True if work ran."""
+        admitted = self._admit()
+        if admitted:
+            self._prefill(admitted)
+        ran_decode = False
+        with self._lock:
+            batch = [r for r in self.running if not r.done.is_set()]
+        if batch:
+            self._decode(batch[: self.max_decode_batch])
+            ran_decode = True
+        self.steps += 1 if (admitted or ran_decode) else 0
+        return bool(admitted) or ran_decode
+
+    def _admit(self) -> List[GenRequest]:
+        out: List[GenRequest] = []
+        budget = self.max_batch_tokens
+        while budget > 0:
+            try:
+                req = self.waiting.get_nowait()
+            except queue.Empty:
+                break
+            b = self._bindings.get(req.agent_id)
+            if b is None:
+                req.error = "agent detached"
+                req.done.set()
+                continue
+            need = len(req.prompt_tokens)
+            if need > self.max_batch_tokens:
+                req.error = f"prompt too long ({need} tokens)"
+                req.done.set()
+                continue
+            if need > budget and out:
+                # put back; try next step
+                self.waiting.put(req)
+                break
+            # KV room: prompt + generation
+            if not self.kvm.can_append(b.seq_id, need + req.max_new):
+                # truncate conversation: reset KV, re-prefill just this turn
+                self.kvm.reset_seq(b.seq_id)
+                if not self.kvm.can_append(b.seq_id, need + req.max_new):
+                    req.error = "KV pool exhausted"
+                    req.done.set()
+                    continue
+            out.append(req)
+            budget -= need
+        return out
+
+    def _sample(self, logits: torch.Tensor, reqs: List[GenRequest]) -> List[int]:
+        B = logits.size(0)
+        out = torch.empty(B, dtype=torch.long, device=logits.device)
+        greedy_rows = [i for i, r in enumerate(reqs) if r.temperature <= 0.0]
+        samp_rows = [i for i, r in enumerate(reqs) if r.temperature > 0.0]
+        lb = logits.to(torch.bfloat16).contiguous()
+        if greedy_rows:
+            if len(greedy_rows) == B:
+                ops.greedy_sample(out, lb)
+            else:
+                sub = torch.empty(len(greedy_rows), dtype=torch.long,
+                                  device=logits.device)
+                ops.greedy_sample(sub, lb[greedy_rows].contiguous())
+                out[greedy_rows] = sub
+        if samp_rows:
+            sub = torch.empty(len(samp_rows), dtype=torch.long, device=logits.device)
+            temps = torch.tensor([reqs[i].temperature for i in samp_rows],
+                                 dtype=torch.float32, device=logits.device)
+            tps = torch.tensor([reqs[i].top_p for i in samp_rows],
+                               dtype=torch.float32, device=logits.device)
+            seeds = torch.tensor(
+                [(reqs[i].seed + len(reqs[i].generated)) & 0x7FFFFFFFFFFFFFFF
+                 for i in samp_rows], dtype=torch.int64, device=logits.device)
+            ops.topp_sample(sub, lb[samp_rows].contiguous(), temps, tps, seeds)
+            out[samp_rows] = sub
+        return out.tolist()
+
+    def _prefill(self, reqs: List[GenRequest]):
+        dev = self.device
+        ids: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        q_starts, q_lens, seq_ids = [], [], []
+        for r in reqs:
+            b = self._bindings[r.agent_id]
+            prev = self.kvm.seq_len(b.seq_id)
+            n = len(r.prompt_tokens)
+            q_starts.append(len(ids))
+            q_lens.append(n)
+            seq_ids.append(b.seq_id)
+            ids.extend(r.prompt_tokens)
+            positions.extend(range(prev, prev + n))
+            slots.extend(self.kvm.append_slots(b.seq_id, n))
+        md = AttnMetadata(
+            page_table=self.kvm.page_table(seq_ids, device=dev),
+            seq_lens=self.kvm.seq_lens(seq_ids, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            is_prefill=True,
+            query_starts=torch.tensor(q_starts, dtype=torch.int32, device=dev),
+            query_lens=torch.tensor(q_lens, dtype=torch.int32, device=dev),
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        last_rows = torch.tensor([s + l - 1 for s, l in zip(q_starts, q_lens)],
+                                 dtype=torch.long, device=dev)
+        logits = self.model(input_ids, md, self.kvm.kv_caches(), last_rows)
+        toks = self._sample(logits, reqs)
+        now = time.time()
+        self.prefill_tokens += len(ids)
+        with self._lock:
+            for r, t in zip(reqs, toks):
+                r.generated.append(int(t))
+                r.first_token_t = now
+                self._finish_or_run(r, int(t))
+
+    def _decode(self, reqs: List[GenRequest]):
+        dev = self.device
+        ids, positions, slots, seq_ids = [], [], [], []
+        for r in reqs:
+            b = self._bindings[r.agent_id]
+            prev = self.kvm.seq_len(b.seq_id)
+            ids.append(r.generated[-1])
+            positions.append(prev)
+            slots.extend(self.kvm.append_slots(b.seq_id, 1))
+            seq_ids.append(b.seq_id)
+        md = AttnMetadata(
+            page_table=self.kvm.page_table(seq_ids, device=dev),
+            seq_lens=self.kvm.seq_lens(seq_ids, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            is_prefill=False,
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        logits = self.model(input_ids, md, self.kvm.kv_caches(), None)
+        toks = self._sample(logits, reqs)
+        self.decode_tokens += len(reqs)
+        self.occupancy_acc += len(reqs) / max(1, self.max_decode_batch)
+        with self._lock:
+            for r, t in zip(reqs, toks):
+                r.generated.append(int(t))
+                self._finish_or_run(r, int(t))
+
+    def _finish_or_run(self, r: GenRequest, tok: int):
+        """Called with lock held, after appending tok."""
+        finished = (tok == self.tokenizer.eos_id or
+                    len(r.generated) >= r.max_new)
+        if r in self.running:
+            if finished:
+                self.running.remove(r)
+        elif not finished:
+            self.running.append(r)
+        if finished:
+            b = self._bindings.get(r.agent_id)
+            if b is not None:
+                b.requests += 1
+                b.tokens += len(r.generated)
+                b.active = None
+            r.fin_t = time.time()
+            r.done.set()
+            self._pump_agent(b)
+
+    def _pump_agent(self, b: Optional[AgentBinding]):
+        """Admit the agent's next queued request (per-agent FIFO — one
+        in-flight request per conversation sequence)."""
+        if b is None or b.paused or b.active is not None:
+            return
+        try:
+            nxt = b.queue.get_nowait()
+        except queue.Empty:
+            return
+        b.active = nxt
+        self.submit(nxt)
+
+    # ---------- binding management ----------
+
+    def bind(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint]):
+        with self._lock:
+            if ckpt is not None:
+                self.kvm.restore(seq_id, ckpt)
+            elif not self.kvm.has_seq(seq_id):
+                self.kvm.create_seq(seq_id)
+            self._bindings[agent.id] = AgentBinding(agent=agent,
+                                                    model_name=self.name,
+                                                    seq_id=seq_id)
+            self.refcount += 1
+
+    def unbind(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:
+        with self._lock:
+            b = self._bindings.pop(agent_id, None)
+            if b is None:
+                return None
+            self.refcount -= 1
+            # drain: fail queued requests (they stay pending in the WAL and
+            # will be replayed after resume — at-least-once contract)
+            if b.active is not None and not b.active.done.is_set():
+                b.active.error = "agent detached"
+                b.active.done.set()
+                if b.active in self.running:
+                    self.running.remove(b.active)
+            while True:
+                try:
+                    r = b.queue.get_nowait()
+                except queue.Empty:
+                    break
+                r.error = "agent detached"
+                r.done.set()
+        ckpt = self.kvm.offload(b.seq_id, free=True) if offload else None
+        if not offload:
+            self.kvm.free_seq(b.seq_id)
+        return ckpt
+
+    def binding(self, agent_id: str) -> Optional[AgentBinding]:
+        with self._lock:
+            return self._bindings.get(agent_id)
+
+
+class LLMEngine:
+    """EngineBackend over ModelInstances. Also serves the echo model so the
+    whole control plane runs against one backend."""
+
+    def __init__(self, store: Store, config=None, device: str = "cpu",
+                 state_root: Optional[str] = None):
+        self.store = store
+        self.device = device
+        self.engine_cfg = dict(config.get("engine")) if config else {}
+        self.state_root = state_root or os.path.expanduser("~/.agentainer_amd")
+        self._instances: Dict[str, ModelInstance] = {}
+        self._agent_model: Dict[str, str] = {}
+        self._ckpts: Dict[str, KVCheckpoint] = {}
+        self._lock = threading.RLock()
+        if device == "cuda" and torch.cuda.is_available():
+            if not ops.hip_available():
+                raise RuntimeError(
+                    "GPU engine requires the gfx950 HIP extension "
+                    "(python -m agentainer_amd.ops.build)")
+
+    # ---------- model registry ----------
+
+    def validate_model(self, model: str) -> None:
+        if model in LLAMA_CONFIGS:
+            return
+        from .echo import ECHO_MODELS
+        if model in ECHO_MODELS:
+            return
+        if os.path.isdir(os.path.expanduser(model)):
+            return  # weights directory
+        raise ModelNotFound(
+            f"unknown model {model!r}; known: {sorted(LLAMA_CONFIGS)} + echo")
+
+    def _get_instance(self, model: str) -> ModelInstance:
+        with self._lock:
+            inst = self._instances.get(model)
+            if inst is None:
+                if model in LLAMA_CONFIGS:
+                    cfg = LLAMA_CONFIGS[model]
+                    inst = ModelInstance(model, cfg, self.device, self.engine_cfg)
+                else:
+                    raise ModelNotFound(f"cannot load {model!r}")
+                self._instances[model] = inst
+                inst.start()
+            return inst
+
+    # ---------- EngineBackend ----------
+
+    def attach(self, agent) -> None:
+        model = agent.model
+        from .echo import ECHO_MODELS
+        if model in ECHO_MODELS:
+            # echo agents need no weights; emulate with tiny state
+            with self._lock:
+                self._agent_model[agent.id] = "echo"
+                self._ckpts.pop(agent.id, None)
+            self.store.hset(f"agent:{agent.id}:metrics", "engine", "echo")
+            self._echo_attached = getattr(self, "_echo_attached", {})
+            self._echo_attached[agent.id] = {"paused": False, "requests": 0,
+                                             "tokens": 0}
+            return
+        inst = self._get_instance(model)
+        ckpt = self._ckpts.pop(agent.id, None)
+        if ckpt is None:
+            ckpt = self._load_disk_ckpt(agent.id)
+        inst.bind(agent, seq_id=agent.id, ckpt=ckpt)
+        with self._lock:
+            self._agent_model[agent.id] = model
+
+    def detach(self, agent_id: str, offload_kv: bool = True) -> bool:
+        with self._lock:
+            model = self._agent_model.pop(agent_id, None)
+        if model is None:
+            return False
+        if model == "echo":
+            getattr(self, "_echo_attached", {}).pop(agent_id, None)
+            return bool(offload_kv)
+        inst = self._instances.get(model)
+        if inst is None:
+            return False
+        ckpt = inst.unbind(agent_id, offload=offload_kv)
+        if ckpt is not None:
+            self._ckpts[agent_id] = ckpt
+            self._save_disk_ckpt(agent_id, ckpt)
+            return True
+        return False
+
+    def pause(self, agent_id: str) -> None:
+        b = self._binding(agent_id)
+        if b is not None:
+            b.paused = True
+        eb = getattr(self, "_echo_attached", {}).get(agent_id)
+        if eb is not None:
+            eb["paused"] = True
+
+    def unpause(self, agent_id: str) -> None:
+        b = self._binding(agent_id)
+        if b is not None:
+            b.paused = False
+            inst = self._instances.get(self._agent_model.get(agent_id, ""))
+            if inst:
+                with inst._lock:
+                    inst._pump_agent(b)
+        eb = getattr(self, "_echo_attached", {}).get(agent_id)
+        if eb is not None:
+            eb["paused"] = False
+
+    def _binding(self, agent_id: str) -> Optional[AgentBinding]:
+        model = self._agent_model.get(agent_id)
+        if model in (None, "echo"):
+            return None
+        inst = self._instances.get(model)
+        return inst.binding(agent_id) if inst else None
+
+    def is_attached(self, agent_id: str) -> bool:
+        return agent_id in self._agent_model
+
+    def attached_ids(self) -> List[str]:
+        with self._lock:
+            return list(self._agent_model)
+
+    def engine_status(self, agent_id: str) -> str:
+        model = self._agent_model.get(agent_id)
+        if model is None:
+            return "missing"
+        if model == "echo":
+            eb = getattr(self, "_echo_attached", {}).get(agent_id)
+            return "paused" if (eb and eb["paused"]) else "running"
+        b = self._binding(agent_id)
+        if b is None:
+            return "missing"
+        return "paused" if b.paused else "running"
+
+    def health_probe(self, agent_id: str) -> bool:
+        model = self._agent_model.get(agent_id)
+        if model is None:
+            return False
+        if model == "echo":
+            return True
+        inst = self._instances.get(model)
+        return inst is not None and inst.alive()
+
+    # ---------- chat ----------
+
+    def chat(self, agent_id: str, message: str, **kwargs: Any) -> Dict[str, Any]:
+        from ..wal import EngineUnavailable
+
+        model = self._agent_model.get(agent_id)
+        if model is None:
+            raise EngineUnavailable(f"agent {agent_id} is not attached")
+        if model == "echo":
+            return self._echo_chat(agent_id, message)
+        inst = self._instances.get(model)
+        b = inst.binding(agent_id) if inst else None
+        if inst is None or b is None or not inst.alive():
+            raise EngineUnavailable(f"engine for {agent_id} is down")
+        if b.paused:
+            raise EngineUnavailable(f"agent {agent_id} is paused")
+        agent = b.agent
+        sampling = dict(agent.sampling or {})
+        sampling.update(kwargs)
+        prompt = self._build_prompt(agent, message)
+        req = GenRequest(
+            agent_id=agent_id,
+            prompt_tokens=inst.tokenizer.encode(prompt),
+            max_new=int(sampling.get("max_tokens", DEFAULT_MAX_NEW)),
+            temperature=float(sampling.get("temperature", 0.0)),
+            top_p=float(sampling.get("top_p", 1.0)),
+            seed=int(sampling.get("seed", 0)),
+        )
+        with inst._lock:
+            b.queue.put(req)
+            inst._pump_agent(b)
+        timeout = float(sampling.get("timeout_s", 120.0))
+        if not req.done.wait(timeout):
+            raise EngineUnavailable(f"generation timed out after {timeout}s")
+        if req.error:
+            raise EngineUnavailable(req.error)
+        text = inst.tokenizer.decode(req.generated)
+        hist_key = f"agent:{agent_id}:conversations"
+        self.store.rpush(hist_key, {"user": message, "assistant": text,
+                                    "ts": time.time()})
+        n = self.store.llen(hist_key)
+        if n > 50:
+            self.store.ltrim(hist_key, n - 50, -1)
+        self.store.hset(f"agent:{agent_id}:metrics", "total_requests", b.requests)
+        return {
+            "response": text,
+            "model": model,
+            "tokens": len(req.generated),
+            "ttft_s": (req.first_token_t - req.enq_t) if req.first_token_t else None,
+            "e2e_s": (req.fin_t - req.enq_t) if req.fin_t else None,
+        }
+
+    def _build_prompt(self, agent, message: str) -> str:
+        """This turn's incremental prompt; earlier turns are already in the
+        agent's KV sequence. Redis-analog history provides the replayable
+        text context (examples/gpt-agent/app.py:89-92 last-3 pattern)."""
+        parts = []
+        if agent.system_prompt and self.store.llen(
+                f"agent:{agent.id}:conversations") == 0:
+            parts.append(f"[system] {agent.system_prompt}\n")
+        parts.append(f"[user] {message}\n[assistant] ")
+        return "".join(parts)
+
+    def _echo_chat(self, agent_id: str, message: str) -> Dict[str, Any]:
+        eb = getattr(self, "_echo_attached", {}).get(agent_id)
+        from ..wal import EngineUnavailable
+        if eb is None:
+            raise EngineUnavailable(f"agent {agent_id} not attached")
+        if eb["paused"]:
+            raise EngineUnavailable(f"agent {agent_id} is paused")
+        hist_key = f"agent:{agent_id}:conversations"
+        ctx = self.store.lrange(hist_key, -3, -1)
+        reply = f"echo({len(ctx)}): {message}"
+        self.store.rpush(hist_key, {"user": message, "assistant": reply,
+                                    "ts": time.time()})
+        eb["requests"] += 1
+        eb["tokens"] += len(reply.split())
+        return {"response": reply, "model": "echo", "tokens": eb["tokens"]}
+
+    # ---------- checkpoint persistence (server-restart recovery) ----------
+
+    def _ckpt_path(self, agent_id: str) -> str:
+        d = os.path.join(self.state_root, "kv_ckpt")
+        os.makedirs(d, exist_ok=True)
+        return os.path.join(d, f"{agent_id}.pt")
+
+    def _save_disk_ckpt(self, agent_id: str, ckpt: KVCheckpoint) -> None:
+        torch.save({"length": ckpt.length, "n_pages": ckpt.n_pages,
+                    "data": ckpt.data.cpu()}, self._ckpt_path(agent_id))
+
+    def _load_disk_ckpt(self, agent_id: str) -> Optional[KVCheckpoint]:
+        p = self._ckpt_path(agent_id)
+        if not os.path.exists(p):
+            return None
+        try:
+            d = torch.load(p, map_location="cpu", weights_only=True)
+            return KVCheckpoint(length=d["length"], n_pages=d["n_pages"],
+                                data=d["data"])
+        except Exception:
+            return None
+
+    # ---------- stats ----------
+
+    def stats(self) -> Dict[str, Any]:
+        agents: Dict[str, Any] = {}
+        for aid, st in getattr(self, "_echo_attached", {}).items():
+            agents[aid] = {"requests": st["requests"], "tokens": st["tokens"],
+                           "paused": st["paused"]}
+        models = {}
+        for name, inst in self._instances.items():
+            with inst._lock:
+                for aid, b in inst._bindings.items():
+                    agents[aid] = {
+                        "requests": b.requests, "tokens": b.tokens,
+                        "paused": b.paused,
+                        "kv_bytes": inst.kvm.seq_bytes(b.seq_id),
+                        "kv_pages": len(inst.kvm._seqs.get(b.seq_id).pages)
+                        if inst.kvm.has_seq(b.seq_id) else 0,
+                        "batch_occupancy": (inst.occupancy_acc / inst.steps)
+                        if inst.steps else 0.0,
+                    }
+            models[name] = {
+                "steps": inst.steps,
+                "decode_tokens": inst.decode_tokens,
+                "prefill_tokens": inst.prefill_tokens,
+                "kv_pages_used": inst.kvm.used_pages,
+                "kv_pages_free": inst.kvm.free_pages,
+                "running": len(inst.running),
+            }
+        return {"engine": "llm", "device": self.device, "agents": agents,
+                "models": models}
+
+    def shutdown(self):
+        for inst in self._instances.values():
+            inst.stop()

@@ -1,0 +1,260 @@
+"""Llama-3 family — MI355X-first serving forward.
+
+The model is written for the paged multi-tenant engine, not as a generic
+HF module: every hot op dispatches to the gfx950 HIP kernels
+(agentainer_amd.ops), plain GEMMs go to hipBLASLt via F.linear, and the
+forward takes paged-KV attention metadata (prefill: varlen packed rows;
+decode: one row per sequence). Weights are bf16, random-init by default
+(BASELINE.json: synthetic prompts / random-init weights; no network for
+checkpoints) with a safetensors loader for real weights.
+
+Tensor-parallel sharding (TP over RCCL/xGMI) slices n_heads/intermediate
+by rank: column-parallel qkv/gate/up, row-parallel o/down with all-reduce
+(agentainer_amd.parallel).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "tiny-llama"
+    vocab_size: int = 512
+    hidden_size: int = 512
+    n_layers: int = 2
+    n_heads: int = 4
+    n_kv_heads: int = 1
+    head_dim: int = 128
+    intermediate_size: int = 1024
+    rope_theta: float = 500000.0
+    max_position: int = 8192
+    norm_eps: float = 1e-5
+    tie_embeddings: bool = True
+
+    @property
+    def q_size(self):
+        return self.n_heads * self.head_dim
+
+    @property
+    def kv_size(self):
+        return self.n_kv_heads * self.head_dim
+
+    def kv_bytes_per_token(self, dtype_bytes: int = 2) -> int:
+        return 2 * self.n_layers * self.n_kv_heads * self.head_dim * dtype_bytes
+
+
+@dataclass
+class AttnMetadata:
+    """Paged-KV attention metadata for one engine step."""
+
+    page_table: torch.Tensor      # [B, max_pages] int32
+    seq_lens: torch.Tensor        # [B] int32 — total context length per seq
+    slot_mapping: torch.Tensor    # [T] int64 — where each new token's K/V goes
+    positions: torch.Tensor       # [T] int32
+    is_prefill: bool = False
+    query_starts: Optional[torch.Tensor] = None  # [B] int32 (prefill)
+    query_lens: Optional[torch.Tensor] = None    # [B] int32 (prefill)
+
+
+class LlamaAttention(torch.nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp_rank: int = 0, tp_size: int = 1):
+        super().__init__()
+        assert cfg.n_heads % tp_size == 0, "n_heads must divide tp_size"
+        assert cfg.n_kv_heads % tp_size == 0 or tp_size == 1, \
+            "n_kv_heads must divide tp_size"
+        self.cfg = cfg
+        self.tp_size = tp_size
+        self.n_heads = cfg.n_heads // tp_size
+        self.n_kv = max(cfg.n_kv_heads // tp_size, 1)
+        self.head_dim = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        q_out = self.n_heads * cfg.head_dim
+        kv_out = self.n_kv * cfg.head_dim
+        self.qkv_proj = torch.nn.Parameter(
+            torch.empty(q_out + 2 * kv_out, cfg.hidden_size, dtype=torch.bfloat16))
+        self.o_proj = torch.nn.Parameter(
+            torch.empty(cfg.hidden_size, q_out, dtype=torch.bfloat16))
+
+    def forward(self, h, k_cache, v_cache, md: AttnMetadata, cos_sin, tp_group=None):
+        T = h.size(0)
+        qkv = F.linear(h, self.qkv_proj)
+        q_sz = self.n_heads * self.head_dim
+        kv_sz = self.n_kv * self.head_dim
+        q = qkv[:, :q_sz].view(T, self.n_heads, self.head_dim).contiguous()
+        k = qkv[:, q_sz:q_sz + kv_sz].view(T, self.n_kv, self.head_dim).contiguous()
+        v = qkv[:, q_sz + kv_sz:].view(T, self.n_kv, self.head_dim).contiguous()
+        ops.rope_inplace(q, k, cos_sin, md.positions)
+        ops.kv_append(k_cache, v_cache, k, v, md.slot_mapping)
+        out = torch.empty_like(q)
+        if md.is_prefill:
+            ops.paged_prefill_attention(out, q, k_cache, v_cache, md.page_table,
+                                        md.seq_lens, md.query_starts,
+                                        md.query_lens, self.scale)
+        else:
+            ops.paged_decode_attention(out, q, k_cache, v_cache, md.page_table,
+                                       md.seq_lens, self.scale)
+        o = F.linear(out.view(T, q_sz), self.o_proj)
+        if tp_group is not None:
+            torch.distributed.all_reduce(o, group=tp_group)
+        return o
+
+
+class LlamaMLP(torch.nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp_size: int = 1):
+        super().__init__()
+        inter = cfg.intermediate_size // tp_size
+        self.gate_up = torch.nn.Parameter(
+            torch.empty(2 * inter, cfg.hidden_size, dtype=torch.bfloat16))
+        self.down = torch.nn.Parameter(
+            torch.empty(cfg.hidden_size, inter, dtype=torch.bfloat16))
+        self.inter = inter
+
+    def forward(self, h, tp_group=None):
+        gu = F.linear(h, self.gate_up)
+        gate, up = gu[:, :self.inter].contiguous(), gu[:, self.inter:].contiguous()
+        act = torch.empty_like(gate)
+        ops.silu_mul(act, gate, up)
+        out = F.linear(act, self.down)
+        if tp_group is not None:
+            torch.distributed.all_reduce(out, group=tp_group)
+        return out
+
+
+class LlamaLayer(torch.nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp_rank: int = 0, tp_size: int = 1):
+        super().__init__()
+        self.attn = LlamaAttention(cfg, tp_rank, tp_size)
+        self.mlp = LlamaMLP(cfg, tp_size)
+        self.input_ln = torch.nn.Parameter(torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+        self.post_ln = torch.nn.Parameter(torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+        self.eps = cfg.norm_eps
+
+    def forward(self, hidden, residual, k_cache, v_cache, md, cos_sin, tp_group=None):
+        normed = torch.empty_like(hidden)
+        if residual is None:
+            residual = hidden.clone()
+            ops.rmsnorm(normed, hidden, self.input_ln, self.eps)
+        else:
+            ops.fused_add_rmsnorm(normed, hidden, residual, self.input_ln, self.eps)
+        attn_out = self.attn(normed, k_cache, v_cache, md, cos_sin, tp_group)
+        normed2 = torch.empty_like(hidden)
+        ops.fused_add_rmsnorm(normed2, attn_out, residual, self.post_ln, self.eps)
+        mlp_out = self.mlp(normed2, tp_group)
+        return mlp_out, residual
+
+
+class LlamaForCausalLM(torch.nn.Module):
+    def __init__(self, cfg: LlamaConfig, device="cpu", tp_rank: int = 0,
+                 tp_size: int = 1, seed: int = 0):
+        super().__init__()
+        self.cfg = cfg
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.tp_group = None
+        self.embed = torch.nn.Parameter(
+            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+        self.layers = torch.nn.ModuleList(
+            [LlamaLayer(cfg, tp_rank, tp_size) for _ in range(cfg.n_layers)])
+        self.final_ln = torch.nn.Parameter(torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+        if cfg.tie_embeddings:
+            self.lm_head = self.embed
+        else:
+            self.lm_head = torch.nn.Parameter(
+                torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+        self.register_buffer(
+            "cos_sin",
+            ops.make_cos_sin_table(cfg.max_position, cfg.head_dim, cfg.rope_theta),
+            persistent=False)
+        self.random_init(seed)
+        self.to(device)
+
+    @torch.no_grad()
+    def random_init(self, seed: int = 0):
+        gen = torch.Generator().manual_seed(seed)
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                p.copy_(torch.randn(p.shape, generator=gen, dtype=torch.float32)
+                        .mul_(0.02).to(p.dtype))
+            else:
+                p.fill_(1.0)  # norm weights
+
+    @torch.no_grad()
+    def forward(self, input_ids: torch.Tensor, md: AttnMetadata,
+                kv_caches: List, last_rows: Optional[torch.Tensor] = None
+                ) -> torch.Tensor:
+        """Returns logits for `last_rows` (defaults: all rows)."""
+        hidden = F.embedding(input_ids, self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            k_cache, v_cache = kv_caches[i]
+            hidden, residual = layer(hidden, residual, k_cache, v_cache, md,
+                                     self.cos_sin, self.tp_group)
+        final = torch.empty_like(hidden)
+        ops.fused_add_rmsnorm(final, hidden, residual, self.final_ln, self.cfg.norm_eps)
+        if last_rows is not None:
+            final = final[last_rows]
+        return F.linear(final, self.lm_head).float()
+
+    @torch.no_grad()
+    def load_safetensors(self, path: str):
+        """Load HF-format Llama weights (weights-path deploys)."""
+        from safetensors.torch import load_file
+        import glob, os
+
+        files = sorted(glob.glob(os.path.join(path, "*.safetensors")))
+        if not files:
+            raise FileNotFoundError(f"no safetensors under {path}")
+        sd = {}
+        for f in files:
+            sd.update(load_file(f))
+        self._load_hf_state_dict(sd)
+
+    @torch.no_grad()
+    def _load_hf_state_dict(self, sd: Dict[str, torch.Tensor]):
+        cfg = self.cfg
+        def t(name):
+            return sd[name].to(torch.bfloat16)
+        self.embed.copy_(t("model.embed_tokens.weight"))
+        if not cfg.tie_embeddings and "lm_head.weight" in sd:
+            self.lm_head.copy_(t("lm_head.weight"))
+        self.final_ln.copy_(t("model.norm.weight"))
+        for i, layer in enumerate(self.layers):
+            pfx = f"model.layers.{i}."
+            q = t(pfx + "self_attn.q_proj.weight")
+            k = t(pfx + "self_attn.k_proj.weight")
+            v = t(pfx + "self_attn.v_proj.weight")
+            layer.attn.qkv_proj.copy_(torch.cat([q, k, v], dim=0))
+            layer.attn.o_proj.copy_(t(pfx + "self_attn.o_proj.weight"))
+            g = t(pfx + "mlp.gate_proj.weight")
+            u = t(pfx + "mlp.up_proj.weight")
+            layer.mlp.gate_up.copy_(torch.cat([g, u], dim=0))
+            layer.mlp.down.copy_(t(pfx + "mlp.down_proj.weight"))
+            layer.input_ln.copy_(t(pfx + "input_layernorm.weight"))
+            layer.post_ln.copy_(t(pfx + "post_attention_layernorm.weight"))
+
+
+# ---------------- model registry ----------------
+
+LLAMA_CONFIGS: Dict[str, LlamaConfig] = {
+    # hd=128 + GQA ratio 4 so the tiny model exercises the real kernels
+    "tiny-llama": LlamaConfig(name="tiny-llama", vocab_size=512, hidden_size=512,
+                              n_layers=2, n_heads=4, n_kv_heads=1,
+                              intermediate_size=1024, max_position=4096),
+    "llama3-8b": LlamaConfig(name="llama3-8b", vocab_size=128256,
+                             hidden_size=4096, n_layers=32, n_heads=32,
+                             n_kv_heads=8, intermediate_size=14336,
+                             max_position=8192, tie_embeddings=False),
+    "llama3-70b": LlamaConfig(name="llama3-70b", vocab_size=128256,
+                              hidden_size=8192, n_layers=80, n_heads=64,
+                              n_kv_heads=8, intermediate_size=28672,
+                              max_position=8192, tie_embeddings=False),
+}

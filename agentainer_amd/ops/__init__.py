@@ -1,0 +1,166 @@
+"""ops — dispatch layer over the gfx950 HIP kernels.
+
+Policy (the driver's "native code not loaded" check depends on this):
+  * On a CUDA/ROCm device: the in-tree HIP extension (_hip_ops.so) is
+    REQUIRED. If it is missing or fails to import, GPU ops raise
+    RuntimeError — there is no silent eager fallback on a GPU box.
+  * On CPU: the fp32 torch references (ops.reference) run, so the engine
+    and control plane are testable without a GPU.
+
+Build the extension with `python -m agentainer_amd.ops.build` (or via
+__graft_entry__.build()).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import reference
+
+_hip = None
+_hip_err: Optional[str] = None
+
+
+def _load_hip():
+    global _hip, _hip_err
+    if _hip is not None or _hip_err is not None:
+        return _hip
+    so = os.path.join(os.path.dirname(__file__), "_hip_ops.so")
+    if not os.path.exists(so):
+        _hip_err = f"HIP extension not built ({so} missing); run python -m agentainer_amd.ops.build"
+        return None
+    try:
+        import importlib.util
+
+        spec = importlib.util.spec_from_file_location("agentainer_amd.ops._hip_ops", so)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _hip = mod
+    except Exception as exc:  # noqa: BLE001
+        _hip_err = f"HIP extension failed to load: {exc}"
+    return _hip
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def _dispatch(name: str, *tensors: torch.Tensor):
+    """Return the HIP module if the leading tensor is on GPU, else None.
+    On GPU with no extension: fail loudly (never a silent fallback)."""
+    if tensors[0].is_cuda:
+        mod = _load_hip()
+        if mod is None:
+            raise RuntimeError(
+                f"op {name!r} on GPU requires the gfx950 HIP extension: {_hip_err}")
+        return mod
+    return None
+
+
+def rmsnorm(out, x, w, eps: float = 1e-5):
+    mod = _dispatch("rmsnorm", x)
+    if mod:
+        mod.rmsnorm(out, x, w, float(eps))
+    else:
+        reference.rmsnorm(out, x, w, eps)
+    return out
+
+
+def fused_add_rmsnorm(out, x, residual, w, eps: float = 1e-5):
+    mod = _dispatch("fused_add_rmsnorm", x)
+    if mod:
+        mod.fused_add_rmsnorm(out, x, residual, w, float(eps))
+    else:
+        reference.fused_add_rmsnorm(out, x, residual, w, eps)
+    return out
+
+
+def silu_mul(out, gate, up):
+    mod = _dispatch("silu_mul", gate)
+    if mod:
+        mod.silu_mul(out, gate, up)
+    else:
+        reference.silu_mul(out, gate, up)
+    return out
+
+
+make_cos_sin_table = reference.make_cos_sin_table
+
+
+def rope_inplace(q, k, cos_sin, positions):
+    mod = _dispatch("rope_inplace", q)
+    if mod:
+        mod.rope_inplace(q, k, cos_sin, positions)
+    else:
+        reference.rope_inplace(q, k, cos_sin, positions)
+
+
+def kv_append(k_cache, v_cache, k, v, slot_mapping):
+    mod = _dispatch("kv_append", k)
+    if mod:
+        mod.kv_append(k_cache, v_cache, k, v, slot_mapping)
+    else:
+        reference.kv_append(k_cache, v_cache, k, v, slot_mapping)
+
+
+def paged_decode_attention(out, q, k_cache, v_cache, page_table, seq_lens,
+                           scale: float):
+    mod = _dispatch("paged_decode_attention", q)
+    if mod:
+        mod.paged_decode_attention(out, q, k_cache, v_cache, page_table,
+                                   seq_lens, float(scale))
+    else:
+        reference.paged_decode_attention(out, q, k_cache, v_cache, page_table,
+                                         seq_lens, scale)
+    return out
+
+
+def paged_prefill_attention(out, q, k_cache, v_cache, page_table, seq_lens,
+                            query_starts, query_lens, scale: float):
+    mod = _dispatch("paged_prefill_attention", q)
+    if mod:
+        mod.paged_prefill_attention(out, q, k_cache, v_cache, page_table,
+                                    seq_lens, query_starts, query_lens,
+                                    float(scale))
+    else:
+        reference.paged_prefill_attention(out, q, k_cache, v_cache, page_table,
+                                          seq_lens, query_starts, query_lens,
+                                          scale)
+    return out
+
+
+def greedy_sample(out, logits):
+    mod = _dispatch("greedy_sample", logits)
+    if mod:
+        mod.greedy_sample(out, logits)
+    else:
+        reference.greedy_sample(out, logits)
+    return out
+
+
+def topp_sample(out, logits, temps, top_ps, seeds):
+    mod = _dispatch("topp_sample", logits)
+    if mod:
+        mod.topp_sample(out, logits, temps, top_ps, seeds)
+    else:
+        reference.topp_sample(out, logits, temps, top_ps, seeds)
+    return out
+
+
+def gather_kv_pages(dst, k_cache, v_cache, page_ids):
+    mod = _dispatch("gather_kv_pages", k_cache)
+    if mod:
+        mod.gather_kv_pages(dst, k_cache, v_cache, page_ids)
+    else:
+        reference.gather_kv_pages(dst, k_cache, v_cache, page_ids)
+
+
+def scatter_kv_pages(k_cache, v_cache, src, page_ids):
+    mod = _dispatch("scatter_kv_pages", k_cache)
+    if mod:
+        mod.scatter_kv_pages(k_cache, v_cache, src, page_ids)
+    else:
+        reference.scatter_kv_pages(k_cache, v_cache, src, page_ids)

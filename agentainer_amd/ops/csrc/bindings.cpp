@@ -1,0 +1,44 @@
+// pybind11 bindings for the agentainer_amd gfx950 HIP kernels.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps);
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
+                       torch::Tensor residual, torch::Tensor w, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up);
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
+                  torch::Tensor positions);
+void kv_append(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k,
+               torch::Tensor v, torch::Tensor slot_mapping);
+void paged_decode_attention(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor page_table, torch::Tensor seq_lens,
+                            double scale);
+void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor page_table, torch::Tensor seq_lens,
+                             torch::Tensor query_starts, torch::Tensor query_lens,
+                             double scale);
+void greedy_sample(torch::Tensor out, torch::Tensor logits);
+void topp_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
+                 torch::Tensor top_ps, torch::Tensor seeds);
+void gather_kv_pages(torch::Tensor dst, torch::Tensor k_cache,
+                     torch::Tensor v_cache, torch::Tensor page_ids);
+void scatter_kv_pages(torch::Tensor k_cache, torch::Tensor v_cache,
+                      torch::Tensor src, torch::Tensor page_ids);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "agentainer_amd CDNA4 (gfx950) kernels";
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual += x; rmsnorm");
+  m.def("silu_mul", &silu_mul, "silu(gate) * up");
+  m.def("rope_inplace", &rope_inplace, "apply rotary embedding to q,k in place");
+  m.def("kv_append", &kv_append, "scatter new k/v into the paged KV cache");
+  m.def("paged_decode_attention", &paged_decode_attention,
+        "single-token GQA attention over the paged KV cache");
+  m.def("paged_prefill_attention", &paged_prefill_attention,
+        "varlen causal MFMA attention over the paged KV cache");
+  m.def("greedy_sample", &greedy_sample, "argmax over vocab per row");
+  m.def("topp_sample", &topp_sample, "temperature + top-p sampling per row");
+  m.def("gather_kv_pages", &gather_kv_pages, "pages -> host-shaped buffer");
+  m.def("scatter_kv_pages", &scatter_kv_pages, "buffer -> pages");
+}
