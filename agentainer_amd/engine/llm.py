@@ -105,6 +105,7 @@ class ModelInstance:
         self.decode_tokens = 0
         self.prefill_tokens = 0
         self.occupancy_acc = 0.0
+        self.sync_mode = bool(engine_cfg.get("sync_mode", False))
 
     @staticmethod
     def _pool_pages(cfg: LlamaConfig, page_size: int, device: str,
@@ -123,6 +124,8 @@ class ModelInstance:
     # ---------- scheduler thread ----------
 
     def start(self):
+        if self.sync_mode:
+            return  # bench/test harnesses drive step() directly
         if self._thread is None:
             self._stop.clear()
             self._thread = threading.Thread(target=self._loop,
@@ -137,6 +140,8 @@ class ModelInstance:
             self._thread = None
 
     def alive(self) -> bool:
+        if self.sync_mode:
+            return True
         return self._thread is not None and self._thread.is_alive()
 
     def _loop(self):

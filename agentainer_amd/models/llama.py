@@ -160,30 +160,34 @@ class LlamaForCausalLM(torch.nn.Module):
         self.tp_rank = tp_rank
         self.tp_size = tp_size
         self.tp_group = None
-        self.embed = torch.nn.Parameter(
-            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
-        self.layers = torch.nn.ModuleList(
-            [LlamaLayer(cfg, tp_rank, tp_size) for _ in range(cfg.n_layers)])
-        self.final_ln = torch.nn.Parameter(torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
-        if cfg.tie_embeddings:
-            self.lm_head = self.embed
-        else:
-            self.lm_head = torch.nn.Parameter(
+        # construct parameters directly on the target device (an 8B/70B
+        # shard must never round-trip through host RAM)
+        with torch.device(device):
+            self.embed = torch.nn.Parameter(
                 torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+            self.layers = torch.nn.ModuleList(
+                [LlamaLayer(cfg, tp_rank, tp_size) for _ in range(cfg.n_layers)])
+            self.final_ln = torch.nn.Parameter(
+                torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+            if cfg.tie_embeddings:
+                self.lm_head = self.embed
+            else:
+                self.lm_head = torch.nn.Parameter(
+                    torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
         self.register_buffer(
             "cos_sin",
-            ops.make_cos_sin_table(cfg.max_position, cfg.head_dim, cfg.rope_theta),
+            ops.make_cos_sin_table(cfg.max_position, cfg.head_dim, cfg.rope_theta,
+                                   device=device),
             persistent=False)
         self.random_init(seed)
-        self.to(device)
 
     @torch.no_grad()
     def random_init(self, seed: int = 0):
-        gen = torch.Generator().manual_seed(seed)
+        dev = self.embed.device
+        gen = torch.Generator(device=dev).manual_seed(seed)
         for name, p in self.named_parameters():
             if p.dim() >= 2:
-                p.copy_(torch.randn(p.shape, generator=gen, dtype=torch.float32)
-                        .mul_(0.02).to(p.dtype))
+                p.data.normal_(0.0, 0.02, generator=gen)
             else:
                 p.fill_(1.0)  # norm weights
 

@@ -109,11 +109,14 @@ class Store:
             self._compact_locked()
 
     def flush(self) -> None:
-        """fsync the AOF — request-WAL callers use this as their commit point."""
+        """WAL commit point. Always pushes buffered ops to the OS (survives
+        a SIGKILL of this process); fsyncs to media only under sync=always
+        (power-loss durability — same split as Redis AOF everysec/always)."""
         with self._lock:
             if self._aof is not None:
                 self._aof.flush()
-                os.fsync(self._aof.fileno())
+                if self._sync == "always":
+                    os.fsync(self._aof.fileno())
 
     def _compact_locked(self) -> None:
         if self._path is None:

@@ -1,0 +1,257 @@
+"""GPU kernel numerics: every gfx950 HIP kernel vs the plain-PyTorch fp32
+reference of the same op (SURVEY.md §4 implication (3)). All @pytest.mark.gpu."""
+
+import math
+
+import pytest
+import torch
+
+from agentainer_amd import ops
+from agentainer_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+
+@pytest.fixture(autouse=True, scope="module")
+def _require_hip():
+    assert ops.hip_available(), "HIP extension must be present on the GPU box"
+
+
+DEV = "cuda"
+
+
+def test_rmsnorm_matches_reference():
+    torch.manual_seed(0)
+    for T, H in [(1, 4096), (33, 4096), (256, 8192), (7, 512)]:
+        x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+        out = torch.empty_like(x)
+        ops.rmsnorm(out, x, w, 1e-5)
+        want = torch.empty_like(x)
+        ref.rmsnorm(want, x.cpu(), w.cpu(), 1e-5)
+        assert torch.allclose(out.float().cpu(), want.float(), atol=0.02, rtol=0.02), \
+            f"rmsnorm mismatch T={T} H={H}"
+
+
+def test_fused_add_rmsnorm_matches_reference():
+    torch.manual_seed(1)
+    T, H = 64, 4096
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    x_c, res_c = x.cpu().clone(), res.cpu().clone()
+    out = torch.empty_like(x)
+    ops.fused_add_rmsnorm(out, x, res, w, 1e-5)
+    want = torch.empty_like(x_c)
+    ref.fused_add_rmsnorm(want, x_c, res_c, w.cpu(), 1e-5)
+    assert torch.allclose(res.float().cpu(), res_c.float(), atol=0.02, rtol=0.02)
+    assert torch.allclose(out.float().cpu(), want.float(), atol=0.02, rtol=0.02)
+
+
+def test_silu_mul_matches_reference():
+    g = torch.randn(1024, 128, dtype=torch.bfloat16, device=DEV)
+    u = torch.randn(1024, 128, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty_like(g)
+    ops.silu_mul(out, g, u)
+    want = torch.empty_like(g.cpu())
+    ref.silu_mul(want, g.cpu(), u.cpu())
+    assert torch.allclose(out.float().cpu(), want.float(), atol=0.02, rtol=0.05)
+
+
+def test_rope_matches_reference():
+    torch.manual_seed(2)
+    T, nq, nkv, D = 17, 8, 2, 128
+    q = torch.randn(T, nq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, nkv, D, dtype=torch.bfloat16, device=DEV)
+    qc, kc = q.cpu().clone(), k.cpu().clone()
+    tab = ops.make_cos_sin_table(256, D, device=DEV)
+    pos = torch.randint(0, 256, (T,), dtype=torch.int32, device=DEV)
+    ops.rope_inplace(q, k, tab, pos)
+    ref.rope_inplace(qc, kc, tab.cpu(), pos.cpu())
+    assert torch.allclose(q.float().cpu(), qc.float(), atol=0.03, rtol=0.03)
+    assert torch.allclose(k.float().cpu(), kc.float(), atol=0.03, rtol=0.03)
+
+
+def _gpu_caches(P, n_kv, D, PS):
+    k = torch.zeros(P, n_kv, D // 8, PS, 8, dtype=torch.bfloat16, device=DEV)
+    v = torch.zeros(P, n_kv, PS, D, dtype=torch.bfloat16, device=DEV)
+    return k, v
+
+
+def test_kv_append_matches_reference():
+    torch.manual_seed(3)
+    P, n_kv, D, PS, T = 8, 8, 128, 16, 20
+    kc, vc = _gpu_caches(P, n_kv, D, PS)
+    k = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    slots = torch.arange(T, dtype=torch.long, device=DEV) * 3 % (P * PS)
+    ops.kv_append(kc, vc, k, v, slots)
+    kr = torch.zeros_like(kc, device="cpu")
+    vr = torch.zeros_like(vc, device="cpu")
+    ref.kv_append(kr, vr, k.cpu(), v.cpu(), slots.cpu())
+    assert torch.equal(kc.cpu(), kr)
+    assert torch.equal(vc.cpu(), vr)
+
+
+def _fill_seq(kc, vc, page_table_row, length, n_kv, D, PS, seed):
+    gen = torch.Generator(device="cpu").manual_seed(seed)
+    K = torch.randn(length, n_kv, D, generator=gen, dtype=torch.float32)
+    V = torch.randn(length, n_kv, D, generator=gen, dtype=torch.float32)
+    Kb, Vb = K.to(torch.bfloat16).to(DEV), V.to(torch.bfloat16).to(DEV)
+    slots = []
+    for t in range(length):
+        page = int(page_table_row[t // PS])
+        slots.append(page * PS + t % PS)
+    ops.kv_append(kc, vc, Kb, Vb, torch.tensor(slots, dtype=torch.long, device=DEV))
+    return Kb.cpu(), Vb.cpu()
+
+
+@pytest.mark.parametrize("lens", [[5], [64], [1, 77, 200, 33], [129] * 8])
+def test_decode_attention_matches_reference(lens):
+    torch.manual_seed(4)
+    n_q, n_kv, D, PS = 32, 8, 128, 16
+    B = len(lens)
+    max_pages = max(-(-l // PS) for l in lens)
+    P = B * max_pages + 1
+    kc, vc = _gpu_caches(P, n_kv, D, PS)
+    page_table = torch.zeros(B, max_pages, dtype=torch.int32)
+    next_page = 1
+    for b in range(B):
+        for i in range(-(-lens[b] // PS)):
+            page_table[b, i] = next_page
+            next_page += 1
+    KVs = [_fill_seq(kc, vc, page_table[b], lens[b], n_kv, D, PS, seed=100 + b)
+           for b in range(B)]
+    q = torch.randn(B, n_q, D, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty_like(q)
+    scale = 1.0 / math.sqrt(D)
+    pt_dev = page_table.to(DEV)
+    sl_dev = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    ops.paged_decode_attention(out, q, kc, vc, pt_dev, sl_dev, scale)
+    want = torch.empty(B, n_q, D, dtype=torch.bfloat16)
+    ref.paged_decode_attention(want, q.cpu(), kc.cpu(), vc.cpu(), page_table,
+                               torch.tensor(lens, dtype=torch.int32), scale)
+    diff = (out.float().cpu() - want.float()).abs().max().item()
+    assert diff < 0.05, f"decode attn max abs diff {diff} (lens={lens})"
+
+
+@pytest.mark.parametrize("ctx,new", [(0, 32), (0, 77), (48, 64), (130, 31), (0, 1)])
+def test_prefill_attention_matches_reference(ctx, new):
+    torch.manual_seed(5)
+    n_q, n_kv, D, PS = 32, 8, 128, 16
+    total = ctx + new
+    max_pages = -(-total // PS)
+    kc, vc = _gpu_caches(max_pages + 2, n_kv, D, PS)
+    page_table = torch.arange(1, max_pages + 1, dtype=torch.int32).unsqueeze(0)
+    _fill_seq(kc, vc, page_table[0], total, n_kv, D, PS, seed=7)
+    q = torch.randn(new, n_q, D, dtype=torch.bfloat16, device=DEV)
+    out = torch.zeros_like(q)
+    scale = 1.0 / math.sqrt(D)
+    args = (page_table.to(DEV), torch.tensor([total], dtype=torch.int32, device=DEV),
+            torch.tensor([0], dtype=torch.int32, device=DEV),
+            torch.tensor([new], dtype=torch.int32, device=DEV))
+    ops.paged_prefill_attention(out, q, kc, vc, *args, scale)
+    want = torch.zeros(new, n_q, D, dtype=torch.bfloat16)
+    ref.paged_prefill_attention(want, q.cpu(), kc.cpu(), vc.cpu(), page_table,
+                                torch.tensor([total], dtype=torch.int32),
+                                torch.tensor([0], dtype=torch.int32),
+                                torch.tensor([new], dtype=torch.int32), scale)
+    diff = (out.float().cpu() - want.float()).abs().max().item()
+    assert diff < 0.05, f"prefill attn max abs diff {diff} (ctx={ctx} new={new})"
+
+
+def test_prefill_attention_varlen_batch():
+    torch.manual_seed(6)
+    n_q, n_kv, D, PS = 32, 8, 128, 16
+    specs = [(0, 40), (16, 16), (64, 100)]  # (ctx, new)
+    max_pages = max(-(-(c + n) // PS) for c, n in specs)
+    kc, vc = _gpu_caches(len(specs) * max_pages + 1, n_kv, D, PS)
+    page_table = torch.zeros(len(specs), max_pages, dtype=torch.int32)
+    nxt = 1
+    for b, (c, n) in enumerate(specs):
+        for i in range(-(-(c + n) // PS)):
+            page_table[b, i] = nxt
+            nxt += 1
+        _fill_seq(kc, vc, page_table[b], c + n, n_kv, D, PS, seed=50 + b)
+    total_new = sum(n for _, n in specs)
+    starts = []
+    acc = 0
+    for _, n in specs:
+        starts.append(acc)
+        acc += n
+    q = torch.randn(total_new, n_q, D, dtype=torch.bfloat16, device=DEV)
+    out = torch.zeros_like(q)
+    scale = 1.0 / math.sqrt(D)
+    seq_lens = torch.tensor([c + n for c, n in specs], dtype=torch.int32)
+    q_starts = torch.tensor(starts, dtype=torch.int32)
+    q_lens = torch.tensor([n for _, n in specs], dtype=torch.int32)
+    ops.paged_prefill_attention(out, q, kc, vc, page_table.to(DEV),
+                                seq_lens.to(DEV), q_starts.to(DEV),
+                                q_lens.to(DEV), scale)
+    want = torch.zeros_like(q.cpu())
+    ref.paged_prefill_attention(want, q.cpu(), kc.cpu(), vc.cpu(), page_table,
+                                seq_lens, q_starts, q_lens, scale)
+    diff = (out.float().cpu() - want.float()).abs().max().item()
+    assert diff < 0.05, f"varlen prefill max abs diff {diff}"
+
+
+def test_greedy_sample_matches_argmax():
+    torch.manual_seed(7)
+    logits = torch.randn(64, 128256, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty(64, dtype=torch.long, device=DEV)
+    ops.greedy_sample(out, logits)
+    want = logits.float().argmax(-1)
+    assert torch.equal(out, want)
+
+
+def test_topp_sample_nucleus_property():
+    torch.manual_seed(8)
+    B, V = 32, 128256
+    logits = torch.randn(B, V, dtype=torch.bfloat16, device=DEV)
+    peak = torch.randint(0, V, (B,))
+    for b in range(B):
+        logits[b, peak[b]] += 12.0  # overwhelming mass on one token
+    out = torch.empty(B, dtype=torch.long, device=DEV)
+    temps = torch.full((B,), 0.8, device=DEV)
+    tps = torch.full((B,), 0.3, device=DEV)
+    seeds = torch.arange(B, dtype=torch.int64, device=DEV)
+    ops.topp_sample(out, logits, temps, tps, seeds)
+    assert torch.equal(out.cpu(), peak.long())
+
+
+def test_topp_sample_distribution():
+    """With top_p=1, temp=1, draws should roughly follow softmax over a
+    3-token-dominated distribution."""
+    B, V = 512, 1024
+    logits = torch.full((B, V), -10.0, dtype=torch.bfloat16, device=DEV)
+    logits[:, 0] = 1.0
+    logits[:, 1] = 1.0
+    logits[:, 2] = 1.0
+    out = torch.empty(B, dtype=torch.long, device=DEV)
+    temps = torch.ones(B, device=DEV)
+    tps = torch.ones(B, device=DEV)
+    seeds = torch.arange(B, dtype=torch.int64, device=DEV) * 7919
+    ops.topp_sample(out, logits, temps, tps, seeds)
+    picks = out.cpu()
+    assert (picks < 3).float().mean() > 0.98
+    counts = torch.bincount(picks.clamp(max=3), minlength=4)[:3].float()
+    assert counts.min() > B / 3 * 0.5  # each of the 3 tokens drawn often
+
+
+def test_gather_scatter_roundtrip_gpu():
+    P, n_kv, D, PS = 8, 8, 128, 16
+    kc, vc = _gpu_caches(P, n_kv, D, PS)
+    kc.normal_()
+    vc.normal_()
+    plane = n_kv * D * PS
+    ids = torch.tensor([2, 5, 7], dtype=torch.int32, device=DEV)
+    buf = torch.zeros(3 * 2 * plane, dtype=torch.bfloat16, device=DEV)
+    ops.gather_kv_pages(buf, kc, vc, ids)
+    kc2, vc2 = _gpu_caches(P, n_kv, D, PS)
+    ops.scatter_kv_pages(kc2, vc2, buf, ids)
+    for p in [2, 5, 7]:
+        assert torch.equal(kc2[p], kc[p])
+        assert torch.equal(vc2[p], vc[p])
