@@ -219,3 +219,31 @@ void topp_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                      temps.data_ptr<float>(), top_ps.data_ptr<float>(),
                      (const unsigned long long*)seeds.data_ptr(), V);
 }
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe (test-only): one v_mfma_f32_16x16x32_bf16 on prepacked
+// per-lane fragments. Host packs candidate layouts and checks which
+// reconstruction matches a torch matmul — pins down the gfx950 fragment
+// maps empirically (used by tools/mfma_probe.py and the kernel tests).
+namespace {
+typedef __bf16 probe_bf16v8 __attribute__((ext_vector_type(8)));
+__global__ void mfma_probe_kernel(const short* __restrict__ a,
+                                  const short* __restrict__ b,
+                                  float* __restrict__ c) {
+  const int lane = threadIdx.x;
+  probe_bf16v8 av = *reinterpret_cast<const probe_bf16v8*>(a + lane * 8);
+  probe_bf16v8 bv = *reinterpret_cast<const probe_bf16v8*>(b + lane * 8);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) c[lane * 4 + r] = acc[r];
+}
+}  // namespace
+
+void mfma_probe(torch::Tensor c, torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.numel() == 512 && b.numel() == 512 && c.numel() == 256);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const short*)a.data_ptr(), (const short*)b.data_ptr(),
+                     c.data_ptr<float>());
+}
