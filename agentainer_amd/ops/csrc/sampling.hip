@@ -6,13 +6,15 @@
 // vectorized loads.
 //
 // topp_sample: per row — temperature scale, streaming max + exp-sum
-// (softmax denominator), then candidate harvest: every token with
-// p >= P_FLOOR * p_max goes to an LDS buffer that is bitonic-sorted by
-// probability; the top-p nucleus is cut there and one draw is taken by
-// inverse CDF. With top_p <= 1 - P_FLOOR*CAND_CAP this is exact; if the
-// candidate buffer overflows (pathologically flat distributions) the
-// kernel falls back to sampling the full softmax without the nucleus cut
-// and reports it via the overflow flag semantics documented in ops/__init__.
+// (softmax denominator), then ADAPTIVE candidate harvest: token counts are
+// taken at four probability floors (p >= {1e-1,1e-2,1e-3,1e-4} * p_max) in
+// one pass and the loosest floor whose candidate set fits the LDS buffer
+// is harvested, bitonic-sorted by probability, nucleus-cut at top_p and
+// drawn by inverse CDF. Exact whenever the nucleus lies above the chosen
+// floor (always, for peaked LLM logits); for pathologically flat rows the
+// draw truncates to the top-CAND_CAP tokens (a p<=1e-1*p_max tail), and if
+// even the tightest floor overflows it falls back to a full categorical
+// draw without the nucleus cut.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -90,6 +92,8 @@ __global__ __launch_bounds__(256) void topp_kernel(
   __shared__ float cand_p[CAND_CAP];
   __shared__ int cand_i[CAND_CAP];
   __shared__ int n_cand;
+  __shared__ int tier_counts[4];
+  __shared__ float chosen_floor_s;
   __shared__ float row_max_s, row_sum_s;
 
   // pass 1: max of scaled logits
@@ -105,31 +109,58 @@ __global__ __launch_bounds__(256) void topp_kernel(
   if ((tid & 63) == 0) red[tid / 64] = mx;
   __syncthreads();
   mx = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
-  if (tid == 0) { row_max_s = mx; n_cand = 0; }
+  if (tid == 0) {
+    row_max_s = mx;
+    n_cand = 0;
+    for (int t = 0; t < 4; ++t) tier_counts[t] = 0;
+  }
   __syncthreads();
 
-  // pass 2: exp-sum + candidate harvest
+  // pass 2: exp-sum + candidate counts at the four floors
+  // floors: p >= 10^-(t+1) * p_max  =>  s - mx >= -(t+1)*ln(10)
+  const float LN10 = 2.302585093f;
   float sum = 0.f;
-  const float floor_logit = logf(P_FLOOR);  // harvest if (s - mx) >= ln floor
+  int my_counts[4] = {0, 0, 0, 0};
   for (int i = tid; i < V; i += blockDim.x) {
     float s = bits2f(lp[i]) * invT - mx;
-    float e = __expf(s);
-    sum += e;
-    if (s >= floor_logit) {
-      int slot = atomicAdd(&n_cand, 1);
-      if (slot < CAND_CAP) { cand_p[slot] = e; cand_i[slot] = i; }
-    }
+    sum += __expf(s);
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+      if (s >= -(float)(t + 1) * LN10) ++my_counts[t];
   }
   sum = wave_sum(sum);
   if ((tid & 63) == 0) red[tid / 64] = sum;
+#pragma unroll
+  for (int t = 0; t < 4; ++t)
+    if (my_counts[t]) atomicAdd(&tier_counts[t], my_counts[t]);
   __syncthreads();
   sum = red[0] + red[1] + red[2] + red[3];
-  if (tid == 0) row_sum_s = sum;
+  if (tid == 0) {
+    row_sum_s = sum;
+    // loosest floor (largest t) whose candidate set fits
+    float floor_logit = 1.0f;  // sentinel: none fits
+    for (int t = 3; t >= 0; --t)
+      if (tier_counts[t] <= CAND_CAP) { floor_logit = -(float)(t + 1) * LN10; break; }
+    chosen_floor_s = floor_logit;
+  }
   __syncthreads();
   const float total = row_sum_s;
   const float u = uniform01(seeds[row], row);
+  const float floor_logit = chosen_floor_s;
 
-  if (n_cand <= CAND_CAP) {
+  if (floor_logit <= 0.f) {
+    // pass 3: harvest at the chosen floor
+    for (int i = tid; i < V; i += blockDim.x) {
+      float s = bits2f(lp[i]) * invT - mx;
+      if (s >= floor_logit) {
+        int slot = atomicAdd(&n_cand, 1);
+        if (slot < CAND_CAP) { cand_p[slot] = __expf(s); cand_i[slot] = i; }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (floor_logit <= 0.f && n_cand <= CAND_CAP) {
     // bitonic sort candidates by p descending (padded to pow2 with -1)
     int n = n_cand;
     int npow = 1;

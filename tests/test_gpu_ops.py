@@ -30,7 +30,7 @@ def test_rmsnorm_matches_reference():
         w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
         out = torch.empty_like(x)
         ops.rmsnorm(out, x, w, 1e-5)
-        want = torch.empty_like(x)
+        want = torch.empty_like(x, device="cpu")
         ref.rmsnorm(want, x.cpu(), w.cpu(), 1e-5)
         assert torch.allclose(out.float().cpu(), want.float(), atol=0.02, rtol=0.02), \
             f"rmsnorm mismatch T={T} H={H}"
