@@ -50,7 +50,8 @@ class KVCheckpoint:
 
 class KVCacheManager:
     def __init__(self, n_layers: int, n_kv: int, head_dim: int, page_size: int,
-                 n_pages: int, device="cpu", dtype=torch.bfloat16):
+                 n_pages: int, device="cpu", dtype=torch.bfloat16,
+                 max_slots: int = 1024, max_pages_per_seq: int = 512):
         assert head_dim % 8 == 0
         self.n_layers = n_layers
         self.n_kv = n_kv
@@ -66,11 +67,29 @@ class KVCacheManager:
                 n_pages, n_kv, head_dim // 8, page_size, 8, dtype=dtype, device=device))
             self.v_caches.append(torch.zeros(
                 n_pages, n_kv, page_size, head_dim, dtype=dtype, device=device))
-        self._free: List[int] = list(range(n_pages - 1, -1, -1))
+        # page 0 is reserved as the scratch page: hipGraph-padded decode rows
+        # write their garbage K/V there (see llm.ModelInstance._device_decode)
+        self._free: List[int] = list(range(n_pages - 1, 0, -1))
         self._seqs: Dict[str, Sequence] = {}
         self._lock = threading.RLock()
         # shorts per page per layer: K plane + V plane
         self.page_shorts = 2 * n_kv * head_dim * page_size
+        # ---- device mirrors for the in-graph decode step (GPU only) ----
+        # dev_page_table[slot] / dev_seq_lens[slot] are the ground truth the
+        # captured decode graph reads; host bookkeeping mirrors them.
+        self.is_gpu = str(device) not in ("cpu",)
+        self.max_slots = max_slots
+        self.max_pages_per_seq = max_pages_per_seq
+        if self.is_gpu:
+            self.dev_page_table = torch.zeros(max_slots, max_pages_per_seq,
+                                              dtype=torch.int32, device=device)
+            self.dev_seq_lens = torch.full((max_slots,), -1, dtype=torch.int32,
+                                           device=device)
+        else:
+            self.dev_page_table = None
+            self.dev_seq_lens = None
+        self._free_slots: List[int] = list(range(max_slots - 1, -1, -1))
+        self._slot_of: Dict[str, int] = {}
 
     # ---------- queries ----------
 
@@ -110,13 +129,60 @@ class KVCacheManager:
                 raise ValueError(f"sequence {seq_id} exists")
             s = Sequence(seq_id)
             self._seqs[seq_id] = s
+            if self.is_gpu:
+                if not self._free_slots:
+                    raise OutOfPages("no free sequence slots")
+                slot = self._free_slots.pop()
+                self._slot_of[seq_id] = slot
+                self.dev_seq_lens[slot] = 0
             return s
+
+    def slot(self, seq_id: str) -> int:
+        return self._slot_of[seq_id]
+
+    def push_dev(self, seq_id: str) -> None:
+        """Sync a sequence's page row + length to the device mirrors."""
+        if not self.is_gpu:
+            return
+        with self._lock:
+            s = self._seqs[seq_id]
+            slot = self._slot_of[seq_id]
+            if s.pages:
+                self.dev_page_table[slot, :len(s.pages)] = torch.tensor(
+                    s.pages, dtype=torch.int32, device=self.device)
+            self.dev_seq_lens[slot] = s.length
+
+    def ensure_decode_page(self, seq_id: str) -> None:
+        """Host half of the in-graph decode append: guarantee the page for
+        the NEXT token exists (the graph computes the slot on device)."""
+        with self._lock:
+            s = self._seqs[seq_id]
+            page_idx = s.length // self.page_size
+            if page_idx >= self.max_pages_per_seq:
+                raise OutOfPages("sequence exceeded max_pages_per_seq")
+            if page_idx >= len(s.pages):
+                if not self._free:
+                    raise OutOfPages(f"KV pool exhausted ({self.n_pages} pages)")
+                page = self._free.pop()
+                s.pages.append(page)
+                if self.is_gpu:
+                    self.dev_page_table[self._slot_of[seq_id], page_idx] = page
+
+    def advance_host(self, seq_id: str) -> None:
+        """Host length += 1 after an in-graph decode step incremented the
+        device length."""
+        with self._lock:
+            self._seqs[seq_id].length += 1
 
     def free_seq(self, seq_id: str) -> None:
         with self._lock:
             s = self._seqs.pop(seq_id, None)
             if s:
                 self._free.extend(reversed(s.pages))
+            slot = self._slot_of.pop(seq_id, None)
+            if slot is not None:
+                self.dev_seq_lens[slot] = -1
+                self._free_slots.append(slot)
 
     def reset_seq(self, seq_id: str) -> None:
         """Drop a sequence's KV but keep it registered (context truncation)."""
@@ -125,6 +191,8 @@ class KVCacheManager:
             self._free.extend(reversed(s.pages))
             s.pages = []
             s.length = 0
+            if self.is_gpu:
+                self.dev_seq_lens[self._slot_of[seq_id]] = 0
 
     def can_append(self, seq_id: str, n_tokens: int) -> bool:
         with self._lock:
@@ -221,5 +289,6 @@ class KVCacheManager:
             staging = ckpt.data[li].to(self.device, non_blocking=is_gpu)
             ops.scatter_kv_pages(self.k_caches[li], self.v_caches[li],
                                  staging, page_ids)
+        self.push_dev(seq_id)
         if is_gpu:
             torch.cuda.synchronize()

@@ -106,6 +106,21 @@ class ModelInstance:
         self.prefill_tokens = 0
         self.occupancy_acc = 0.0
         self.sync_mode = bool(engine_cfg.get("sync_mode", False))
+        # hipGraph-captured decode step (SURVEY.md north star): the whole
+        # batched decode forward — page-table gather, slot computation,
+        # RoPE/append/attention/GEMMs, device-side length increment — is
+        # captured once per batch bucket and replayed with two small H2D
+        # copies (row slots + last tokens) per step.
+        self.is_gpu = device.startswith("cuda")
+        self.use_graph = self.is_gpu and bool(engine_cfg.get("graph_capture", True))
+        self._graphs: Dict[int, Dict[str, Any]] = {}
+        self._pad_slot = -1
+        if self.is_gpu:
+            # reserve one sequence slot as the graph's pad row target
+            self._pad_slot = self.kvm._free_slots.pop()
+            self.kvm.dev_seq_lens[self._pad_slot] = -1
+            self._pad_slot_t = torch.tensor([self._pad_slot], dtype=torch.long,
+                                            device=device)
 
     @staticmethod
     def _pool_pages(cfg: LlamaConfig, page_size: int, device: str,
@@ -255,6 +270,9 @@ class ModelInstance:
             ids.extend(r.prompt_tokens)
             positions.extend(range(prev, prev + n))
             slots.extend(self.kvm.append_slots(b.seq_id, n))
+        if self.is_gpu:
+            for sid in seq_ids:
+                self.kvm.push_dev(sid)
         md = AttnMetadata(
             page_table=self.kvm.page_table(seq_ids, device=dev),
             seq_lens=self.kvm.seq_lens(seq_ids, device=dev),
@@ -277,7 +295,97 @@ class ModelInstance:
                 r.first_token_t = now
                 self._finish_or_run(r, int(t))
 
+    # ---------- device-side decode (hipGraph) ----------
+
+    def _device_decode_fwd(self, rows: torch.Tensor, input_ids: torch.Tensor,
+                           inc: torch.Tensor) -> torch.Tensor:
+        """The captured body: everything reads device state only."""
+        kvm = self.kvm
+        PS = kvm.page_size
+        lens = kvm.dev_seq_lens.index_select(0, rows)      # int32 [Bk]
+        pos = lens.clamp(min=0)
+        page_idx = torch.div(pos, PS, rounding_mode="floor").long()
+        pt = kvm.dev_page_table.index_select(0, rows)      # [Bk, MP] int32
+        pages = pt.gather(1, page_idx.unsqueeze(1)).squeeze(1).long()
+        slots = pages * PS + (pos % PS).long()             # pad rows -> page 0
+        md = AttnMetadata(page_table=pt, seq_lens=lens + 1,
+                          slot_mapping=slots, positions=pos, is_prefill=False)
+        logits = self.model(input_ids, md, self.kvm.kv_caches(), None)
+        kvm.dev_seq_lens.index_add_(0, rows, inc)
+        kvm.dev_seq_lens.index_fill_(0, self._pad_slot_t, -1)
+        return logits
+
+    @staticmethod
+    def _bucket(n: int) -> int:
+        b = 1
+        while b < n:
+            b <<= 1
+        return b
+
+    def _get_graph(self, bucket: int) -> Dict[str, Any]:
+        entry = self._graphs.get(bucket)
+        if entry is not None:
+            return entry
+        dev = self.device
+        rows = torch.full((bucket,), self._pad_slot, dtype=torch.long, device=dev)
+        ids = torch.zeros(bucket, dtype=torch.long, device=dev)
+        inc = torch.ones(bucket, dtype=torch.int32, device=dev)
+        rows_pin = torch.full((bucket,), self._pad_slot, dtype=torch.long,
+                              pin_memory=True)
+        ids_pin = torch.zeros(bucket, dtype=torch.long, pin_memory=True)
+        entry = {"rows": rows, "ids": ids, "inc": inc, "rows_pin": rows_pin,
+                 "ids_pin": ids_pin, "graph": None, "logits": None}
+        if self.use_graph:
+            for _ in range(2):  # warmup before capture
+                self._device_decode_fwd(rows, ids, inc)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                entry["logits"] = self._device_decode_fwd(rows, ids, inc)
+            entry["graph"] = g
+        self._graphs[bucket] = entry
+        return entry
+
+    def _decode_gpu(self, reqs: List[GenRequest]):
+        kvm = self.kvm
+        B = len(reqs)
+        bucket = min(self._bucket(B), max(self.max_decode_batch, 1))
+        if bucket < B:
+            reqs = reqs[:bucket]
+            B = bucket
+        row_ids = []
+        for r in reqs:
+            b = self._bindings[r.agent_id]
+            kvm.ensure_decode_page(b.seq_id)   # host half of the append
+            row_ids.append(kvm.slot(b.seq_id))
+        entry = self._get_graph(bucket)
+        entry["rows_pin"][:B] = torch.tensor(row_ids, dtype=torch.long)
+        entry["rows_pin"][B:] = self._pad_slot
+        entry["ids_pin"][:B] = torch.tensor([r.generated[-1] for r in reqs],
+                                            dtype=torch.long)
+        entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
+        entry["ids"].copy_(entry["ids_pin"], non_blocking=True)
+        if entry["graph"] is not None:
+            entry["graph"].replay()
+            logits = entry["logits"][:B]
+        else:
+            logits = self._device_decode_fwd(entry["rows"], entry["ids"],
+                                             entry["inc"])[:B]
+        toks = self._sample(logits, reqs)
+        self.decode_tokens += B
+        self.occupancy_acc += B / max(1, self.max_decode_batch)
+        with self._lock:
+            for r in reqs:
+                b = self._bindings.get(r.agent_id)
+                if b is not None:
+                    kvm.advance_host(b.seq_id)
+            for r, t in zip(reqs, toks):
+                r.generated.append(int(t))
+                self._finish_or_run(r, int(t))
+
     def _decode(self, reqs: List[GenRequest]):
+        if self.is_gpu:
+            return self._decode_gpu(reqs)
         dev = self.device
         ids, positions, slots, seq_ids = [], [], [], []
         for r in reqs:

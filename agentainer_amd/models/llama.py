@@ -206,7 +206,7 @@ class LlamaForCausalLM(torch.nn.Module):
         ops.fused_add_rmsnorm(final, hidden, residual, self.final_ln, self.cfg.norm_eps)
         if last_rows is not None:
             final = final[last_rows]
-        return F.linear(final, self.lm_head).float()
+        return F.linear(final, self.lm_head)  # bf16; samplers take bf16
 
     @torch.no_grad()
     def load_safetensors(self, path: str):
