@@ -89,12 +89,15 @@ class LlamaAttention(torch.nn.Module):
         qkv = F.linear(h, self.qkv_proj)
         q_sz = self.n_heads * self.head_dim
         kv_sz = self.n_kv * self.head_dim
-        q = qkv[:, :q_sz].view(T, self.n_heads, self.head_dim).contiguous()
-        k = qkv[:, q_sz:q_sz + kv_sz].view(T, self.n_kv, self.head_dim).contiguous()
-        v = qkv[:, q_sz + kv_sz:].view(T, self.n_kv, self.head_dim).contiguous()
+        # strided views into the fused GEMM output — the HIP kernels take
+        # token strides, so no contiguous copies are materialized
+        q = qkv[:, :q_sz].view(T, self.n_heads, self.head_dim)
+        k = qkv[:, q_sz:q_sz + kv_sz].view(T, self.n_kv, self.head_dim)
+        v = qkv[:, q_sz + kv_sz:].view(T, self.n_kv, self.head_dim)
         ops.rope_inplace(q, k, cos_sin, md.positions)
         ops.kv_append(k_cache, v_cache, k, v, md.slot_mapping)
-        out = torch.empty_like(q)
+        out = torch.empty(T, self.n_heads, self.head_dim, dtype=q.dtype,
+                          device=q.device)
         if md.is_prefill:
             ops.paged_prefill_attention(out, q, k_cache, v_cache, md.page_table,
                                         md.seq_lens, md.query_starts,
@@ -120,8 +123,9 @@ class LlamaMLP(torch.nn.Module):
 
     def forward(self, h, tp_group=None):
         gu = F.linear(h, self.gate_up)
-        gate, up = gu[:, :self.inter].contiguous(), gu[:, self.inter:].contiguous()
-        act = torch.empty_like(gate)
+        gate, up = gu[:, :self.inter], gu[:, self.inter:]
+        act = torch.empty(gu.size(0), self.inter, dtype=gu.dtype,
+                          device=gu.device)
         ops.silu_mul(act, gate, up)
         out = F.linear(act, self.down)
         if tp_group is not None:

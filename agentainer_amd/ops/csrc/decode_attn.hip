@@ -1,23 +1,29 @@
-// Paged GQA decode attention (gfx950) — one new token per sequence.
+// Paged GQA decode attention (gfx950) — split-KV two-kernel design.
 //
 // Serves the concurrent-agent decode step (SURVEY.md §2.3 "Decode attention
-// kernel" row; BASELINE.json configs 2-5). The op is HBM-bound: each
-// (sequence, kv-head) streams seq_len * 2 * D bf16 of K/V once. Design:
+// kernel" row; BASELINE.json configs 2-5). HBM-bound: the whole op streams
+// seq_len * 2 * D bf16 of K/V per (seq, kv-head) ONCE — the GQA group's
+// `ratio` query heads share every K/V byte loaded.
 //
-//   grid:  (n_seqs, n_kv_heads); block: 4 waves (256 threads).
-//   Each wave owns one query head of the GQA group (ratio > 4: waves loop).
-//   KV is consumed in 64-token chunks with two lane roles:
-//     score phase: lane = token. K layout [page][h][D/8][ps][8] makes the
-//       16-lane page subgroups read 16 B contiguous per (d8, token) — fully
-//       coalesced. q (f32) lives in LDS and broadcasts (all lanes read the
-//       same address = LDS broadcast, conflict-free).
-//     PV phase:   lane = dim pair. V layout [page][h][ps][D] makes lanes
-//       read 4 B contiguous across a token row. p broadcasts from LDS.
-//   Online softmax (m, l running; rescale the 2-f32 o accumulator).
+// Kernel 1 (partials): grid (n_seqs, n_kv, SPLITS), ONE wave each.
+//   A wave owns a chunk-aligned KV span and ALL `ratio` q heads of its
+//   kv head (RATIO is a template constant so the per-head accumulators
+//   stay in registers — runtime indexing would spill to scratch).
+//     score phase: lane = token. K layout [page][h][D/8][ps][8] gives
+//       16-lane-contiguous 16 B loads; each loaded k-vector feeds RATIO
+//       dot products against q rows broadcast from LDS.
+//     PV phase: lane = dim pair; V rows read once (4 B/lane contiguous),
+//       each value feeds RATIO accumulators via p broadcast from LDS.
+//   Online softmax per span; unnormalized (m, l, o) partials to workspace.
+// Kernel 2 (combine): grid (n_seqs, n_q), one wave: merge the SPLITS
+//   partials with the standard log-sum-exp rescale, normalize, write bf16.
 //
-// Wave-shape notes: all reductions are 64-wide __shfl_xor (wave64, never
-// warp-32 idioms); per-wave LDS slices avoid __syncthreads in the KV loop
-// (s_waitcnt lgkmcnt orders each wave's own LDS accesses).
+// Parallelism: B*n_kv*SPLITS waves (e.g. 64 seqs x 8 kv x 8 = 4096 waves
+// on 256 CUs) — the v1 single-kernel design peaked at 2 workgroups/CU and
+// 0.36 TB/s; this shape exists to keep every CU's memory queue full.
+//
+// Workspace layout: part[b][g][split][h][PART_STRIDE] f32 where
+// [0..127] = o pairs (lane d), [128] = m, [129] = l.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -27,114 +33,251 @@
 namespace {
 
 constexpr int CHUNK = 64;
+constexpr int SPLITS = 8;
+constexpr int PART_STRIDE = 132;  // 128 o + m + l (+2 pad)
 
-template <int D>
-__global__ __launch_bounds__(256) void decode_attn_kernel(
-    short* __restrict__ out,            // [B, n_q, D]
+template <int RATIO, int D>
+__global__ __launch_bounds__(64) void decode_partial_kernel(
+    float* __restrict__ part,           // [B, n_kv, SPLITS, RATIO, PART_STRIDE]
     const short* __restrict__ q,        // [B, n_q, D]
     const short* __restrict__ k_cache,  // [P, n_kv, D/8, PS, 8]
     const short* __restrict__ v_cache,  // [P, n_kv, PS, D]
     const int* __restrict__ page_table, // [B, max_pages]
     const int* __restrict__ seq_lens,   // [B]
-    float scale, int n_q, int n_kv, int PS, int max_pages, int ratio) {
+    float scale, int n_q, int n_kv, int PS, int max_pages, long q_ts) {
   const int b = blockIdx.x;
-  const int g = blockIdx.y;  // kv head
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
+  const int g = blockIdx.y;
+  const int split = blockIdx.z;
+  const int lane = threadIdx.x;
   const int len = seq_lens[b];
-  if (len <= 0) return;
   constexpr int D8 = D / 8;
 
-  // LDS: per-wave q (f32[D]) and p (f32[CHUNK]) slices.
-  __shared__ float q_lds[4][D];
-  __shared__ float p_lds[4][CHUNK];
+  float* my_part = part +
+      ((((long)b * gridDim.y + g) * SPLITS + split) * RATIO) * PART_STRIDE;
 
-  const int* pt = page_table + (long)b * max_pages;
   const int n_chunks = (len + CHUNK - 1) / CHUNK;
+  const int cpw = (n_chunks + SPLITS - 1) / SPLITS;  // chunks per wave
+  const int c0 = split * cpw;
+  const int c1 = min(c0 + cpw, n_chunks);
+  if (len <= 0 || c0 >= c1) {
+    // empty span: publish neutral partials
+    for (int h = 0; h < RATIO; ++h) {
+      float* pp = my_part + h * PART_STRIDE;
+      pp[2 * lane] = 0.f;
+      pp[2 * lane + 1] = 0.f;
+      if (lane == 0) { pp[128] = -FLT_MAX; pp[129] = 0.f; }
+    }
+    return;
+  }
 
-  for (int qh = g * ratio + wid; qh < (g + 1) * ratio; qh += 4) {
-    // load q for this head into LDS as f32
-    const short* qp = q + ((long)b * n_q + qh) * D;
+  __shared__ float q_lds[RATIO][D];
+  __shared__ float p_lds[RATIO][CHUNK];
+  const int qh0 = g * RATIO;
+#pragma unroll
+  for (int h = 0; h < RATIO; ++h) {
+    const short* qp = q + (long)b * q_ts + (long)(qh0 + h) * D;
 #pragma unroll
     for (int r = 0; r < D / WAVE; ++r)
-      q_lds[wid][r * WAVE + lane] = bits2f(qp[r * WAVE + lane]);
-    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): q_lds visible to this wave
+      q_lds[h][r * WAVE + lane] = bits2f(qp[r * WAVE + lane]);
+  }
+  __builtin_amdgcn_s_waitcnt(0);
 
-    float m = -FLT_MAX, l = 0.f;
-    float o0 = 0.f, o1 = 0.f;  // this lane's two output dims
-    const int d0 = lane * 2;
+  float m[RATIO], l[RATIO], o0[RATIO], o1[RATIO];
+#pragma unroll
+  for (int h = 0; h < RATIO; ++h) {
+    m[h] = -FLT_MAX; l[h] = 0.f; o0[h] = 0.f; o1[h] = 0.f;
+  }
+  const int* pt = page_table + (long)b * max_pages;
+  const int d0 = lane * 2;
 
-    for (int c = 0; c < n_chunks; ++c) {
-      const int tok = c * CHUNK + lane;  // this lane's token (score phase)
-      float s = -FLT_MAX;
-      if (tok < len) {
-        const long page = pt[tok / PS];
-        const int off = tok % PS;
-        const short* kp = k_cache + (((long)page * n_kv + g) * D8 * PS + off) * 8;
-        float acc = 0.f;
+  for (int c = c0; c < c1; ++c) {
+    const int tok = c * CHUNK + lane;
+    float s[RATIO];
 #pragma unroll
-        for (int d8 = 0; d8 < D8; ++d8) {
-          bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(kp + (long)d8 * PS * 8);
+    for (int h = 0; h < RATIO; ++h) s[h] = -FLT_MAX;
+    if (tok < len) {
+      const long page = pt[tok / PS];
+      const short* kp = k_cache + (((long)page * n_kv + g) * D8 * PS + tok % PS) * 8;
+      float acc[RATIO];
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            acc = fmaf(bits2f(kv8[j]), q_lds[wid][d8 * 8 + j], acc);
+      for (int h = 0; h < RATIO; ++h) acc[h] = 0.f;
+#pragma unroll
+      for (int d8 = 0; d8 < D8; ++d8) {
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(kp + (long)d8 * PS * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float kf = bits2f(kv8[j]);
+#pragma unroll
+          for (int h = 0; h < RATIO; ++h)
+            acc[h] = fmaf(kf, q_lds[h][d8 * 8 + j], acc[h]);
         }
-        s = acc * scale;
       }
-      // online softmax update across the wave
-      const float cmax = wave_max(s);
-      const float mn = fmaxf(m, cmax);
-      const float p = (tok < len) ? __expf(s - mn) : 0.f;
+#pragma unroll
+      for (int h = 0; h < RATIO; ++h) s[h] = acc[h] * scale;
+    }
+    // per-head online softmax update
+    const int c_len = min(CHUNK, len - c * CHUNK);
+#pragma unroll
+    for (int h = 0; h < RATIO; ++h) {
+      const float cmax = wave_max(s[h]);
+      const float mn = fmaxf(m[h], cmax);
+      const float p = (tok < len) ? __expf(s[h] - mn) : 0.f;
       const float csum = wave_sum(p);
-      const float alpha = (m == -FLT_MAX) ? 0.f : __expf(m - mn);
-      l = l * alpha + csum;
-      o0 *= alpha;
-      o1 *= alpha;
-      m = mn;
-      p_lds[wid][lane] = p;
-      __builtin_amdgcn_s_waitcnt(0);
+      const float alpha = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
+      l[h] = l[h] * alpha + csum;
+      o0[h] *= alpha;
+      o1[h] *= alpha;
+      m[h] = mn;
+      p_lds[h][lane] = p;
+    }
+    __builtin_amdgcn_s_waitcnt(0);
 
-      // PV phase: lane = dim pair d0, d0+1
-      const int c_len = min(CHUNK, len - c * CHUNK);
-      for (int t = 0; t < c_len; ++t) {
-        const int gt = c * CHUNK + t;
+    // PV: lane = dim pair; V row loaded once, feeds all RATIO heads.
+    // 8-token blocks: preload 8 independent rows, then FMA.
+    const int base_tok = c * CHUNK;
+    int t = 0;
+    for (; t + 8 <= c_len; t += 8) {
+      float v0[8], v1[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int gt = base_tok + t + u;
         const long page = pt[gt / PS];
-        const int off = gt % PS;
-        const short* vp = v_cache + (((long)page * n_kv + g) * PS + off) * D + d0;
-        const float pw = p_lds[wid][t];
-        o0 = fmaf(pw, bits2f(vp[0]), o0);
-        o1 = fmaf(pw, bits2f(vp[1]), o1);
+        const short* vp = v_cache + (((long)page * n_kv + g) * PS + gt % PS) * D + d0;
+        v0[u] = bits2f(vp[0]);
+        v1[u] = bits2f(vp[1]);
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+#pragma unroll
+        for (int h = 0; h < RATIO; ++h) {
+          const float pw = p_lds[h][t + u];
+          o0[h] = fmaf(pw, v0[u], o0[h]);
+          o1[h] = fmaf(pw, v1[u], o1[h]);
+        }
       }
     }
-    const float inv = (l > 0.f) ? 1.0f / l : 0.f;
-    short* op = out + ((long)b * n_q + qh) * D + d0;
-    op[0] = f2bits(o0 * inv);
-    op[1] = f2bits(o1 * inv);
+    for (; t < c_len; ++t) {
+      const int gt = base_tok + t;
+      const long page = pt[gt / PS];
+      const short* vp = v_cache + (((long)page * n_kv + g) * PS + gt % PS) * D + d0;
+      const float v0 = bits2f(vp[0]), v1 = bits2f(vp[1]);
+#pragma unroll
+      for (int h = 0; h < RATIO; ++h) {
+        const float pw = p_lds[h][t];
+        o0[h] = fmaf(pw, v0, o0[h]);
+        o1[h] = fmaf(pw, v1, o1[h]);
+      }
+    }
   }
+
+  // publish partials (unnormalized)
+#pragma unroll
+  for (int h = 0; h < RATIO; ++h) {
+    float* pp = my_part + h * PART_STRIDE;
+    pp[2 * lane] = o0[h];
+    pp[2 * lane + 1] = o1[h];
+    if (lane == 0) { pp[128] = m[h]; pp[129] = l[h]; }
+  }
+}
+
+template <int D>
+__global__ __launch_bounds__(64) void decode_combine_kernel(
+    short* __restrict__ out,          // [B, n_q, D]
+    const float* __restrict__ part,   // [B, n_kv, SPLITS, RATIO, PART_STRIDE]
+    const int* __restrict__ seq_lens, int n_q, int n_kv) {
+  const int b = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int lane = threadIdx.x;
+  if (seq_lens[b] <= 0) return;
+  const int ratio = n_q / n_kv;
+  const int g = qh / ratio;
+  const int h = qh % ratio;
+  const float* pp = part +
+      ((((long)b * n_kv + g) * SPLITS) * ratio + h) * PART_STRIDE;
+  const long step = (long)ratio * PART_STRIDE;
+
+  float M = -FLT_MAX;
+#pragma unroll
+  for (int s = 0; s < SPLITS; ++s) M = fmaxf(M, pp[s * step + 128]);
+  float L = 0.f, O0 = 0.f, O1 = 0.f;
+#pragma unroll
+  for (int s = 0; s < SPLITS; ++s) {
+    const float ms = pp[s * step + 128];
+    if (ms == -FLT_MAX) continue;
+    const float w = __expf(ms - M);
+    L += w * pp[s * step + 129];
+    O0 = fmaf(w, pp[s * step + 2 * lane], O0);
+    O1 = fmaf(w, pp[s * step + 2 * lane + 1], O1);
+  }
+  const float inv = (L > 0.f) ? 1.0f / L : 0.f;
+  short* op = out + ((long)b * n_q + qh) * D + 2 * lane;
+  op[0] = f2bits(O0 * inv);
+  op[1] = f2bits(O1 * inv);
 }
 
 }  // namespace
 
-void paged_decode_attention(torch::Tensor out, torch::Tensor q,
-                            torch::Tensor k_cache, torch::Tensor v_cache,
-                            torch::Tensor page_table, torch::Tensor seq_lens,
-                            double scale) {
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+void paged_decode_attention_ws(torch::Tensor out, torch::Tensor q,
+                               torch::Tensor k_cache, torch::Tensor v_cache,
+                               torch::Tensor page_table, torch::Tensor seq_lens,
+                               torch::Tensor workspace, double scale) {
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16);
   TORCH_CHECK(page_table.scalar_type() == at::kInt &&
               seq_lens.scalar_type() == at::kInt);
+  TORCH_CHECK(workspace.scalar_type() == at::kFloat);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   const int B = q.size(0), n_q = q.size(1), D = q.size(2);
   const int n_kv = k_cache.size(1), PS = k_cache.size(3);
   const int max_pages = page_table.size(1);
   const int ratio = n_q / n_kv;
   TORCH_CHECK(D == 128, "decode attention: head_dim 128 only");
   TORCH_CHECK(n_q % n_kv == 0);
+  TORCH_CHECK(workspace.numel() >=
+              (long)B * n_kv * SPLITS * ratio * PART_STRIDE,
+              "decode workspace too small");
   if (B == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL((decode_attn_kernel<128>), dim3(B, n_kv), dim3(256), 0,
-                     stream, (short*)out.data_ptr(), (const short*)q.data_ptr(),
-                     (const short*)k_cache.data_ptr(),
-                     (const short*)v_cache.data_ptr(),
-                     page_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                     (float)scale, n_q, n_kv, PS, max_pages, ratio);
+  dim3 grid1(B, n_kv, SPLITS);
+#define LAUNCH_RATIO(R)                                                        \
+  hipLaunchKernelGGL((decode_partial_kernel<R, 128>), grid1, dim3(64), 0,      \
+                     stream, workspace.data_ptr<float>(),                      \
+                     (const short*)q.data_ptr(),                               \
+                     (const short*)k_cache.data_ptr(),                         \
+                     (const short*)v_cache.data_ptr(),                         \
+                     page_table.data_ptr<int>(), seq_lens.data_ptr<int>(),     \
+                     (float)scale, n_q, n_kv, PS, max_pages,                    \
+                     (long)q.stride(0))
+  switch (ratio) {
+    case 1: LAUNCH_RATIO(1); break;
+    case 2: LAUNCH_RATIO(2); break;
+    case 4: LAUNCH_RATIO(4); break;
+    case 8: LAUNCH_RATIO(8); break;
+    default:
+      TORCH_CHECK(false, "decode attention: GQA ratio must be 1/2/4/8, got ",
+                  ratio);
+  }
+#undef LAUNCH_RATIO
+  hipLaunchKernelGGL((decode_combine_kernel<128>), dim3(B, n_q), dim3(64), 0,
+                     stream, (short*)out.data_ptr(),
+                     workspace.data_ptr<float>(), seq_lens.data_ptr<int>(),
+                     n_q, n_kv);
+}
+
+// Workspace floats needed for (B, n_q/n_kv ratio, n_kv).
+long decode_attention_workspace_size(long B, long n_q, long n_kv) {
+  return B * n_kv * SPLITS * (n_q / n_kv) * PART_STRIDE;
+}
+
+void paged_decode_attention(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor page_table, torch::Tensor seq_lens,
+                            double scale) {
+  const int B = q.size(0), n_q = q.size(1);
+  const int n_kv = k_cache.size(1);
+  auto ws = torch::empty(
+      {decode_attention_workspace_size(B, n_q, n_kv)},
+      torch::TensorOptions().dtype(torch::kFloat).device(q.device()));
+  paged_decode_attention_ws(out, q, k_cache, v_cache, page_table, seq_lens,
+                            ws, scale);
 }

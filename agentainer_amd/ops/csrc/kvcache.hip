@@ -27,7 +27,7 @@ __global__ void kv_append_kernel(short* __restrict__ k_cache,
                                  const short* __restrict__ k,
                                  const short* __restrict__ v,
                                  const long* __restrict__ slots, int n_kv,
-                                 int D, int PS) {
+                                 int D, int PS, long kts, long vts) {
   const int t = blockIdx.x;
   const long slot = slots[t];
   if (slot < 0) return;
@@ -37,11 +37,11 @@ __global__ void kv_append_kernel(short* __restrict__ k_cache,
   const int chunks = n_kv * D8;
   for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
     const int h = c / D8, d8 = c % D8;
-    bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + ((long)t * n_kv + h) * D + d8 * 8);
+    bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + (long)t * kts + (long)h * D + d8 * 8);
     // k_cache[page][h][d8][off][0..8]
     long kidx = ((((page * n_kv + h) * D8 + d8) * PS) + off) * 8;
     *reinterpret_cast<bf16x8*>(k_cache + kidx) = kv8;
-    bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + ((long)t * n_kv + h) * D + d8 * 8);
+    bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + (long)t * vts + (long)h * D + d8 * 8);
     // v_cache[page][h][off][d8*8..]
     long vidx = (((page * n_kv + h) * PS) + off) * D + d8 * 8;
     *reinterpret_cast<bf16x8*>(v_cache + vidx) = vv8;
@@ -78,7 +78,6 @@ __global__ void page_copy_kernel(short* __restrict__ k_cache,
 
 void kv_append(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k,
                torch::Tensor v, torch::Tensor slot_mapping) {
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
   TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16);
   TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
   const int T = k.size(0), n_kv = k.size(1), D = k.size(2);
@@ -86,12 +85,15 @@ void kv_append(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k,
   TORCH_CHECK(k_cache.size(1) == n_kv && k_cache.size(2) == D / 8);
   TORCH_CHECK(v_cache.size(2) == PS && v_cache.size(3) == D);
   TORCH_CHECK(D % 8 == 0);
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == D);
+  TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == D);
   if (T == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(kv_append_kernel, dim3(T), dim3(256), 0, stream,
                      (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
                      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     slot_mapping.data_ptr<long>(), n_kv, D, PS);
+                     slot_mapping.data_ptr<long>(), n_kv, D, PS,
+                     (long)k.stride(0), (long)v.stride(0));
 }
 
 static void page_copy(torch::Tensor k_cache, torch::Tensor v_cache,

@@ -52,7 +52,7 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
     const int* __restrict__ seq_lens,   // [B] total context length
     const int* __restrict__ q_starts,   // [B] row offset into q
     const int* __restrict__ q_lens,     // [B]
-    float scale, int n_q, int n_kv, int PS, int max_pages) {
+    float scale, int n_q, int n_kv, int PS, int max_pages, long q_ts) {
   const int qtile = blockIdx.x;
   const int b = blockIdx.y;
   const int head = blockIdx.z * 4 + threadIdx.x / WAVE;
@@ -82,7 +82,7 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
 #pragma unroll
     for (int dc = 0; dc < 4; ++dc) {
       if (qr < qlen) {
-        const short* p = q + ((q_row0 + qr) * n_q + head) * D + dc * 32 + lg * 8;
+        const short* p = q + (q_row0 + qr) * q_ts + (long)head * D + dc * 32 + lg * 8;
         qf[qs][dc] = *reinterpret_cast<const bf16v8*>(p);
       } else {
         qf[qs][dc] = bf16v8{};
@@ -259,8 +259,9 @@ void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
                              torch::Tensor page_table, torch::Tensor seq_lens,
                              torch::Tensor query_starts, torch::Tensor query_lens,
                              double scale) {
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   const int n_q = q.size(1), Dh = q.size(2);
   const int n_kv = k_cache.size(1), PS = k_cache.size(3);
   const int B = seq_lens.size(0);
@@ -285,5 +286,6 @@ void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
                      (const short*)v_cache.data_ptr(),
                      page_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                      query_starts.data_ptr<int>(), query_lens.data_ptr<int>(),
-                     (float)scale, n_q, n_kv, PS, max_pages);
+                     (float)scale, n_q, n_kv, PS, max_pages,
+                     (long)q.stride(0));
 }

@@ -17,17 +17,19 @@ namespace {
 // grid: (T, n_q + n_kv); block: D/2 lanes (<=256). Each lane rotates one
 // (x1, x2) pair; each pair of bf16 loads is 2 B, cos/sin are two f32x1
 // loads from a row the whole block shares (L1-resident).
+// q/k may be strided views into the fused qkv GEMM output (token stride
+// qs/ks) — avoids materializing contiguous q/k copies per layer.
 __global__ void rope_kernel(short* __restrict__ q, short* __restrict__ k,
                             const float* __restrict__ cos_sin,
                             const int* __restrict__ positions, int n_q,
-                            int n_kv, int D) {
+                            int n_kv, int D, long qs, long ks) {
   const int t = blockIdx.x;
   const int h = blockIdx.y;
   const int i = threadIdx.x;  // pair index < D/2
   const int half = D / 2;
   if (i >= half) return;
-  short* base = (h < n_q) ? q + ((long)t * n_q + h) * D
-                          : k + ((long)t * n_kv + (h - n_q)) * D;
+  short* base = (h < n_q) ? q + (long)t * qs + (long)h * D
+                          : k + (long)t * ks + (long)(h - n_q) * D;
   const int pos = positions[t];
   const float c = cos_sin[(long)pos * D + i];
   const float s = cos_sin[(long)pos * D + half + i];
@@ -41,7 +43,7 @@ __global__ void rope_kernel(short* __restrict__ q, short* __restrict__ k,
 
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
                   torch::Tensor positions) {
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && cos_sin.is_contiguous());
+  TORCH_CHECK(cos_sin.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16 && k.scalar_type() == at::kBFloat16);
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
   TORCH_CHECK(positions.scalar_type() == at::kInt);
@@ -49,9 +51,12 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
   const int n_kv = k.size(1);
   TORCH_CHECK(k.size(0) == T && k.size(2) == D);
   TORCH_CHECK(D % 2 == 0 && D / 2 <= 1024);
+  // strided views allowed: last dim contiguous, head stride == D
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D);
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == D);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(rope_kernel, dim3(T, n_q + n_kv), dim3(D / 2), 0, stream,
                      (short*)q.data_ptr(), (short*)k.data_ptr(),
                      cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
-                     n_q, n_kv, D);
+                     n_q, n_kv, D, (long)q.stride(0), (long)k.stride(0));
 }

@@ -71,7 +71,7 @@ def kv_append(k_cache: torch.Tensor, v_cache: torch.Tensor, k: torch.Tensor,
             continue
         page, off = slot // PS, slot % PS
         # k_cache [P, n_kv, D/8, PS, 8]
-        k_cache[page, :, :, off, :] = k[t].view(n_kv, D // 8, 8)
+        k_cache[page, :, :, off, :] = k[t].reshape(n_kv, D // 8, 8)
         v_cache[page, :, off, :] = v[t]
 
 
