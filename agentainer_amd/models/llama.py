@@ -86,7 +86,7 @@ class LlamaAttention(torch.nn.Module):
 
     def forward(self, h, k_cache, v_cache, md: AttnMetadata, cos_sin, tp_group=None):
         T = h.size(0)
-        qkv = F.linear(h, self.qkv_proj)
+        qkv = ops.linear(h, self.qkv_proj)
         q_sz = self.n_heads * self.head_dim
         kv_sz = self.n_kv * self.head_dim
         # strided views into the fused GEMM output — the HIP kernels take
@@ -105,7 +105,7 @@ class LlamaAttention(torch.nn.Module):
         else:
             ops.paged_decode_attention(out, q, k_cache, v_cache, md.page_table,
                                        md.seq_lens, self.scale)
-        o = F.linear(out.view(T, q_sz), self.o_proj)
+        o = ops.linear(out.view(T, q_sz), self.o_proj)
         if tp_group is not None:
             torch.distributed.all_reduce(o, group=tp_group)
         return o
@@ -122,12 +122,12 @@ class LlamaMLP(torch.nn.Module):
         self.inter = inter
 
     def forward(self, h, tp_group=None):
-        gu = F.linear(h, self.gate_up)
+        gu = ops.linear(h, self.gate_up)
         gate, up = gu[:, :self.inter], gu[:, self.inter:]
         act = torch.empty(gu.size(0), self.inter, dtype=gu.dtype,
                           device=gu.device)
         ops.silu_mul(act, gate, up)
-        out = F.linear(act, self.down)
+        out = ops.linear(act, self.down)
         if tp_group is not None:
             torch.distributed.all_reduce(out, group=tp_group)
         return out
@@ -210,7 +210,7 @@ class LlamaForCausalLM(torch.nn.Module):
         ops.fused_add_rmsnorm(final, hidden, residual, self.final_ln, self.cfg.norm_eps)
         if last_rows is not None:
             final = final[last_rows]
-        return F.linear(final, self.lm_head)  # bf16; samplers take bf16
+        return ops.linear(final, self.lm_head)  # bf16; samplers take bf16
 
     @torch.no_grad()
     def load_safetensors(self, path: str):

@@ -150,6 +150,44 @@ def topp_sample(out, logits, temps, top_ps, seeds):
     return out
 
 
+_ws_cache = {}
+_EMPTY_WS = {}
+
+
+def _skinny_split(ntiles: int, K: int) -> int:
+    split = 1
+    while split < 8 and ntiles * split * 2 < 512 and (K // (split * 2)) % 256 == 0:
+        split *= 2
+    return split
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """x[M,K] @ w[N,K]^T. Decode-shaped (M<=64) GPU GEMMs go to the
+    hand-written skinny MFMA kernel (weight streaming at the HBM
+    roofline); everything else to hipBLASLt via F.linear."""
+    M, K = x.shape
+    N = w.size(0)
+    if x.is_cuda and M <= 64 and N % 64 == 0 and K % 32 == 0:
+        mod = _dispatch("skinny_gemm", x)
+        split = _skinny_split(N // 64, K)
+        out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+        if split > 1:
+            key = (x.device, N, split)
+            ws = _ws_cache.get(key)
+            if ws is None or ws.numel() < split * 64 * N:
+                ws = torch.empty(split * 64 * N, dtype=torch.float32,
+                                 device=x.device)
+                _ws_cache[key] = ws
+        else:
+            ws = _EMPTY_WS.get(x.device)
+            if ws is None:
+                ws = torch.empty(1, dtype=torch.float32, device=x.device)
+                _EMPTY_WS[x.device] = ws
+        mod.skinny_gemm(out, x.contiguous(), w, ws, split)
+        return out
+    return torch.nn.functional.linear(x, w)
+
+
 def gather_kv_pages(dst, k_cache, v_cache, page_ids):
     mod = _dispatch("gather_kv_pages", k_cache)
     if mod:

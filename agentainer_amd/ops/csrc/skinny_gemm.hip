@@ -1,0 +1,129 @@
+// Skinny-M GEMM (gfx950, MFMA bf16) — the decode-step projection GEMM.
+//
+// out[M,N] = x[M,K] @ W[N,K]^T for M <= 64 (torch Linear weight layout).
+// At decode batch sizes hipBLASLt reaches only ~2-4.5 TB/s of weight
+// streaming on these shapes (profiled); this kernel exists to stream W at
+// the HBM roofline:
+//
+//   * W rows feed MFMA B-fragments STRAIGHT from HBM: B lane l holds
+//     W[n0 + l%16][k + (l/16)*8 + j] — a single contiguous 16 B load per
+//     fragment, no LDS staging (cdna_hip_programming.md §5 "GEMV / M<=16
+//     decode weights: load straight to VGPRs" — generalized to MFMA).
+//   * x (<=512 KB) is L2-resident and re-read by every n-tile block —
+//     A-fragments also load straight from global.
+//   * Grid: (N/64) x SPLITK blocks of 4 waves; each wave owns 16 n-cols
+//     and 4 m-sub accumulators (16x16x32 MFMA). Split-K keeps >=256
+//     blocks in flight for small N; fp32 partials combine in a second
+//     elementwise kernel.
+//
+// Used by the engine for decode projections; prefill (large M) stays on
+// hipBLASLt (compute-bound regime where the library is at ~1.2 PF).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+typedef __bf16 bf16v8 __attribute__((ext_vector_type(8)));
+
+template <bool SPLIT>
+__global__ __launch_bounds__(256) void skinny_gemm_kernel(
+    void* __restrict__ out,            // bf16 [M,N] or f32 ws [S,M,N]
+    const short* __restrict__ x,       // [M,K]
+    const short* __restrict__ w,       // [N,K]
+    int M, int N, int K, int k_per_split) {
+  const int n0 = blockIdx.x * 64 + (threadIdx.x / WAVE) * 16;
+  const int split = blockIdx.y;
+  const int lane = threadIdx.x % WAVE;
+  const int l16 = lane % 16;
+  const int lg = lane / 16;
+  const int k0 = split * k_per_split;
+  const int k1 = min(k0 + k_per_split, K);
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) acc[ms] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const short* wp = w + (long)(n0 + l16) * K;
+  // A rows for the 4 m-subs this lane reads
+  const int arow[4] = {l16, 16 + l16, 32 + l16, 48 + l16};
+
+  for (int k = k0; k < k1; k += 32) {
+    const long koff = k + lg * 8;
+    bf16v8 bfrag = *reinterpret_cast<const bf16v8*>(wp + koff);
+#pragma unroll
+    for (int ms = 0; ms < 4; ++ms) {
+      bf16v8 afrag;
+      if (arow[ms] < M) {
+        afrag = *reinterpret_cast<const bf16v8*>(x + (long)arow[ms] * K + koff);
+      } else {
+        afrag = bf16v8{};
+      }
+      acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[ms],
+                                                        0, 0, 0);
+    }
+  }
+
+  // epilogue: C lane holds rows (lg*4 + r) of each m-sub, col l16
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = ms * 16 + lg * 4 + r;
+      if (row >= M) continue;
+      const int col = blockIdx.x * 64 + (threadIdx.x / WAVE) * 16 + l16;
+      if (SPLIT) {
+        float* o = reinterpret_cast<float*>(out);
+        o[((long)split * M + row) * N + col] = acc[ms][r];
+      } else {
+        short* o = reinterpret_cast<short*>(out);
+        o[(long)row * N + col] = f2bits(acc[ms][r]);
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void splitk_combine_kernel(
+    short* __restrict__ out, const float* __restrict__ ws, long mn, int S) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= mn) return;
+  float acc = 0.f;
+  for (int s = 0; s < S; ++s) acc += ws[(long)s * mn + i];
+  out[i] = f2bits(acc);
+}
+
+}  // namespace
+
+void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                 torch::Tensor ws, long split) {
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16);
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K);
+  TORCH_CHECK(M <= 64, "skinny_gemm: M <= 64");
+  TORCH_CHECK(N % 64 == 0, "skinny_gemm: N % 64 == 0");
+  TORCH_CHECK(K % 32 == 0, "skinny_gemm: K % 32 == 0");
+  auto stream = at::hip::getCurrentHIPStream();
+  const int ntiles = N / 64;
+  TORCH_CHECK(split >= 1 && (K % (32 * split)) == 0,
+              "invalid split for K");
+  const int kps = K / (int)split;
+  if (split == 1) {
+    hipLaunchKernelGGL((skinny_gemm_kernel<false>), dim3(ntiles, 1), dim3(256),
+                       0, stream, out.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), M, N, K, kps);
+  } else {
+    TORCH_CHECK(ws.numel() >= (long)split * M * N,
+                "skinny_gemm split-K workspace too small");
+    TORCH_CHECK(ws.scalar_type() == at::kFloat);
+    hipLaunchKernelGGL((skinny_gemm_kernel<true>), dim3(ntiles, split),
+                       dim3(256), 0, stream, ws.data_ptr(),
+                       (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), M, N, K, kps);
+    const long mn = (long)M * N;
+    hipLaunchKernelGGL(splitk_combine_kernel,
+                       dim3((mn + 255) / 256), dim3(256), 0, stream,
+                       (short*)out.data_ptr(), ws.data_ptr<float>(), mn, split);
+  }
+}

@@ -255,3 +255,18 @@ def test_gather_scatter_roundtrip_gpu():
     for p in [2, 5, 7]:
         assert torch.equal(kc2[p], kc[p])
         assert torch.equal(vc2[p], vc[p])
+
+
+@pytest.mark.parametrize("M,N,K", [(64, 4096, 4096), (64, 6144, 4096),
+                                   (64, 4096, 14336), (33, 28672, 4096),
+                                   (1, 4096, 4096), (64, 128256, 4096)])
+def test_skinny_gemm_matches_torch(M, N, K):
+    torch.manual_seed(10)
+    x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.1)
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.1)
+    from agentainer_amd.ops import linear
+    out = linear(x, w)
+    want = torch.nn.functional.linear(x.float(), w.float())
+    diff = (out.float() - want).abs()
+    rel = diff.max().item() / max(want.abs().max().item(), 1e-6)
+    assert rel < 0.02, f"skinny_gemm rel diff {rel} (M={M} N={N} K={K})"
