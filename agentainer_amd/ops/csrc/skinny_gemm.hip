@@ -27,7 +27,7 @@ namespace {
 
 typedef __bf16 bf16v8 __attribute__((ext_vector_type(8)));
 
-template <bool SPLIT>
+template <bool SPLIT, bool FULL>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     void* __restrict__ out,            // bf16 [M,N] or f32 ws [S,M,N]
     const short* __restrict__ x,       // [M,K]
@@ -46,23 +46,38 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   for (int ms = 0; ms < 4; ++ms) acc[ms] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const short* wp = w + (long)(n0 + l16) * K;
-  // A rows for the 4 m-subs this lane reads
-  const int arow[4] = {l16, 16 + l16, 32 + l16, 48 + l16};
+  // A-row base pointers for the 4 m-subs this lane reads. FULL (M == 64)
+  // is branch-free: a per-lane row<M guard here would make hipcc branch
+  // around every load and drain vmcnt per element (guide §5 trap (c)).
+  const short* xp[4];
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) {
+    const int row = ms * 16 + l16;
+    xp[ms] = x + (long)(FULL ? row : min(row, M - 1)) * K;
+  }
+  const bool alive[4] = {FULL || l16 < M, FULL || 16 + l16 < M,
+                         FULL || 32 + l16 < M, FULL || 48 + l16 < M};
 
+#pragma unroll 2
   for (int k = k0; k < k1; k += 32) {
     const long koff = k + lg * 8;
-    bf16v8 bfrag = *reinterpret_cast<const bf16v8*>(wp + koff);
+    // weights stream once per CU: non-temporal (guide: nt-weights)
+    bf16v8 bfrag = __builtin_nontemporal_load(
+        reinterpret_cast<const bf16v8*>(wp + koff));
+    bf16v8 a[4];
 #pragma unroll
-    for (int ms = 0; ms < 4; ++ms) {
-      bf16v8 afrag;
-      if (arow[ms] < M) {
-        afrag = *reinterpret_cast<const bf16v8*>(x + (long)arow[ms] * K + koff);
-      } else {
-        afrag = bf16v8{};
-      }
-      acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[ms],
-                                                        0, 0, 0);
+    for (int ms = 0; ms < 4; ++ms)
+      a[ms] = *reinterpret_cast<const bf16v8*>(xp[ms] + koff);
+    if (!FULL) {
+      // zero out clamped duplicate rows (still branch-free: csel)
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        if (!alive[ms]) a[ms] = bf16v8{};
     }
+#pragma unroll
+    for (int ms = 0; ms < 4; ++ms)
+      acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bfrag, acc[ms],
+                                                        0, 0, 0);
   }
 
   // epilogue: C lane holds rows (lg*4 + r) of each m-sub, col l16
@@ -109,21 +124,25 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(split >= 1 && (K % (32 * split)) == 0,
               "invalid split for K");
   const int kps = K / (int)split;
+  const bool full = (M == 64);
+#define SG_LAUNCH(SPLIT_, FULL_, OUTP)                                         \
+  hipLaunchKernelGGL((skinny_gemm_kernel<SPLIT_, FULL_>),                      \
+                     dim3(ntiles, SPLIT_ ? (int)split : 1), dim3(256), 0,      \
+                     stream, OUTP, (const short*)x.data_ptr(),                 \
+                     (const short*)w.data_ptr(), M, N, K, kps)
   if (split == 1) {
-    hipLaunchKernelGGL((skinny_gemm_kernel<false>), dim3(ntiles, 1), dim3(256),
-                       0, stream, out.data_ptr(), (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), M, N, K, kps);
+    if (full) SG_LAUNCH(false, true, out.data_ptr());
+    else SG_LAUNCH(false, false, out.data_ptr());
   } else {
     TORCH_CHECK(ws.numel() >= (long)split * M * N,
                 "skinny_gemm split-K workspace too small");
     TORCH_CHECK(ws.scalar_type() == at::kFloat);
-    hipLaunchKernelGGL((skinny_gemm_kernel<true>), dim3(ntiles, split),
-                       dim3(256), 0, stream, ws.data_ptr(),
-                       (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), M, N, K, kps);
+    if (full) SG_LAUNCH(true, true, ws.data_ptr());
+    else SG_LAUNCH(true, false, ws.data_ptr());
     const long mn = (long)M * N;
     hipLaunchKernelGGL(splitk_combine_kernel,
                        dim3((mn + 255) / 256), dim3(256), 0, stream,
                        (short*)out.data_ptr(), ws.data_ptr<float>(), mn, split);
   }
+#undef SG_LAUNCH
 }
