@@ -46,38 +46,64 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   for (int ms = 0; ms < 4; ++ms) acc[ms] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const short* wp = w + (long)(n0 + l16) * K;
-  // A-row base pointers for the 4 m-subs this lane reads. FULL (M == 64)
-  // is branch-free: a per-lane row<M guard here would make hipcc branch
-  // around every load and drain vmcnt per element (guide §5 trap (c)).
-  const short* xp[4];
-#pragma unroll
-  for (int ms = 0; ms < 4; ++ms) {
-    const int row = ms * 16 + l16;
-    xp[ms] = x + (long)(FULL ? row : min(row, M - 1)) * K;
-  }
-  const bool alive[4] = {FULL || l16 < M, FULL || 16 + l16 < M,
-                         FULL || 32 + l16 < M, FULL || 48 + l16 < M};
 
-#pragma unroll 2
-  for (int k = k0; k < k1; k += 32) {
-    const long koff = k + lg * 8;
-    // weights stream once per CU: non-temporal (guide: nt-weights)
-    bf16v8 bfrag = __builtin_nontemporal_load(
-        reinterpret_cast<const bf16v8*>(wp + koff));
-    bf16v8 a[4];
-#pragma unroll
-    for (int ms = 0; ms < 4; ++ms)
-      a[ms] = *reinterpret_cast<const bf16v8*>(xp[ms] + koff);
-    if (!FULL) {
-      // zero out clamped duplicate rows (still branch-free: csel)
-#pragma unroll
-      for (int ms = 0; ms < 4; ++ms)
-        if (!alive[ms]) a[ms] = bf16v8{};
+  // x tile [64][KC] staged through LDS once per BLOCK per k-chunk — the
+  // 4 waves reading x straight from global exceeded the per-CU L2
+  // bandwidth (x re-read once per wave instead of once per block).
+  // Double-buffered, T14 split: issue next tile's loads early, ds_write
+  // after the MFMAs (guide §6 G15). Row stride 40 shorts (80 B) makes the
+  // b128 fragment reads conflict-free (row*20 mod 64 distinct over 16).
+  constexpr int KC = 64;
+  constexpr int XS = 40;
+  __shared__ short x_lds[2][64 * XS];
+  const int s_row = threadIdx.x % 64;        // staging: this thread's x row
+  const int s_col8 = threadIdx.x / 64;       // covers cols {0..3}*8, +32 next
+  const short* s_xp = x + (long)(FULL ? s_row : min(s_row, M - 1)) * K;
+  const bool s_alive = FULL || s_row < M;
+
+  bf16x8 st0{}, st1{};
+  // prologue: stage first tile
+  if (s_alive) {
+    st0 = *reinterpret_cast<const bf16x8*>(s_xp + k0 + s_col8 * 8);
+    st1 = *reinterpret_cast<const bf16x8*>(s_xp + k0 + 32 + s_col8 * 8);
+  }
+  *reinterpret_cast<bf16x8*>(&x_lds[0][s_row * XS + s_col8 * 8]) = st0;
+  *reinterpret_cast<bf16x8*>(&x_lds[0][s_row * XS + 32 + s_col8 * 8]) = st1;
+  __syncthreads();
+
+  int buf = 0;
+  for (int k = k0; k < k1; k += KC) {
+    const bool has_next = (k + KC < k1);
+    if (has_next) {  // issue next tile's global loads early
+      st0 = bf16x8{};
+      st1 = bf16x8{};
+      if (s_alive) {
+        st0 = *reinterpret_cast<const bf16x8*>(s_xp + k + KC + s_col8 * 8);
+        st1 = *reinterpret_cast<const bf16x8*>(s_xp + k + KC + 32 + s_col8 * 8);
+      }
     }
 #pragma unroll
-    for (int ms = 0; ms < 4; ++ms)
-      acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bfrag, acc[ms],
-                                                        0, 0, 0);
+    for (int kk = 0; kk < KC; kk += 32) {
+      const long koff = k + kk + lg * 8;
+      // weights stream once per CU: non-temporal (guide: nt-weights)
+      bf16v8 bfrag = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16v8*>(wp + koff));
+      bf16v8 a[4];
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        a[ms] = *reinterpret_cast<const bf16v8*>(
+            &x_lds[buf][(ms * 16 + l16) * XS + kk + lg * 8]);
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bfrag,
+                                                          acc[ms], 0, 0, 0);
+    }
+    if (has_next) {  // write pass after the MFMAs (T14)
+      *reinterpret_cast<bf16x8*>(&x_lds[buf ^ 1][s_row * XS + s_col8 * 8]) = st0;
+      *reinterpret_cast<bf16x8*>(&x_lds[buf ^ 1][s_row * XS + 32 + s_col8 * 8]) = st1;
+    }
+    __syncthreads();
+    buf ^= 1;
   }
 
   // epilogue: C lane holds rows (lg*4 + r) of each m-sub, col l16
@@ -118,11 +144,11 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(w.size(1) == K);
   TORCH_CHECK(M <= 64, "skinny_gemm: M <= 64");
   TORCH_CHECK(N % 64 == 0, "skinny_gemm: N % 64 == 0");
-  TORCH_CHECK(K % 32 == 0, "skinny_gemm: K % 32 == 0");
+  TORCH_CHECK(K % 64 == 0, "skinny_gemm: K % 64 == 0");
   auto stream = at::hip::getCurrentHIPStream();
   const int ntiles = N / 64;
-  TORCH_CHECK(split >= 1 && (K % (32 * split)) == 0,
-              "invalid split for K");
+  TORCH_CHECK(split >= 1 && (K % (64 * split)) == 0,
+              "invalid split for K (k slices must be 64-aligned)");
   const int kps = K / (int)split;
   const bool full = (M == 64);
 #define SG_LAUNCH(SPLIT_, FULL_, OUTP)                                         \
