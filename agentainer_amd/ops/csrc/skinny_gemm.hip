@@ -51,10 +51,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   // 4 waves reading x straight from global exceeded the per-CU L2
   // bandwidth (x re-read once per wave instead of once per block).
   // Double-buffered, T14 split: issue next tile's loads early, ds_write
-  // after the MFMAs (guide §6 G15). Row stride 40 shorts (80 B) makes the
-  // b128 fragment reads conflict-free (row*20 mod 64 distinct over 16).
+  // after the MFMAs (guide §6 G15). Row stride 72 shorts (144 B) makes the
+  // b128 fragment reads conflict-free (row*36 mod 64 distinct over 16).
   constexpr int KC = 64;
-  constexpr int XS = 40;
+  constexpr int XS = KC + 8;
   __shared__ short x_lds[2][64 * XS];
   const int s_row = threadIdx.x % 64;        // staging: this thread's x row
   const int s_col8 = threadIdx.x / 64;       // covers cols {0..3}*8, +32 next
