@@ -167,7 +167,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     roofline); everything else to hipBLASLt via F.linear."""
     M, K = x.shape
     N = w.size(0)
-    if x.is_cuda and M <= 64 and N % 64 == 0 and K % 64 == 0:
+    if x.is_cuda and M <= 64 and N % 64 == 0 and K % 256 == 0:
         mod = _dispatch("skinny_gemm", x)
         split = _skinny_split(N // 64, K)
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)

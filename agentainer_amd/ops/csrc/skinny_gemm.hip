@@ -47,61 +47,65 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 
   const short* wp = w + (long)(n0 + l16) * K;
 
-  // x tile [64][KC] staged through LDS once per BLOCK per k-chunk — the
-  // 4 waves reading x straight from global exceeded the per-CU L2
-  // bandwidth (x re-read once per wave instead of once per block).
+  // x tile [64][KC] staged through LDS once per BLOCK per k-chunk (waves
+  // reading x straight from global exceeded the per-CU L2 bandwidth).
   // Double-buffered, T14 split: issue next tile's loads early, ds_write
-  // after the MFMAs (guide §6 G15). Row stride 72 shorts (144 B) makes the
-  // b128 fragment reads conflict-free (row*36 mod 64 distinct over 16).
-  constexpr int KC = 64;
+  // after the MFMAs (guide §6 G15). KC=256 so each wave holds EIGHT
+  // in-flight nt W loads per chunk (deep unroll, late vmcnt — the
+  // decode-weights idiom): ~8 KB of W in flight per CU covers HBM latency.
+  // Row stride 264 shorts: fragment-read bank offsets row*132 dwords
+  // (mod 64 = row*4) are distinct over a 16-lane group — conflict-free.
+  constexpr int KC = 256;
   constexpr int XS = KC + 8;
   __shared__ short x_lds[2][64 * XS];
   const int s_row = threadIdx.x % 64;        // staging: this thread's x row
-  const int s_col8 = threadIdx.x / 64;       // covers cols {0..3}*8, +32 next
+  const int s_col0 = (threadIdx.x / 64) * 8; // + 32*i, i in 0..7
   const short* s_xp = x + (long)(FULL ? s_row : min(s_row, M - 1)) * K;
   const bool s_alive = FULL || s_row < M;
 
-  bf16x8 st0{}, st1{};
-  // prologue: stage first tile
-  if (s_alive) {
-    st0 = *reinterpret_cast<const bf16x8*>(s_xp + k0 + s_col8 * 8);
-    st1 = *reinterpret_cast<const bf16x8*>(s_xp + k0 + 32 + s_col8 * 8);
-  }
-  *reinterpret_cast<bf16x8*>(&x_lds[0][s_row * XS + s_col8 * 8]) = st0;
-  *reinterpret_cast<bf16x8*>(&x_lds[0][s_row * XS + 32 + s_col8 * 8]) = st1;
+  bf16x8 st[8];
+  auto stage_load = [&](int k) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      st[i] = bf16x8{};
+      if (s_alive)
+        st[i] = *reinterpret_cast<const bf16x8*>(s_xp + k + s_col0 + 32 * i);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      *reinterpret_cast<bf16x8*>(
+          &x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
+  };
+
+  stage_load(k0);
+  stage_write(0);
   __syncthreads();
 
   int buf = 0;
   for (int k = k0; k < k1; k += KC) {
     const bool has_next = (k + KC < k1);
-    if (has_next) {  // issue next tile's global loads early
-      st0 = bf16x8{};
-      st1 = bf16x8{};
-      if (s_alive) {
-        st0 = *reinterpret_cast<const bf16x8*>(s_xp + k + KC + s_col8 * 8);
-        st1 = *reinterpret_cast<const bf16x8*>(s_xp + k + KC + 32 + s_col8 * 8);
-      }
-    }
+    if (has_next) stage_load(k + KC);  // x loads for tile t+1 in flight
+    // all 8 W fragments of this chunk issued before any MFMA consumes one
+    bf16v8 bw[8];
 #pragma unroll
-    for (int kk = 0; kk < KC; kk += 32) {
-      const long koff = k + kk + lg * 8;
-      // weights stream once per CU: non-temporal (guide: nt-weights)
-      bf16v8 bfrag = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16v8*>(wp + koff));
+    for (int i = 0; i < 8; ++i)
+      bw[i] = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16v8*>(wp + k + i * 32 + lg * 8));
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
       bf16v8 a[4];
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms)
         a[ms] = *reinterpret_cast<const bf16v8*>(
-            &x_lds[buf][(ms * 16 + l16) * XS + kk + lg * 8]);
+            &x_lds[buf][(ms * 16 + l16) * XS + i * 32 + lg * 8]);
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms)
-        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bfrag,
+        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bw[i],
                                                           acc[ms], 0, 0, 0);
     }
-    if (has_next) {  // write pass after the MFMAs (T14)
-      *reinterpret_cast<bf16x8*>(&x_lds[buf ^ 1][s_row * XS + s_col8 * 8]) = st0;
-      *reinterpret_cast<bf16x8*>(&x_lds[buf ^ 1][s_row * XS + 32 + s_col8 * 8]) = st1;
-    }
+    if (has_next) stage_write(buf ^ 1);  // write pass after the MFMAs (T14)
     __syncthreads();
     buf ^= 1;
   }
@@ -144,11 +148,11 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(w.size(1) == K);
   TORCH_CHECK(M <= 64, "skinny_gemm: M <= 64");
   TORCH_CHECK(N % 64 == 0, "skinny_gemm: N % 64 == 0");
-  TORCH_CHECK(K % 64 == 0, "skinny_gemm: K % 64 == 0");
+  TORCH_CHECK(K % 256 == 0, "skinny_gemm: K % 256 == 0");
   auto stream = at::hip::getCurrentHIPStream();
   const int ntiles = N / 64;
-  TORCH_CHECK(split >= 1 && (K % (64 * split)) == 0,
-              "invalid split for K (k slices must be 64-aligned)");
+  TORCH_CHECK(split >= 1 && (K % (256 * split)) == 0,
+              "invalid split for K (k slices must be 256-aligned)");
   const int kps = K / (int)split;
   const bool full = (M == 64);
 #define SG_LAUNCH(SPLIT_, FULL_, OUTP)                                         \
