@@ -87,6 +87,8 @@ class ModelInstance:
         self.model = LlamaForCausalLM(cfg, device=device)
         if weights_path:
             self.model.load_safetensors(weights_path)
+        if device.startswith("cuda"):
+            self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))
         n_pages = self._pool_pages(cfg, page_size, device, engine_cfg)
         self.kvm = KVCacheManager(cfg.n_layers, cfg.n_kv_heads, cfg.head_dim,

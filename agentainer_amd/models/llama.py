@@ -83,10 +83,12 @@ class LlamaAttention(torch.nn.Module):
             torch.empty(q_out + 2 * kv_out, cfg.hidden_size, dtype=torch.bfloat16))
         self.o_proj = torch.nn.Parameter(
             torch.empty(cfg.hidden_size, q_out, dtype=torch.bfloat16))
+        self.qkv_packed = None
+        self.o_packed = None
 
     def forward(self, h, k_cache, v_cache, md: AttnMetadata, cos_sin, tp_group=None):
         T = h.size(0)
-        qkv = ops.linear(h, self.qkv_proj)
+        qkv = ops.linear(h, self.qkv_proj, self.qkv_packed)
         q_sz = self.n_heads * self.head_dim
         kv_sz = self.n_kv * self.head_dim
         # strided views into the fused GEMM output — the HIP kernels take
@@ -105,7 +107,7 @@ class LlamaAttention(torch.nn.Module):
         else:
             ops.paged_decode_attention(out, q, k_cache, v_cache, md.page_table,
                                        md.seq_lens, self.scale)
-        o = ops.linear(out.view(T, q_sz), self.o_proj)
+        o = ops.linear(out.view(T, q_sz), self.o_proj, self.o_packed)
         if tp_group is not None:
             torch.distributed.all_reduce(o, group=tp_group)
         return o
@@ -120,14 +122,16 @@ class LlamaMLP(torch.nn.Module):
         self.down = torch.nn.Parameter(
             torch.empty(cfg.hidden_size, inter, dtype=torch.bfloat16))
         self.inter = inter
+        self.gate_up_packed = None
+        self.down_packed = None
 
     def forward(self, h, tp_group=None):
-        gu = ops.linear(h, self.gate_up)
+        gu = ops.linear(h, self.gate_up, self.gate_up_packed)
         gate, up = gu[:, :self.inter], gu[:, self.inter:]
         act = torch.empty(gu.size(0), self.inter, dtype=gu.dtype,
                           device=gu.device)
         ops.silu_mul(act, gate, up)
-        out = ops.linear(act, self.down)
+        out = ops.linear(act, self.down, self.down_packed)
         if tp_group is not None:
             torch.distributed.all_reduce(out, group=tp_group)
         return out
@@ -210,7 +214,23 @@ class LlamaForCausalLM(torch.nn.Module):
         ops.fused_add_rmsnorm(final, hidden, residual, self.final_ln, self.cfg.norm_eps)
         if last_rows is not None:
             final = final[last_rows]
-        return ops.linear(final, self.lm_head)  # bf16; samplers take bf16
+        return ops.linear(final, self.lm_head,
+                          getattr(self, 'lm_head_packed', None))  # bf16
+
+    @torch.no_grad()
+    def pack_decode_weights(self):
+        """Pre-shuffle every projection weight into the fragment-linear
+        layout the skinny decode GEMM streams (ops.pack_weight). Keeps the
+        natural-layout Parameters for the prefill hipBLASLt path — the
+        packed copies double weight memory, well inside 288 GB HBM3E."""
+        for layer in self.layers:
+            layer.attn.qkv_packed = ops.pack_weight(layer.attn.qkv_proj.data)
+            layer.attn.o_packed = ops.pack_weight(layer.attn.o_proj.data)
+            layer.mlp.gate_up_packed = ops.pack_weight(layer.mlp.gate_up.data)
+            layer.mlp.down_packed = ops.pack_weight(layer.mlp.down.data)
+        self.lm_head_packed = ops.pack_weight(
+            self.lm_head.data if isinstance(self.lm_head, torch.nn.Parameter)
+            else self.lm_head)
 
     @torch.no_grad()
     def load_safetensors(self, path: str):

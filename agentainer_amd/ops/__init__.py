@@ -161,29 +161,48 @@ def _skinny_split(ntiles: int, K: int) -> int:
     return split
 
 
-def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+def pack_weight(w: torch.Tensor) -> torch.Tensor:
+    """Shuffle a static [N, K] weight into MFMA-fragment-linear layout
+    packed[n_tile][k_chunk][lane][8] so the skinny GEMM's weight stream is
+    contiguous per wave. Done once at model load; flat [N*K] result."""
+    N, K = w.shape
+    assert N % 16 == 0 and K % 32 == 0
+    p = w.view(N // 16, 16, K // 32, 4, 8).permute(0, 2, 3, 1, 4).contiguous()
+    return p.view(-1)
+
+
+def _skinny_ws(device, N: int, split: int) -> torch.Tensor:
+    if split > 1:
+        key = (device, N, split)
+        ws = _ws_cache.get(key)
+        if ws is None or ws.numel() < split * 64 * N:
+            ws = torch.empty(split * 64 * N, dtype=torch.float32, device=device)
+            _ws_cache[key] = ws
+        return ws
+    ws = _EMPTY_WS.get(device)
+    if ws is None:
+        ws = torch.empty(1, dtype=torch.float32, device=device)
+        _EMPTY_WS[device] = ws
+    return ws
+
+
+def linear(x: torch.Tensor, w: torch.Tensor,
+           w_packed: Optional[torch.Tensor] = None) -> torch.Tensor:
     """x[M,K] @ w[N,K]^T. Decode-shaped (M<=64) GPU GEMMs go to the
-    hand-written skinny MFMA kernel (weight streaming at the HBM
-    roofline); everything else to hipBLASLt via F.linear."""
+    hand-written skinny MFMA kernel — packed-weight variant when the
+    caller pre-packed W (streams at the HBM roofline); everything else to
+    hipBLASLt via F.linear."""
     M, K = x.shape
     N = w.size(0)
     if x.is_cuda and M <= 64 and N % 64 == 0 and K % 256 == 0:
         mod = _dispatch("skinny_gemm", x)
         split = _skinny_split(N // 64, K)
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
-        if split > 1:
-            key = (x.device, N, split)
-            ws = _ws_cache.get(key)
-            if ws is None or ws.numel() < split * 64 * N:
-                ws = torch.empty(split * 64 * N, dtype=torch.float32,
-                                 device=x.device)
-                _ws_cache[key] = ws
+        ws = _skinny_ws(x.device, N, split)
+        if w_packed is not None:
+            mod.skinny_gemm_packed(out, x.contiguous(), w_packed, N, K, ws, split)
         else:
-            ws = _EMPTY_WS.get(x.device)
-            if ws is None:
-                ws = torch.empty(1, dtype=torch.float32, device=x.device)
-                _EMPTY_WS[x.device] = ws
-        mod.skinny_gemm(out, x.contiguous(), w, ws, split)
+            mod.skinny_gemm(out, x.contiguous(), w, ws, split)
         return out
     return torch.nn.functional.linear(x, w)
 

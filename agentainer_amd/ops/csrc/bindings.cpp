@@ -25,6 +25,9 @@ void gather_kv_pages(torch::Tensor dst, torch::Tensor k_cache,
                      torch::Tensor v_cache, torch::Tensor page_ids);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws, long split);
+void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
+                        torch::Tensor w_packed, long N, long K,
+                        torch::Tensor ws, long split);
 void mfma_probe(torch::Tensor c, torch::Tensor a, torch::Tensor b);
 void scatter_kv_pages(torch::Tensor k_cache, torch::Tensor v_cache,
                       torch::Tensor src, torch::Tensor page_ids);
@@ -44,6 +47,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topp_sample", &topp_sample, "temperature + top-p sampling per row");
   m.def("gather_kv_pages", &gather_kv_pages, "pages -> host-shaped buffer");
   m.def("skinny_gemm", &skinny_gemm, "out[M,N] = x[M,K] @ W[N,K]^T, M<=64");
+  m.def("skinny_gemm_packed", &skinny_gemm_packed, "packed-weight skinny GEMM");
   m.def("mfma_probe", &mfma_probe, "single 16x16x32 bf16 MFMA on prepacked fragments");
   m.def("scatter_kv_pages", &scatter_kv_pages, "buffer -> pages");
 }

@@ -176,3 +176,145 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   }
 #undef SG_LAUNCH
 }
+
+// ---------------------------------------------------------------------------
+// Packed-weight variant: W pre-shuffled at model load into fragment-linear
+// layout packed[n_tile][k_chunk][lane][8] (ops.pack_weight), so each wave
+// streams its 16-row x K weight block CONTIGUOUSLY — 1 KB per load
+// instruction, one stream per wave. This is what buys the HBM roofline;
+// the unpacked variant's 16 scattered 64-B row segments per instruction
+// cap out around 2-3 TB/s.
+
+namespace {
+
+template <bool SPLIT, bool FULL>
+__global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
+    void* __restrict__ out, const short* __restrict__ x,
+    const short* __restrict__ wp_packed, int M, int N, int K,
+    int k_per_split) {
+  const int n_tile = blockIdx.x * 4 + threadIdx.x / WAVE;  // 16-col tile
+  const int split = blockIdx.y;
+  const int lane = threadIdx.x % WAVE;
+  const int l16 = lane % 16;
+  const int lg = lane / 16;
+  const int k0 = split * k_per_split;
+  const int k1 = min(k0 + k_per_split, K);
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) acc[ms] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // this wave's contiguous weight stream
+  const short* wp = wp_packed + ((long)n_tile * (K / 32) + k0 / 32) * 512 +
+                    lane * 8;
+
+  constexpr int KC = 256;
+  constexpr int XS = KC + 8;
+  __shared__ short x_lds[2][64 * XS];
+  const int s_row = threadIdx.x % 64;
+  const int s_col0 = (threadIdx.x / 64) * 8;
+  const short* s_xp = x + (long)(FULL ? s_row : min(s_row, M - 1)) * K;
+  const bool s_alive = FULL || s_row < M;
+
+  bf16x8 st[8];
+  auto stage_load = [&](int k) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      st[i] = bf16x8{};
+      if (s_alive)
+        st[i] = *reinterpret_cast<const bf16x8*>(s_xp + k + s_col0 + 32 * i);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      *reinterpret_cast<bf16x8*>(
+          &x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
+  };
+
+  stage_load(k0);
+  stage_write(0);
+  __syncthreads();
+
+  int buf = 0;
+  long woff = 0;  // shorts into this wave's packed stream
+  for (int k = k0; k < k1; k += KC) {
+    const bool has_next = (k + KC < k1);
+    if (has_next) stage_load(k + KC);
+    bf16v8 bw[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      bw[i] = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16v8*>(wp + woff + (long)i * 512));
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      bf16v8 a[4];
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        a[ms] = *reinterpret_cast<const bf16v8*>(
+            &x_lds[buf][(ms * 16 + l16) * XS + i * 32 + lg * 8]);
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bw[i],
+                                                          acc[ms], 0, 0, 0);
+    }
+    if (has_next) stage_write(buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+    woff += 8 * 512;
+  }
+
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = ms * 16 + lg * 4 + r;
+      if (row >= M) continue;
+      const int col = n_tile * 16 + l16;
+      if (SPLIT) {
+        float* o = reinterpret_cast<float*>(out);
+        o[((long)split * M + row) * N + col] = acc[ms][r];
+      } else {
+        short* o = reinterpret_cast<short*>(out);
+        o[(long)row * N + col] = f2bits(acc[ms][r]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
+                        torch::Tensor w_packed, long N, long K,
+                        torch::Tensor ws, long split) {
+  TORCH_CHECK(x.is_contiguous() && w_packed.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w_packed.scalar_type() == at::kBFloat16);
+  const int M = x.size(0);
+  TORCH_CHECK(x.size(1) == K);
+  TORCH_CHECK(w_packed.numel() == N * K, "packed weight numel mismatch");
+  TORCH_CHECK(M <= 64 && N % 64 == 0 && K % 256 == 0);
+  TORCH_CHECK(split >= 1 && (K % (256 * split)) == 0);
+  auto stream = at::hip::getCurrentHIPStream();
+  const int ntiles = (int)N / 64;
+  const int kps = (int)K / (int)split;
+  const bool full = (M == 64);
+#define SGP_LAUNCH(SPLIT_, FULL_, OUTP)                                        \
+  hipLaunchKernelGGL((skinny_gemm_packed_kernel<SPLIT_, FULL_>),               \
+                     dim3(ntiles, SPLIT_ ? (int)split : 1), dim3(256), 0,      \
+                     stream, OUTP, (const short*)x.data_ptr(),                 \
+                     (const short*)w_packed.data_ptr(), M, (int)N, (int)K, kps)
+  if (split == 1) {
+    if (full) SGP_LAUNCH(false, true, out.data_ptr());
+    else SGP_LAUNCH(false, false, out.data_ptr());
+  } else {
+    TORCH_CHECK(ws.numel() >= (long)split * M * N && ws.scalar_type() == at::kFloat);
+    if (full) SGP_LAUNCH(true, true, ws.data_ptr());
+    else SGP_LAUNCH(true, false, ws.data_ptr());
+    const long mn = (long)M * N;
+    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 255) / 256),
+                       dim3(256), 0, stream, (short*)out.data_ptr(),
+                       ws.data_ptr<float>(), mn, split);
+  }
+#undef SGP_LAUNCH
+}

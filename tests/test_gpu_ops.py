@@ -270,3 +270,16 @@ def test_skinny_gemm_matches_torch(M, N, K):
     diff = (out.float() - want).abs()
     rel = diff.max().item() / max(want.abs().max().item(), 1e-6)
     assert rel < 0.02, f"skinny_gemm rel diff {rel} (M={M} N={N} K={K})"
+
+
+def test_skinny_gemm_packed_matches_torch():
+    torch.manual_seed(11)
+    from agentainer_amd.ops import linear, pack_weight
+    for M, N, K in [(64, 4096, 4096), (64, 6144, 4096), (33, 4096, 14336),
+                    (64, 28672, 4096)]:
+        x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.1)
+        w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.1)
+        out = linear(x, w, pack_weight(w))
+        want = torch.nn.functional.linear(x.float(), w.float())
+        rel = (out.float() - want).abs().max().item() / want.abs().max().item()
+        assert rel < 0.02, f"packed skinny rel {rel} M={M} N={N} K={K}"
