@@ -155,8 +155,9 @@ _EMPTY_WS = {}
 
 
 def _skinny_split(ntiles: int, K: int) -> int:
+    # target >= 512 blocks (2 per CU) so chunk barriers overlap across blocks
     split = 1
-    while split < 8 and ntiles * split * 2 < 512 and (K // (split * 2)) % 256 == 0:
+    while split < 16 and ntiles * split < 512 and K % (256 * split * 2) == 0:
         split *= 2
     return split
 

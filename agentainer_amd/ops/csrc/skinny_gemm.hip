@@ -232,20 +232,31 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
           &x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
   };
 
+  // W stream is software-pipelined ACROSS the per-chunk barrier: chunk
+  // t+1's eight nt loads are issued while chunk t computes, so no chunk
+  // pays the ~900-cycle HBM latency cold (the barrier would otherwise
+  // serialize it at 1-2 blocks/CU).
+  bf16v8 bw_cur[8], bw_nxt[8];
+  auto w_load = [&](bf16v8 (&dst)[8], long woff) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      dst[i] = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16v8*>(wp + woff + (long)i * 512));
+  };
+
   stage_load(k0);
   stage_write(0);
+  w_load(bw_cur, 0);
   __syncthreads();
 
   int buf = 0;
-  long woff = 0;  // shorts into this wave's packed stream
+  long woff = 8 * 512;  // shorts into this wave's packed stream (chunk t+1)
   for (int k = k0; k < k1; k += KC) {
     const bool has_next = (k + KC < k1);
-    if (has_next) stage_load(k + KC);
-    bf16v8 bw[8];
-#pragma unroll
-    for (int i = 0; i < 8; ++i)
-      bw[i] = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16v8*>(wp + woff + (long)i * 512));
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_nxt, woff);
+    }
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       bf16v8 a[4];
@@ -255,13 +266,15 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
             &x_lds[buf][(ms * 16 + l16) * XS + i * 32 + lg * 8]);
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms)
-        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bw[i],
+        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bw_cur[i],
                                                           acc[ms], 0, 0, 0);
     }
     if (has_next) stage_write(buf ^ 1);
     __syncthreads();
     buf ^= 1;
     woff += 8 * 512;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) bw_cur[i] = bw_nxt[i];
   }
 
 #pragma unroll
