@@ -152,6 +152,8 @@ def topp_sample(out, logits, temps, top_ps, seeds):
 
 _ws_cache = {}
 _EMPTY_WS = {}
+SKINNY_DISABLED = os.environ.get("AGENTAINER_DISABLE_SKINNY", "0") == "1"
+SKINNY_NT = os.environ.get("AGENTAINER_SKINNY_NT", "1") == "1"
 
 
 def _skinny_split(ntiles: int, K: int) -> int:
@@ -195,13 +197,15 @@ def linear(x: torch.Tensor, w: torch.Tensor,
     hipBLASLt via F.linear."""
     M, K = x.shape
     N = w.size(0)
-    if x.is_cuda and M <= 64 and N % 64 == 0 and K % 256 == 0:
+    if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 256 == 0
+            and not SKINNY_DISABLED):
         mod = _dispatch("skinny_gemm", x)
         split = _skinny_split(N // 64, K)
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
         ws = _skinny_ws(x.device, N, split)
         if w_packed is not None:
-            mod.skinny_gemm_packed(out, x.contiguous(), w_packed, N, K, ws, split)
+            mod.skinny_gemm_packed(out, x.contiguous(), w_packed, N, K, ws,
+                                   split, SKINNY_NT)
         else:
             mod.skinny_gemm(out, x.contiguous(), w, ws, split)
         return out

@@ -27,7 +27,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws, long split);
 void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
                         torch::Tensor w_packed, long N, long K,
-                        torch::Tensor ws, long split);
+                        torch::Tensor ws, long split, bool nt);
 void mfma_probe(torch::Tensor c, torch::Tensor a, torch::Tensor b);
 void scatter_kv_pages(torch::Tensor k_cache, torch::Tensor v_cache,
                       torch::Tensor src, torch::Tensor page_ids);

@@ -187,7 +187,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
 
 namespace {
 
-template <bool SPLIT, bool FULL>
+template <bool SPLIT, bool FULL, bool NT>
 __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     void* __restrict__ out, const short* __restrict__ x,
     const short* __restrict__ wp_packed, int M, int N, int K,
@@ -239,9 +239,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   bf16v8 bw_cur[8], bw_nxt[8];
   auto w_load = [&](bf16v8 (&dst)[8], long woff) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
-      dst[i] = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16v8*>(wp + woff + (long)i * 512));
+    for (int i = 0; i < 8; ++i) {
+      const bf16v8* p = reinterpret_cast<const bf16v8*>(wp + woff + (long)i * 512);
+      dst[i] = NT ? __builtin_nontemporal_load(p) : *p;
+    }
   };
 
   stage_load(k0);
@@ -299,7 +300,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
 
 void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
                         torch::Tensor w_packed, long N, long K,
-                        torch::Tensor ws, long split) {
+                        torch::Tensor ws, long split, bool nt) {
   TORCH_CHECK(x.is_contiguous() && w_packed.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               w_packed.scalar_type() == at::kBFloat16);
@@ -312,22 +313,28 @@ void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
   const int ntiles = (int)N / 64;
   const int kps = (int)K / (int)split;
   const bool full = (M == 64);
-#define SGP_LAUNCH(SPLIT_, FULL_, OUTP)                                        \
-  hipLaunchKernelGGL((skinny_gemm_packed_kernel<SPLIT_, FULL_>),               \
+#define SGP_LAUNCH(SPLIT_, FULL_, NT_, OUTP)                                   \
+  hipLaunchKernelGGL((skinny_gemm_packed_kernel<SPLIT_, FULL_, NT_>),          \
                      dim3(ntiles, SPLIT_ ? (int)split : 1), dim3(256), 0,      \
                      stream, OUTP, (const short*)x.data_ptr(),                 \
                      (const short*)w_packed.data_ptr(), M, (int)N, (int)K, kps)
+#define SGP_DISPATCH(SPLIT_, OUTP)                                             \
+  do {                                                                         \
+    if (full && nt) SGP_LAUNCH(SPLIT_, true, true, OUTP);                      \
+    else if (full) SGP_LAUNCH(SPLIT_, true, false, OUTP);                      \
+    else if (nt) SGP_LAUNCH(SPLIT_, false, true, OUTP);                        \
+    else SGP_LAUNCH(SPLIT_, false, false, OUTP);                               \
+  } while (0)
   if (split == 1) {
-    if (full) SGP_LAUNCH(false, true, out.data_ptr());
-    else SGP_LAUNCH(false, false, out.data_ptr());
+    SGP_DISPATCH(false, out.data_ptr());
   } else {
     TORCH_CHECK(ws.numel() >= (long)split * M * N && ws.scalar_type() == at::kFloat);
-    if (full) SGP_LAUNCH(true, true, ws.data_ptr());
-    else SGP_LAUNCH(true, false, ws.data_ptr());
+    SGP_DISPATCH(true, ws.data_ptr());
     const long mn = (long)M * N;
     hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 255) / 256),
                        dim3(256), 0, stream, (short*)out.data_ptr(),
                        ws.data_ptr<float>(), mn, split);
   }
+#undef SGP_DISPATCH
 #undef SGP_LAUNCH
 }
