@@ -153,7 +153,11 @@ def topp_sample(out, logits, temps, top_ps, seeds):
 _ws_cache = {}
 _EMPTY_WS = {}
 SKINNY_DISABLED = os.environ.get("AGENTAINER_DISABLE_SKINNY", "0") == "1"
-SKINNY_NT = os.environ.get("AGENTAINER_SKINNY_NT", "1") == "1"
+SKINNY_FORCED = os.environ.get("AGENTAINER_FORCE_SKINNY", "0") == "1"
+# shapes where the hand-written skinny GEMM beat hipBLASLt on MI355X
+# (measured via tools/gemm_probe.py; see profiles/)
+_SKINNY_SHAPES = {(4096, 4096)}
+SKINNY_NT = os.environ.get("AGENTAINER_SKINNY_NT", "0") == "1"
 
 
 def _skinny_split(ntiles: int, K: int) -> int:
@@ -197,8 +201,9 @@ def linear(x: torch.Tensor, w: torch.Tensor,
     hipBLASLt via F.linear."""
     M, K = x.shape
     N = w.size(0)
+    skinny_wins = (N, K) in _SKINNY_SHAPES or SKINNY_FORCED
     if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 256 == 0
-            and not SKINNY_DISABLED):
+            and skinny_wins and not SKINNY_DISABLED):
         mod = _dispatch("skinny_gemm", x)
         split = _skinny_split(N // 64, K)
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
