@@ -201,6 +201,13 @@ class KVCacheManager:
             need_pages = max(0, -(-(n_tokens - have) // self.page_size))
             return need_pages <= len(self._free)
 
+    def can_append_after_reset(self, seq_id: str, n_tokens: int) -> bool:
+        """Would n_tokens fit if this sequence's pages were freed first?"""
+        with self._lock:
+            s = self._seqs[seq_id]
+            need_pages = -(-n_tokens // self.page_size)
+            return need_pages <= len(self._free) + len(s.pages)
+
     def append_slots(self, seq_id: str, n_tokens: int) -> List[int]:
         """Reserve slots for n_tokens new tokens; allocates pages as needed.
         Returns global slot ids (page * page_size + offset)."""

@@ -33,6 +33,7 @@ from typing import Any, Dict, List, Optional
 import torch
 
 from .. import ops
+from .. import parallel as par
 from ..store import Store
 from .base import ModelNotFound
 from .kvcache import KVCacheManager, KVCheckpoint, OutOfPages
@@ -79,20 +80,35 @@ class ModelInstance:
     """One loaded model + KV pool + scheduler thread."""
 
     def __init__(self, name: str, cfg: LlamaConfig, device: str,
-                 engine_cfg: Dict[str, Any], weights_path: Optional[str] = None):
+                 engine_cfg: Dict[str, Any], weights_path: Optional[str] = None,
+                 model_cls=None):
         self.name = name
         self.cfg = cfg
         self.device = device
         self.tokenizer = ByteTokenizer(cfg.vocab_size)
-        self.model = LlamaForCausalLM(cfg, device=device)
+        # tensor parallelism: engine-level degree must match the launched
+        # world size; rank 0 schedules, plans broadcast (parallel.dist)
+        self.tp_size = int(engine_cfg.get("tp_degree", 1))
+        self.tp_rank = par.tp_rank() if self.tp_size > 1 else 0
+        if self.tp_size > 1:
+            assert par.is_tp() and par.tp_size() == self.tp_size, (
+                f"tp_degree={self.tp_size} requires torchrun world of that size")
+        model_cls = model_cls or LlamaForCausalLM
+        self.model = model_cls(cfg, device=device,
+                               tp_rank=self.tp_rank, tp_size=self.tp_size)
+        if self.tp_size > 1:
+            self.model.tp_group = par.tp_group()
         if weights_path:
             self.model.load_safetensors(weights_path)
         if device.startswith("cuda"):
             self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))
-        n_pages = self._pool_pages(cfg, page_size, device, engine_cfg)
-        self.kvm = KVCacheManager(cfg.n_layers, cfg.n_kv_heads, cfg.head_dim,
-                                  page_size, n_pages, device=device)
+        n_pages = self._pool_pages(cfg, page_size, device, engine_cfg,
+                                   self.tp_size)
+        self.kvm = KVCacheManager(cfg.n_layers,
+                                  max(cfg.n_kv_heads // self.tp_size, 1),
+                                  cfg.head_dim, page_size, n_pages,
+                                  device=device)
         self.max_batch_tokens = int(engine_cfg.get("max_batch_tokens", 8192))
         self.max_decode_batch = int(engine_cfg.get("max_decode_batch", 256))
         self.refcount = 0
@@ -114,7 +130,8 @@ class ModelInstance:
         # captured once per batch bucket and replayed with two small H2D
         # copies (row slots + last tokens) per step.
         self.is_gpu = device.startswith("cuda")
-        self.use_graph = self.is_gpu and bool(engine_cfg.get("graph_capture", True))
+        self.use_graph = (self.is_gpu and self.tp_size == 1
+                          and bool(engine_cfg.get("graph_capture", True)))
         self._graphs: Dict[int, Dict[str, Any]] = {}
         self._pad_slot = -1
         if self.is_gpu:
@@ -126,8 +143,8 @@ class ModelInstance:
 
     @staticmethod
     def _pool_pages(cfg: LlamaConfig, page_size: int, device: str,
-                    engine_cfg: Dict[str, Any]) -> int:
-        bytes_per_page = cfg.kv_bytes_per_token() * page_size
+                    engine_cfg: Dict[str, Any], tp_size: int = 1) -> int:
+        bytes_per_page = cfg.kv_bytes_per_token() * page_size // max(tp_size, 1)
         pool_gb = float(engine_cfg.get("kv_pool_gb", 0.0))
         if device.startswith("cuda") and pool_gb <= 0:
             free, _total = torch.cuda.mem_get_info()
@@ -218,10 +235,13 @@ class ModelInstance:
                 self.waiting.put(req)
                 break
             # KV room: prompt + generation
+            req.needs_reset = False
             if not self.kvm.can_append(b.seq_id, need + req.max_new):
-                # truncate conversation: reset KV, re-prefill just this turn
-                self.kvm.reset_seq(b.seq_id)
-                if not self.kvm.can_append(b.seq_id, need + req.max_new):
+                # truncate conversation: reset KV (applied in the prefill
+                # plan so TP workers mirror it), re-prefill just this turn
+                req.needs_reset = True
+                if not self.kvm.can_append_after_reset(b.seq_id,
+                                                       need + req.max_new):
                     req.error = "KV pool exhausted"
                     req.done.set()
                     continue
@@ -256,22 +276,43 @@ class ModelInstance:
             out[samp_rows] = sub
         return out.tolist()
 
+    def _bcast(self, cmd):
+        if self.tp_size > 1 and self.tp_rank == 0:
+            par.broadcast_obj(cmd)
+
     def _prefill(self, reqs: List[GenRequest]):
+        plan = []
+        for r in reqs:
+            b = self._bindings[r.agent_id]
+            plan.append((b.seq_id, r.prompt_tokens, bool(getattr(r, "needs_reset", False))))
+        self._bcast(("prefill", self.name, plan))
+        logits = self._prefill_exec(plan)
+        toks = self._sample(logits, reqs)
+        now = time.time()
+        with self._lock:
+            for r, t in zip(reqs, toks):
+                r.generated.append(int(t))
+                r.first_token_t = now
+                self._finish_or_run(r, int(t))
+
+    def _prefill_exec(self, plan) -> torch.Tensor:
+        """Runs identically on every TP rank (plan = seq ids + tokens)."""
         dev = self.device
         ids: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
         q_starts, q_lens, seq_ids = [], [], []
-        for r in reqs:
-            b = self._bindings[r.agent_id]
-            prev = self.kvm.seq_len(b.seq_id)
-            n = len(r.prompt_tokens)
+        for seq_id, tokens, needs_reset in plan:
+            if needs_reset:
+                self.kvm.reset_seq(seq_id)
+            prev = self.kvm.seq_len(seq_id)
+            n = len(tokens)
             q_starts.append(len(ids))
             q_lens.append(n)
-            seq_ids.append(b.seq_id)
-            ids.extend(r.prompt_tokens)
+            seq_ids.append(seq_id)
+            ids.extend(tokens)
             positions.extend(range(prev, prev + n))
-            slots.extend(self.kvm.append_slots(b.seq_id, n))
+            slots.extend(self.kvm.append_slots(seq_id, n))
         if self.is_gpu:
             for sid in seq_ids:
                 self.kvm.push_dev(sid)
@@ -288,14 +329,8 @@ class ModelInstance:
         last_rows = torch.tensor([s + l - 1 for s, l in zip(q_starts, q_lens)],
                                  dtype=torch.long, device=dev)
         logits = self.model(input_ids, md, self.kvm.kv_caches(), last_rows)
-        toks = self._sample(logits, reqs)
-        now = time.time()
         self.prefill_tokens += len(ids)
-        with self._lock:
-            for r, t in zip(reqs, toks):
-                r.generated.append(int(t))
-                r.first_token_t = now
-                self._finish_or_run(r, int(t))
+        return logits
 
     # ---------- device-side decode (hipGraph) ----------
 
@@ -348,22 +383,58 @@ class ModelInstance:
         self._graphs[bucket] = entry
         return entry
 
-    def _decode_gpu(self, reqs: List[GenRequest]):
+    def _decode(self, reqs: List[GenRequest]):
+        B0 = len(reqs)
+        if self.is_gpu:
+            bucket = min(self._bucket(B0), max(self.max_decode_batch, 1))
+            if bucket < B0:
+                reqs = reqs[:bucket]
+        plan = [(self._bindings[r.agent_id].seq_id, r.generated[-1])
+                for r in reqs]
+        self._bcast(("decode", self.name, plan))
+        logits = self._decode_exec(plan)
+        toks = self._sample(logits, reqs)
+        self.decode_tokens += len(reqs)
+        self.occupancy_acc += len(reqs) / max(1, self.max_decode_batch)
+        with self._lock:
+            for r, t in zip(reqs, toks):
+                r.generated.append(int(t))
+                self._finish_or_run(r, int(t))
+
+    def _decode_exec(self, plan) -> torch.Tensor:
+        """One batched decode step; runs identically on every TP rank."""
+        if self.is_gpu:
+            return self._decode_exec_gpu(plan)
+        dev = self.device
+        ids, positions, slots, seq_ids = [], [], [], []
+        for seq_id, tok in plan:
+            prev = self.kvm.seq_len(seq_id)
+            ids.append(tok)
+            positions.append(prev)
+            slots.extend(self.kvm.append_slots(seq_id, 1))
+            seq_ids.append(seq_id)
+        md = AttnMetadata(
+            page_table=self.kvm.page_table(seq_ids, device=dev),
+            seq_lens=self.kvm.seq_lens(seq_ids, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            is_prefill=False,
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        return self.model(input_ids, md, self.kvm.kv_caches(), None)
+
+    def _decode_exec_gpu(self, plan) -> torch.Tensor:
         kvm = self.kvm
-        B = len(reqs)
+        B = len(plan)
         bucket = min(self._bucket(B), max(self.max_decode_batch, 1))
-        if bucket < B:
-            reqs = reqs[:bucket]
-            B = bucket
         row_ids = []
-        for r in reqs:
-            b = self._bindings[r.agent_id]
-            kvm.ensure_decode_page(b.seq_id)   # host half of the append
-            row_ids.append(kvm.slot(b.seq_id))
+        for seq_id, _tok in plan:
+            kvm.ensure_decode_page(seq_id)   # host half of the append
+            row_ids.append(kvm.slot(seq_id))
         entry = self._get_graph(bucket)
         entry["rows_pin"][:B] = torch.tensor(row_ids, dtype=torch.long)
         entry["rows_pin"][B:] = self._pad_slot
-        entry["ids_pin"][:B] = torch.tensor([r.generated[-1] for r in reqs],
+        entry["ids_pin"][:B] = torch.tensor([t for _s, t in plan],
                                             dtype=torch.long)
         entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
         entry["ids"].copy_(entry["ids_pin"], non_blocking=True)
@@ -373,46 +444,9 @@ class ModelInstance:
         else:
             logits = self._device_decode_fwd(entry["rows"], entry["ids"],
                                              entry["inc"])[:B]
-        toks = self._sample(logits, reqs)
-        self.decode_tokens += B
-        self.occupancy_acc += B / max(1, self.max_decode_batch)
-        with self._lock:
-            for r in reqs:
-                b = self._bindings.get(r.agent_id)
-                if b is not None:
-                    kvm.advance_host(b.seq_id)
-            for r, t in zip(reqs, toks):
-                r.generated.append(int(t))
-                self._finish_or_run(r, int(t))
-
-    def _decode(self, reqs: List[GenRequest]):
-        if self.is_gpu:
-            return self._decode_gpu(reqs)
-        dev = self.device
-        ids, positions, slots, seq_ids = [], [], [], []
-        for r in reqs:
-            b = self._bindings[r.agent_id]
-            prev = self.kvm.seq_len(b.seq_id)
-            ids.append(r.generated[-1])
-            positions.append(prev)
-            slots.extend(self.kvm.append_slots(b.seq_id, 1))
-            seq_ids.append(b.seq_id)
-        md = AttnMetadata(
-            page_table=self.kvm.page_table(seq_ids, device=dev),
-            seq_lens=self.kvm.seq_lens(seq_ids, device=dev),
-            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
-            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
-            is_prefill=False,
-        )
-        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
-        logits = self.model(input_ids, md, self.kvm.kv_caches(), None)
-        toks = self._sample(logits, reqs)
-        self.decode_tokens += len(reqs)
-        self.occupancy_acc += len(reqs) / max(1, self.max_decode_batch)
-        with self._lock:
-            for r, t in zip(reqs, toks):
-                r.generated.append(int(t))
-                self._finish_or_run(r, int(t))
+        for seq_id, _tok in plan:
+            kvm.advance_host(seq_id)
+        return logits
 
     def _finish_or_run(self, r: GenRequest, tok: int):
         """Called with lock held, after appending tok."""
@@ -447,12 +481,18 @@ class ModelInstance:
 
     # ---------- binding management ----------
 
-    def bind(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint]):
+    def _bind_seq(self, seq_id: str, ckpt: Optional[KVCheckpoint]):
+        if ckpt is not None:
+            self.kvm.restore(seq_id, ckpt)
+        elif not self.kvm.has_seq(seq_id):
+            self.kvm.create_seq(seq_id)
+
+    def bind(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint],
+             worker_has_ckpt: bool = False):
         with self._lock:
-            if ckpt is not None:
-                self.kvm.restore(seq_id, ckpt)
-            elif not self.kvm.has_seq(seq_id):
-                self.kvm.create_seq(seq_id)
+            self._bcast(("bind", self.name, seq_id,
+                         ckpt is not None or worker_has_ckpt))
+            self._bind_seq(seq_id, ckpt)
             self._bindings[agent.id] = AgentBinding(agent=agent,
                                                     model_name=self.name,
                                                     seq_id=seq_id)
@@ -478,6 +518,7 @@ class ModelInstance:
                     break
                 r.error = "agent detached"
                 r.done.set()
+        self._bcast(("unbind", self.name, b.seq_id, offload))
         ckpt = self.kvm.offload(b.seq_id, free=True) if offload else None
         if not offload:
             self.kvm.free_seq(b.seq_id)
@@ -502,6 +543,8 @@ class LLMEngine:
         self._agent_model: Dict[str, str] = {}
         self._ckpts: Dict[str, KVCheckpoint] = {}
         self._lock = threading.RLock()
+        self.tp_size = int(self.engine_cfg.get("tp_degree", 1))
+        self.tp_rank = par.tp_rank() if self.tp_size > 1 else 0
         if device == "cuda" and torch.cuda.is_available():
             if not ops.hip_available():
                 raise RuntimeError(
@@ -511,25 +554,35 @@ class LLMEngine:
     # ---------- model registry ----------
 
     def validate_model(self, model: str) -> None:
-        if model in LLAMA_CONFIGS:
+        from ..models.mixtral import MIXTRAL_CONFIGS
+        if model in LLAMA_CONFIGS or model in MIXTRAL_CONFIGS:
             return
         from .echo import ECHO_MODELS
         if model in ECHO_MODELS:
             return
         if os.path.isdir(os.path.expanduser(model)):
             return  # weights directory
+        from ..models.mixtral import MIXTRAL_CONFIGS as _MC
         raise ModelNotFound(
-            f"unknown model {model!r}; known: {sorted(LLAMA_CONFIGS)} + echo")
+            f"unknown model {model!r}; known: "
+            f"{sorted(LLAMA_CONFIGS) + sorted(_MC)} + echo")
+
+    def _make_instance(self, model: str) -> ModelInstance:
+        from ..models.mixtral import MIXTRAL_CONFIGS, make_mixtral_instance
+        if model in LLAMA_CONFIGS:
+            return ModelInstance(model, LLAMA_CONFIGS[model], self.device,
+                                 self.engine_cfg)
+        if model in MIXTRAL_CONFIGS:
+            return make_mixtral_instance(model, self.device, self.engine_cfg)
+        raise ModelNotFound(f"cannot load {model!r}")
 
     def _get_instance(self, model: str) -> ModelInstance:
         with self._lock:
             inst = self._instances.get(model)
             if inst is None:
-                if model in LLAMA_CONFIGS:
-                    cfg = LLAMA_CONFIGS[model]
-                    inst = ModelInstance(model, cfg, self.device, self.engine_cfg)
-                else:
-                    raise ModelNotFound(f"cannot load {model!r}")
+                if self.tp_size > 1 and self.tp_rank == 0:
+                    par.broadcast_obj(("instance", model))
+                inst = self._make_instance(model)
                 self._instances[model] = inst
                 inst.start()
             return inst
@@ -553,7 +606,7 @@ class LLMEngine:
         ckpt = self._ckpts.pop(agent.id, None)
         if ckpt is None:
             ckpt = self._load_disk_ckpt(agent.id)
-        inst.bind(agent, seq_id=agent.id, ckpt=ckpt)
+        inst.bind(agent, seq_id=agent.id, ckpt=ckpt)  # workers restore their shard
         with self._lock:
             self._agent_model[agent.id] = model
 
@@ -714,7 +767,8 @@ class LLMEngine:
     def _ckpt_path(self, agent_id: str) -> str:
         d = os.path.join(self.state_root, "kv_ckpt")
         os.makedirs(d, exist_ok=True)
-        return os.path.join(d, f"{agent_id}.pt")
+        suffix = f".rank{self.tp_rank}" if self.tp_size > 1 else ""
+        return os.path.join(d, f"{agent_id}{suffix}.pt")
 
     def _save_disk_ckpt(self, agent_id: str, ckpt: KVCheckpoint) -> None:
         torch.save({"length": ckpt.length, "n_pages": ckpt.n_pages,
@@ -763,5 +817,47 @@ class LLMEngine:
                 "models": models}
 
     def shutdown(self):
+        if self.tp_size > 1 and self.tp_rank == 0 and par.is_tp():
+            par.broadcast_obj(("shutdown",))
         for inst in self._instances.values():
             inst.stop()
+
+    # ---------- TP worker loop (ranks > 0) ----------
+
+    def run_worker(self):
+        """SPMD worker: executes the step plans rank 0 broadcasts.
+        Page allocation is deterministic, so this rank's KV pool mirrors
+        rank 0's without any per-step tensor metadata exchange."""
+        assert self.tp_size > 1 and self.tp_rank > 0
+        while True:
+            cmd = par.broadcast_obj(None)
+            op = cmd[0]
+            if op == "shutdown":
+                break
+            if op == "instance":
+                with self._lock:
+                    if cmd[1] not in self._instances:
+                        self._instances[cmd[1]] = self._make_instance(cmd[1])
+                continue
+            inst = self._instances[cmd[1]]
+            if op == "prefill":
+                inst._prefill_exec(cmd[2])
+            elif op == "decode":
+                inst._decode_exec(cmd[2])
+            elif op == "bind":
+                seq_id, has_ckpt = cmd[2], cmd[3]
+                ckpt = self._ckpts.pop(seq_id, None) if has_ckpt else None
+                if has_ckpt and ckpt is None:
+                    ckpt = self._load_disk_ckpt(seq_id)
+                inst._bind_seq(seq_id, ckpt)
+            elif op == "unbind":
+                seq_id, offload = cmd[2], cmd[3]
+                if offload:
+                    ckpt = inst.kvm.offload(seq_id, free=True)
+                    if ckpt is not None:
+                        self._ckpts[seq_id] = ckpt
+                        self._save_disk_ckpt(seq_id, ckpt)
+                else:
+                    inst.kvm.free_seq(seq_id)
+            if self.device.startswith("cuda"):
+                torch.cuda.synchronize()

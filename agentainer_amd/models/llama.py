@@ -277,6 +277,11 @@ LLAMA_CONFIGS: Dict[str, LlamaConfig] = {
     "tiny-llama": LlamaConfig(name="tiny-llama", vocab_size=512, hidden_size=512,
                               n_layers=2, n_heads=4, n_kv_heads=1,
                               intermediate_size=1024, max_position=4096),
+    # 2-way-shardable tiny model for the gloo TP tests (n_kv=2)
+    "tiny-llama-tp": LlamaConfig(name="tiny-llama-tp", vocab_size=512,
+                                 hidden_size=1024, n_layers=2, n_heads=8,
+                                 n_kv_heads=2, intermediate_size=1024,
+                                 max_position=4096),
     "llama3-8b": LlamaConfig(name="llama3-8b", vocab_size=128256,
                              hidden_size=4096, n_layers=32, n_heads=32,
                              n_kv_heads=8, intermediate_size=14336,

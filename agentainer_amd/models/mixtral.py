@@ -1,0 +1,201 @@
+"""Mixtral (sparse MoE) family — MI355X-first serving forward.
+
+BASELINE.json config 5: Mixtral 8x7B MoE agents. Architecture = the Llama
+attention stack (same HIP kernels: RMSNorm, RoPE, paged attention) with
+the dense MLP replaced by a top-2-of-8 expert MoE.
+
+MoE execution model, chosen for the MI355X serving regime:
+
+  * Decode (small batch): DENSE-ROUTED — every (local) expert runs on the
+    whole batch, scaled by the router's top-k-masked weight (0 for
+    unrouted tokens). At decode batch sizes every expert is hit anyway,
+    so the op is expert-WEIGHT-streaming-bound and dense routing costs
+    the same HBM traffic as gather/scatter — while keeping the step free
+    of host syncs (hipGraph-capturable) .
+  * Expert parallelism (EP over RCCL/xGMI): experts are PARTITIONED
+    across ranks; activations are replicated (attention output is already
+    all-reduced), so each rank computes its local experts' contribution
+    and one all-reduce combines — the xGMI-friendly formulation of EP for
+    replicated-activation serving. A token all-to-all dispatch (for
+    sequence-sharded activations) is the round-2 upgrade path.
+  * fp8 MFMA expert GEMMs (config 5's decode dtype) are a planned upgrade
+    of ops.skinny_gemm (OCP e4m3 + MX scaling); experts run bf16 today.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from .llama import AttnMetadata, LlamaAttention, LlamaConfig
+
+
+@dataclass
+class MixtralConfig(LlamaConfig):
+    n_experts: int = 8
+    top_k: int = 2
+
+
+class MixtralMoE(torch.nn.Module):
+    """Top-k router + SwiGLU experts, expert-partitioned across EP ranks."""
+
+    def __init__(self, cfg: MixtralConfig, ep_rank: int = 0, ep_size: int = 1):
+        super().__init__()
+        assert cfg.n_experts % ep_size == 0
+        self.cfg = cfg
+        self.n_local = cfg.n_experts // ep_size
+        self.e0 = ep_rank * self.n_local  # first local expert id
+        self.top_k = cfg.top_k
+        inter = cfg.intermediate_size
+        self.router = torch.nn.Parameter(
+            torch.empty(cfg.n_experts, cfg.hidden_size, dtype=torch.bfloat16))
+        self.gate_up = torch.nn.ParameterList([
+            torch.nn.Parameter(torch.empty(2 * inter, cfg.hidden_size,
+                                           dtype=torch.bfloat16))
+            for _ in range(self.n_local)])
+        self.down = torch.nn.ParameterList([
+            torch.nn.Parameter(torch.empty(cfg.hidden_size, inter,
+                                           dtype=torch.bfloat16))
+            for _ in range(self.n_local)])
+        self.inter = inter
+        self.gate_up_packed: List[Optional[torch.Tensor]] = [None] * self.n_local
+        self.down_packed: List[Optional[torch.Tensor]] = [None] * self.n_local
+
+    def forward(self, h: torch.Tensor, ep_group=None) -> torch.Tensor:
+        T = h.size(0)
+        logits = F.linear(h, self.router).float()           # [T, E]
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)         # [T, k]
+        topv = topv / topv.sum(dim=-1, keepdim=True)        # renormalize
+        # dense top-k mask: weight[t, e] (0 if e not in token t's top-k)
+        weight = torch.zeros_like(probs)
+        weight.scatter_(1, topi, topv)
+        out = torch.zeros_like(h, dtype=torch.float32)
+        for i in range(self.n_local):
+            e = self.e0 + i
+            gu = ops.linear(h, self.gate_up[i], self.gate_up_packed[i])
+            gate, up = gu[:, :self.inter], gu[:, self.inter:]
+            act = torch.empty(T, self.inter, dtype=gu.dtype, device=gu.device)
+            ops.silu_mul(act, gate, up)
+            eo = ops.linear(act, self.down[i], self.down_packed[i])
+            out += weight[:, e].unsqueeze(1) * eo.float()
+        out = out.to(h.dtype)
+        if ep_group is not None:
+            torch.distributed.all_reduce(out, group=ep_group)
+        return out
+
+
+class MixtralLayer(torch.nn.Module):
+    def __init__(self, cfg: MixtralConfig, tp_rank: int = 0, tp_size: int = 1):
+        super().__init__()
+        # attention is TP-sharded like Llama; experts are EP-partitioned
+        # over the same ranks
+        self.attn = LlamaAttention(cfg, tp_rank, tp_size)
+        self.moe = MixtralMoE(cfg, ep_rank=tp_rank, ep_size=tp_size)
+        self.input_ln = torch.nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+        self.post_ln = torch.nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+        self.eps = cfg.norm_eps
+
+    def forward(self, hidden, residual, k_cache, v_cache, md, cos_sin,
+                tp_group=None):
+        normed = torch.empty_like(hidden)
+        if residual is None:
+            residual = hidden.clone()
+            ops.rmsnorm(normed, hidden, self.input_ln, self.eps)
+        else:
+            ops.fused_add_rmsnorm(normed, hidden, residual, self.input_ln, self.eps)
+        attn_out = self.attn(normed, k_cache, v_cache, md, cos_sin, tp_group)
+        normed2 = torch.empty_like(hidden)
+        ops.fused_add_rmsnorm(normed2, attn_out, residual, self.post_ln, self.eps)
+        moe_out = self.moe(normed2, ep_group=tp_group)
+        return moe_out, residual
+
+
+class MixtralForCausalLM(torch.nn.Module):
+    def __init__(self, cfg: MixtralConfig, device="cpu", tp_rank: int = 0,
+                 tp_size: int = 1, seed: int = 0):
+        super().__init__()
+        self.cfg = cfg
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.tp_group = None
+        with torch.device(device):
+            self.embed = torch.nn.Parameter(
+                torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+            self.layers = torch.nn.ModuleList(
+                [MixtralLayer(cfg, tp_rank, tp_size) for _ in range(cfg.n_layers)])
+            self.final_ln = torch.nn.Parameter(
+                torch.empty(cfg.hidden_size, dtype=torch.bfloat16))
+            self.lm_head = torch.nn.Parameter(
+                torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16))
+        self.register_buffer(
+            "cos_sin",
+            ops.make_cos_sin_table(cfg.max_position, cfg.head_dim,
+                                   cfg.rope_theta, device=device),
+            persistent=False)
+        self.random_init(seed)
+
+    @torch.no_grad()
+    def random_init(self, seed: int = 0):
+        dev = self.embed.device
+        gen = torch.Generator(device=dev).manual_seed(seed)
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                p.data.normal_(0.0, 0.02, generator=gen)
+            else:
+                p.fill_(1.0)
+
+    @torch.no_grad()
+    def pack_decode_weights(self):
+        for layer in self.layers:
+            layer.attn.qkv_packed = ops.pack_weight(layer.attn.qkv_proj.data)
+            layer.attn.o_packed = ops.pack_weight(layer.attn.o_proj.data)
+            for i in range(layer.moe.n_local):
+                layer.moe.gate_up_packed[i] = ops.pack_weight(
+                    layer.moe.gate_up[i].data)
+                layer.moe.down_packed[i] = ops.pack_weight(layer.moe.down[i].data)
+        self.lm_head_packed = ops.pack_weight(self.lm_head.data)
+
+    @torch.no_grad()
+    def forward(self, input_ids, md: AttnMetadata, kv_caches,
+                last_rows=None) -> torch.Tensor:
+        hidden = F.embedding(input_ids, self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            k_cache, v_cache = kv_caches[i]
+            hidden, residual = layer(hidden, residual, k_cache, v_cache, md,
+                                     self.cos_sin, self.tp_group)
+        final = torch.empty_like(hidden)
+        ops.fused_add_rmsnorm(final, hidden, residual, self.final_ln,
+                              self.cfg.norm_eps)
+        if last_rows is not None:
+            final = final[last_rows]
+        return ops.linear(final, self.lm_head,
+                          getattr(self, "lm_head_packed", None))
+
+
+MIXTRAL_CONFIGS: Dict[str, MixtralConfig] = {
+    "tiny-mixtral": MixtralConfig(
+        name="tiny-mixtral", vocab_size=512, hidden_size=512, n_layers=2,
+        n_heads=4, n_kv_heads=1, intermediate_size=512, max_position=4096,
+        n_experts=4, top_k=2, tie_embeddings=False),
+    "mixtral-8x7b": MixtralConfig(
+        name="mixtral-8x7b", vocab_size=32064, hidden_size=4096, n_layers=32,
+        n_heads=32, n_kv_heads=8, intermediate_size=14336, max_position=8192,
+        rope_theta=1e6, n_experts=8, top_k=2, tie_embeddings=False),
+}
+
+
+def make_mixtral_instance(name: str, device: str, engine_cfg):
+    """ModelInstance over a Mixtral model (engine factory hook)."""
+    from ..engine.llm import ModelInstance
+
+    return ModelInstance(name, MIXTRAL_CONFIGS[name], device, engine_cfg,
+                         model_cls=MixtralForCausalLM)

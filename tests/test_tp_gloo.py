@@ -1,0 +1,224 @@
+"""Tensor-parallel engine over gloo (world_size 2, CPU).
+
+Covers the distributed path the driver's 8-GPU round-end run exercises
+(SURVEY.md §4 implication (4)): TP sharding math, the SPMD worker loop
+(plan broadcast + deterministic page allocation), and TP KV
+checkpoint/restore — all with the same code that runs RCCL on MI355X.
+
+Correctness oracle: a FULL (tp=1) tiny model whose weights are sharded
+by hand onto the two ranks; greedy generation must match token-for-token.
+"""
+
+import os
+import socket
+import tempfile
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+TP_MODEL = "tiny-llama-tp"
+PROMPT = list(range(3, 43))
+MAX_NEW = 6
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _shard_weights(full_sd, model, rank, size):
+    """Slice a full model state dict onto one TP rank's model."""
+    import re
+
+    cfg = model.cfg
+    qh = cfg.n_heads // size
+    kvh = cfg.n_kv_heads // size
+    hd = cfg.head_dim
+    inter = cfg.intermediate_size // size
+    with torch.no_grad():
+        model.embed.copy_(full_sd["embed"])
+        model.final_ln.copy_(full_sd["final_ln"])
+        if not cfg.tie_embeddings:
+            model.lm_head.copy_(full_sd["lm_head"])
+        for i, layer in enumerate(model.layers):
+            pfx = f"layers.{i}."
+            fq = full_sd[pfx + "attn.qkv_proj"]
+            q_full, kv_full = cfg.n_heads * hd, cfg.n_kv_heads * hd
+            q = fq[rank * qh * hd:(rank + 1) * qh * hd]
+            k = fq[q_full + rank * kvh * hd: q_full + (rank + 1) * kvh * hd]
+            v = fq[q_full + kv_full + rank * kvh * hd:
+                   q_full + kv_full + (rank + 1) * kvh * hd]
+            layer.attn.qkv_proj.copy_(torch.cat([q, k, v]))
+            layer.attn.o_proj.copy_(
+                full_sd[pfx + "attn.o_proj"][:, rank * qh * hd:(rank + 1) * qh * hd])
+            fgu = full_sd[pfx + "mlp.gate_up"]
+            fi = cfg.intermediate_size
+            gate = fgu[rank * inter:(rank + 1) * inter]
+            up = fgu[fi + rank * inter: fi + (rank + 1) * inter]
+            layer.mlp.gate_up.copy_(torch.cat([gate, up]))
+            layer.mlp.down.copy_(
+                full_sd[pfx + "mlp.down"][:, rank * inter:(rank + 1) * inter])
+            layer.input_ln.copy_(full_sd[pfx + "input_ln"])
+            layer.post_ln.copy_(full_sd[pfx + "post_ln"])
+
+
+def _reference_tokens(full_sd, tmpdir):
+    """Greedy generation on the full (tp=1) model."""
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import GenRequest, LLMEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    cfg.data["engine"]["sync_mode"] = True
+    cfg.data["engine"]["kv_pool_gb"] = 0.02
+    store = Store(os.path.join(tmpdir, "ref-state"), sync="never")
+    eng = LLMEngine(store, cfg, device="cpu", state_root=tmpdir + "/ref")
+    man = Manager(store, eng, cfg)
+    a = man.deploy(name="ref", model=TP_MODEL, sampling={"max_tokens": MAX_NEW})
+    man.start(a.id)
+    inst = eng._instances[TP_MODEL]
+    inst.model.load_state_dict(
+        {k: v for k, v in full_sd.items()}, strict=False)
+    # state dict keys: convert plain names used in _shard_weights
+    with torch.no_grad():
+        inst.model.embed.copy_(full_sd["embed"])
+        inst.model.final_ln.copy_(full_sd["final_ln"])
+        for i, layer in enumerate(inst.model.layers):
+            pfx = f"layers.{i}."
+            layer.attn.qkv_proj.copy_(full_sd[pfx + "attn.qkv_proj"])
+            layer.attn.o_proj.copy_(full_sd[pfx + "attn.o_proj"])
+            layer.mlp.gate_up.copy_(full_sd[pfx + "mlp.gate_up"])
+            layer.mlp.down.copy_(full_sd[pfx + "mlp.down"])
+            layer.input_ln.copy_(full_sd[pfx + "input_ln"])
+            layer.post_ln.copy_(full_sd[pfx + "post_ln"])
+    req = GenRequest(agent_id=a.id, prompt_tokens=PROMPT, max_new=MAX_NEW,
+                     temperature=0.0, top_p=1.0, seed=0)
+    b = inst.binding(a.id)
+    with inst._lock:
+        b.queue.put(req)
+        inst._pump_agent(b)
+    for _ in range(MAX_NEW + 4):
+        inst.step()
+        if req.done.is_set():
+            break
+    assert req.done.is_set() and not req.error
+    eng.shutdown()
+    store.close()
+    return req.generated
+
+
+def _tp_worker(rank, world, port, tmpdir, result_file):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+
+    from agentainer_amd import parallel as par
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import GenRequest, LLMEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    par.init_distributed(backend="gloo")
+    full_sd = torch.load(os.path.join(tmpdir, "full_sd.pt"), weights_only=True)
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    cfg.data["engine"]["sync_mode"] = True
+    cfg.data["engine"]["kv_pool_gb"] = 0.02
+    cfg.data["engine"]["tp_degree"] = world
+    store = Store(os.path.join(tmpdir, f"state-{rank}"), sync="never")
+    eng = LLMEngine(store, cfg, device="cpu", state_root=f"{tmpdir}/r{rank}")
+    if rank != 0:
+        # build the instance when rank0 broadcasts, then serve plans;
+        # shard weights when the instance appears
+        import threading
+
+        def patch_weights():
+            # run_worker creates the instance on the "instance" cmd; patch
+            # by wrapping _make_instance
+            orig = eng._make_instance
+            def wrapped(model):
+                inst = orig(model)
+                _shard_weights(full_sd, inst.model, rank, world)
+                return inst
+            eng._make_instance = wrapped
+        patch_weights()
+        eng.run_worker()
+        dist.barrier()
+        return
+    man = Manager(store, eng, cfg)
+    a = man.deploy(name="tp", model=TP_MODEL, sampling={"max_tokens": MAX_NEW})
+    man.start(a.id)
+    inst = eng._instances[TP_MODEL]
+    _shard_weights(full_sd, inst.model, 0, world)
+    req = GenRequest(agent_id=a.id, prompt_tokens=PROMPT, max_new=MAX_NEW,
+                     temperature=0.0, top_p=1.0, seed=0)
+    b = inst.binding(a.id)
+    with inst._lock:
+        b.queue.put(req)
+        inst._pump_agent(b)
+    for _ in range(MAX_NEW + 4):
+        inst.step()
+        if req.done.is_set():
+            break
+    # KV checkpoint roundtrip under TP: stop + resume, then one more turn
+    man.stop(a.id)
+    man.resume(a.id)
+    req2 = GenRequest(agent_id=a.id, prompt_tokens=list(range(50, 70)),
+                      max_new=MAX_NEW, temperature=0.0, top_p=1.0, seed=0)
+    b = inst.binding(a.id)
+    with inst._lock:
+        b.queue.put(req2)
+        inst._pump_agent(b)
+    for _ in range(MAX_NEW + 4):
+        inst.step()
+        if req2.done.is_set():
+            break
+    torch.save({"tokens": req.generated, "err": req.error,
+                "tokens2": req2.generated, "err2": req2.error,
+                "len_after": inst.kvm.seq_len(a.id)}, result_file)
+    eng.shutdown()  # broadcasts shutdown to the worker
+    dist.barrier()
+
+
+@pytest.mark.timeout(300)
+def test_tp2_matches_full_model(tmp_path):
+    from agentainer_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+
+    tmpdir = str(tmp_path)
+    # build the full model once; save a plain-named state dict
+    full = LlamaForCausalLM(LLAMA_CONFIGS[TP_MODEL], device="cpu", seed=3)
+    sd = {"embed": full.embed.data, "final_ln": full.final_ln.data}
+    for i, layer in enumerate(full.layers):
+        pfx = f"layers.{i}."
+        sd[pfx + "attn.qkv_proj"] = layer.attn.qkv_proj.data
+        sd[pfx + "attn.o_proj"] = layer.attn.o_proj.data
+        sd[pfx + "mlp.gate_up"] = layer.mlp.gate_up.data
+        sd[pfx + "mlp.down"] = layer.mlp.down.data
+        sd[pfx + "input_ln"] = layer.input_ln.data
+        sd[pfx + "post_ln"] = layer.post_ln.data
+    torch.save(sd, os.path.join(tmpdir, "full_sd.pt"))
+
+    want = _reference_tokens(sd, tmpdir)
+    assert len(want) == MAX_NEW
+
+    result_file = os.path.join(tmpdir, "result.pt")
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_tp_worker, args=(r, 2, port, tmpdir, result_file))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0, f"worker exit {p.exitcode}"
+    res = torch.load(result_file, weights_only=True)
+    assert res["err"] is None and res["err2"] is None
+    assert res["tokens"] == want, f"TP tokens {res['tokens']} != full {want}"
+    assert len(res["tokens2"]) == MAX_NEW  # post-restore turn completed
+    assert res["len_after"] > len(PROMPT) + MAX_NEW  # KV restored + extended
