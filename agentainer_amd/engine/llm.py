@@ -124,6 +124,13 @@ class ModelInstance:
         self.prefill_tokens = 0
         self.occupancy_acc = 0.0
         self.sync_mode = bool(engine_cfg.get("sync_mode", False))
+        # engine watchdog (SURVEY.md §5 failure detection): a step that has
+        # been in flight longer than this marks the instance unhealthy
+        # (stuck kernel / GPU fell off the bus) so the health monitor's
+        # auto-restart path fires
+        self.watchdog_timeout_s = float(engine_cfg.get("watchdog_timeout_s", 120.0))
+        self._step_started: Optional[float] = None
+        self.last_step_t = time.time()
         # hipGraph-captured decode step (SURVEY.md north star): the whole
         # batched decode forward — page-table gather, slot computation,
         # RoPE/append/attention/GEMMs, device-side length increment — is
@@ -174,9 +181,16 @@ class ModelInstance:
             self._thread = None
 
     def alive(self) -> bool:
+        if self.stuck():
+            return False
         if self.sync_mode:
             return True
         return self._thread is not None and self._thread.is_alive()
+
+    def stuck(self) -> bool:
+        """Watchdog verdict: one engine step exceeding the timeout."""
+        st = self._step_started
+        return st is not None and (time.time() - st) > self.watchdog_timeout_s
 
     def _loop(self):
         while not self._stop.is_set():
@@ -200,6 +214,7 @@ class ModelInstance:
 
     def step(self) -> bool:
         """Admit + prefill, then one decode step. Returns True if work ran."""
+        self._step_started = time.time()
         admitted = self._admit()
         if admitted:
             self._prefill(admitted)
@@ -210,6 +225,8 @@ class ModelInstance:
             self._decode(batch[: self.max_decode_batch])
             ran_decode = True
         self.steps += 1 if (admitted or ran_decode) else 0
+        self._step_started = None
+        self.last_step_t = time.time()
         return bool(admitted) or ran_decode
 
     def _admit(self) -> List[GenRequest]:
@@ -812,9 +829,18 @@ class LLMEngine:
                 "kv_pages_used": inst.kvm.used_pages,
                 "kv_pages_free": inst.kvm.free_pages,
                 "running": len(inst.running),
+                "stuck": inst.stuck(),
+                "last_step_age_s": round(time.time() - inst.last_step_t, 3),
             }
-        return {"engine": "llm", "device": self.device, "agents": agents,
-                "models": models}
+        hbm = {}
+        if self.device.startswith("cuda") and torch.cuda.is_available():
+            free, total = torch.cuda.mem_get_info()
+            hbm = {"hbm_total_bytes": total, "hbm_free_bytes": free,
+                   "hbm_torch_allocated": torch.cuda.memory_allocated()}
+        out = {"engine": "llm", "device": self.device, "agents": agents,
+               "models": models}
+        out.update(hbm)
+        return out
 
     def shutdown(self):
         if self.tp_size > 1 and self.tp_rank == 0 and par.is_tp():

@@ -115,3 +115,35 @@ def test_metrics_history_trim(runtime):
     runtime.metrics.sample_agent(a.id, {"tokens": 2, "requests": 2}, now=time.time())
     hist = runtime.metrics.get_metrics_history(a.id, duration_s=48 * 3600)
     assert len(hist) == 1  # 24h retention trim
+
+
+# ---------- engine watchdog ----------
+
+def test_watchdog_marks_stuck_engine_unhealthy(tmp_path):
+    """A step stuck past watchdog_timeout_s fails the health probe, and
+    the health monitor's retry counter picks it up (stuck-kernel path)."""
+    import time as _time
+
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import LLMEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    cfg.data["engine"]["sync_mode"] = True
+    cfg.data["engine"]["kv_pool_gb"] = 0.01
+    cfg.data["engine"]["watchdog_timeout_s"] = 0.05
+    store = Store(str(tmp_path / "state"), sync="never")
+    eng = LLMEngine(store, cfg, device="cpu", state_root=str(tmp_path))
+    man = Manager(store, eng, cfg)
+    a = man.deploy(name="w", model="tiny-llama")
+    man.start(a.id)
+    assert eng.health_probe(a.id) is True
+    inst = eng._instances["tiny-llama"]
+    inst._step_started = _time.time() - 1.0  # simulate a wedged step
+    assert inst.stuck() is True
+    assert eng.health_probe(a.id) is False
+    assert eng.stats()["models"]["tiny-llama"]["stuck"] is True
+    inst._step_started = None
+    assert eng.health_probe(a.id) is True
+    store.close()
