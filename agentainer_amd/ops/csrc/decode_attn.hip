@@ -70,17 +70,23 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
     return;
   }
 
-  __shared__ float q_lds[RATIO][D];
   __shared__ float p_lds[RATIO][CHUNK];
   const int qh0 = g * RATIO;
+  // score-phase lane split: ds = lane/16 owns a 32-dim slice, t16 = lane%16
+  // walks tokens. q lives in REGISTERS as bf16 pairs and the dots run on
+  // v_dot2_f32_bf16 — the previous LDS-broadcast-per-element formulation
+  // was instruction-issue-bound (512 ds_read + 512 fma per token).
+  const int ds = lane / 16;
+  const int t16 = lane % 16;
+  typedef __bf16 bf2 __attribute__((ext_vector_type(2)));
+  bf16x8 qv[RATIO][4];
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
-    const short* qp = q + (long)b * q_ts + (long)(qh0 + h) * D;
+    const short* qp = q + (long)b * q_ts + (long)(qh0 + h) * D + ds * 32;
 #pragma unroll
-    for (int r = 0; r < D / WAVE; ++r)
-      q_lds[h][r * WAVE + lane] = bits2f(qp[r * WAVE + lane]);
+    for (int r = 0; r < 4; ++r)
+      qv[h][r] = *reinterpret_cast<const bf16x8*>(qp + r * 8);
   }
-  __builtin_amdgcn_s_waitcnt(0);
 
   float m[RATIO], l[RATIO], o0[RATIO], o1[RATIO];
 #pragma unroll
@@ -91,44 +97,66 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
   const int d0 = lane * 2;
 
   for (int c = c0; c < c1; ++c) {
-    const int tok = c * CHUNK + lane;
-    float s[RATIO];
+    float s[4][RATIO];  // [16-token sub-pass][head]
 #pragma unroll
-    for (int h = 0; h < RATIO; ++h) s[h] = -FLT_MAX;
-    if (tok < len) {
-      const long page = pt[tok / PS];
-      const short* kp = k_cache + (((long)page * n_kv + g) * D8 * PS + tok % PS) * 8;
+    for (int sub = 0; sub < 4; ++sub) {
+      const int tok = c * CHUNK + sub * 16 + t16;
       float acc[RATIO];
 #pragma unroll
       for (int h = 0; h < RATIO; ++h) acc[h] = 0.f;
+      if (tok < len) {
+        const long page = pt[tok / PS];
+        // this lane's 4 d8 groups: d8 = ds*4 + r
+        const short* kp = k_cache +
+            (((long)page * n_kv + g) * D8 + ds * 4) * PS * 8 + (tok % PS) * 8;
+        bf16x8 kv[4];
 #pragma unroll
-      for (int d8 = 0; d8 < D8; ++d8) {
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(kp + (long)d8 * PS * 8);
+        for (int r = 0; r < 4; ++r)
+          kv[r] = *reinterpret_cast<const bf16x8*>(kp + (long)r * PS * 8);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float kf = bits2f(kv8[j]);
+        for (int r = 0; r < 4; ++r) {
+          const bf2* kp2 = reinterpret_cast<const bf2*>(&kv[r]);
 #pragma unroll
-          for (int h = 0; h < RATIO; ++h)
-            acc[h] = fmaf(kf, q_lds[h][d8 * 8 + j], acc[h]);
+          for (int j = 0; j < 4; ++j) {
+#pragma unroll
+            for (int h = 0; h < RATIO; ++h) {
+              const bf2* qp2 = reinterpret_cast<const bf2*>(&qv[h][r]);
+              acc[h] = __builtin_amdgcn_fdot2_f32_bf16(kp2[j], qp2[j],
+                                                       acc[h], false);
+            }
+          }
         }
       }
+      // reduce the 4 dim-slices (lanes differing in bits 4..5)
 #pragma unroll
-      for (int h = 0; h < RATIO; ++h) s[h] = acc[h] * scale;
+      for (int h = 0; h < RATIO; ++h) {
+        acc[h] += __shfl_xor(acc[h], 16, WAVE);
+        acc[h] += __shfl_xor(acc[h], 32, WAVE);
+        s[sub][h] = (tok < len) ? acc[h] * scale : -FLT_MAX;
+      }
     }
-    // per-head online softmax update
+    // per-head online softmax update over the whole 64-token chunk
     const int c_len = min(CHUNK, len - c * CHUNK);
 #pragma unroll
     for (int h = 0; h < RATIO; ++h) {
-      const float cmax = wave_max(s[h]);
-      const float mn = fmaxf(m[h], cmax);
-      const float p = (tok < len) ? __expf(s[h] - mn) : 0.f;
-      const float csum = wave_sum(p);
+      float cm = fmaxf(fmaxf(s[0][h], s[1][h]), fmaxf(s[2][h], s[3][h]));
+      cm = wave_max(cm);
+      const float mn = fmaxf(m[h], cm);
+      float psum = 0.f;
+#pragma unroll
+      for (int sub = 0; sub < 4; ++sub) {
+        const float p = (s[sub][h] == -FLT_MAX) ? 0.f : __expf(s[sub][h] - mn);
+        if (ds == 0) {
+          p_lds[h][sub * 16 + t16] = p;
+          psum += p;
+        }
+      }
+      const float csum = wave_sum(psum);
       const float alpha = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
       l[h] = l[h] * alpha + csum;
       o0[h] *= alpha;
       o1[h] *= alpha;
       m[h] = mn;
-      p_lds[h][lane] = p;
     }
     __builtin_amdgcn_s_waitcnt(0);
 
