@@ -79,13 +79,16 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
   const int ds = lane / 16;
   const int t16 = lane % 16;
   typedef __bf16 bf2 __attribute__((ext_vector_type(2)));
+  // lane (ds, t16) owns dims {(4r+ds)*8 .. +8} for r in 0..3: for each r
+  // the wave's 64 lanes cover 4 CONSECUTIVE d8 groups x 16 consecutive
+  // tokens = one contiguous 1 KB run per page (full coalescing)
   bf16x8 qv[RATIO][4];
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
-    const short* qp = q + (long)b * q_ts + (long)(qh0 + h) * D + ds * 32;
+    const short* qp = q + (long)b * q_ts + (long)(qh0 + h) * D;
 #pragma unroll
     for (int r = 0; r < 4; ++r)
-      qv[h][r] = *reinterpret_cast<const bf16x8*>(qp + r * 8);
+      qv[h][r] = *reinterpret_cast<const bf16x8*>(qp + (r * 4 + ds) * 8);
   }
 
   float m[RATIO], l[RATIO], o0[RATIO], o1[RATIO];
@@ -106,13 +109,14 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       for (int h = 0; h < RATIO; ++h) acc[h] = 0.f;
       if (tok < len) {
         const long page = pt[tok / PS];
-        // this lane's 4 d8 groups: d8 = ds*4 + r
+        // this lane's 4 d8 groups: d8 = r*4 + ds (coalesced per r)
         const short* kp = k_cache +
-            (((long)page * n_kv + g) * D8 + ds * 4) * PS * 8 + (tok % PS) * 8;
+            ((long)page * n_kv + g) * D8 * PS * 8 + (tok % PS) * 8;
         bf16x8 kv[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          kv[r] = *reinterpret_cast<const bf16x8*>(kp + (long)r * PS * 8);
+          kv[r] = *reinterpret_cast<const bf16x8*>(
+              kp + (long)(r * 4 + ds) * PS * 8);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const bf2* kp2 = reinterpret_cast<const bf2*>(&kv[r]);
