@@ -101,7 +101,13 @@ class ModelInstance:
         if weights_path:
             self.model.load_safetensors(weights_path)
         if device.startswith("cuda"):
-            self.model.pack_decode_weights()
+            import inspect
+            if "expert_fp8" in inspect.signature(
+                    self.model.pack_decode_weights).parameters:
+                self.model.pack_decode_weights(
+                    expert_fp8=bool(engine_cfg.get("expert_fp8", False)))
+            else:
+                self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))
         n_pages = self._pool_pages(cfg, page_size, device, engine_cfg,
                                    self.tp_size)

@@ -96,8 +96,8 @@ class LlamaAttention(torch.nn.Module):
         q = qkv[:, :q_sz].view(T, self.n_heads, self.head_dim)
         k = qkv[:, q_sz:q_sz + kv_sz].view(T, self.n_kv, self.head_dim)
         v = qkv[:, q_sz + kv_sz:].view(T, self.n_kv, self.head_dim)
-        ops.rope_inplace(q, k, cos_sin, md.positions)
-        ops.kv_append(k_cache, v_cache, k, v, md.slot_mapping)
+        ops.rope_append(q, k, v, k_cache, v_cache, cos_sin, md.positions,
+                        md.slot_mapping)
         out = torch.empty(T, self.n_heads, self.head_dim, dtype=q.dtype,
                           device=q.device)
         if md.is_prefill:

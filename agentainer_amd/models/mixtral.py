@@ -65,6 +65,9 @@ class MixtralMoE(torch.nn.Module):
         self.inter = inter
         self.gate_up_packed: List[Optional[torch.Tensor]] = [None] * self.n_local
         self.down_packed: List[Optional[torch.Tensor]] = [None] * self.n_local
+        # fp8 expert path (config 5): (packed bytes, scales) per expert
+        self.gate_up_fp8: List[Optional[tuple]] = [None] * self.n_local
+        self.down_fp8: List[Optional[tuple]] = [None] * self.n_local
 
     def forward(self, h: torch.Tensor, ep_group=None) -> torch.Tensor:
         T = h.size(0)
@@ -76,13 +79,22 @@ class MixtralMoE(torch.nn.Module):
         weight = torch.zeros_like(probs)
         weight.scatter_(1, topi, topv)
         out = torch.zeros_like(h, dtype=torch.float32)
+        use_fp8 = (h.is_cuda and T <= 64 and self.gate_up_fp8[0] is not None)
         for i in range(self.n_local):
             e = self.e0 + i
-            gu = ops.linear(h, self.gate_up[i], self.gate_up_packed[i])
+            if use_fp8:
+                wp, sw = self.gate_up_fp8[i]
+                gu = ops.linear_fp8(h, wp, sw, 2 * self.inter)
+            else:
+                gu = ops.linear(h, self.gate_up[i], self.gate_up_packed[i])
             gate, up = gu[:, :self.inter], gu[:, self.inter:]
             act = torch.empty(T, self.inter, dtype=gu.dtype, device=gu.device)
             ops.silu_mul(act, gate, up)
-            eo = ops.linear(act, self.down[i], self.down_packed[i])
+            if use_fp8:
+                wp, sw = self.down_fp8[i]
+                eo = ops.linear_fp8(act, wp, sw, h.size(1))
+            else:
+                eo = ops.linear(act, self.down[i], self.down_packed[i])
             out += weight[:, e].unsqueeze(1) * eo.float()
         out = out.to(h.dtype)
         if ep_group is not None:
@@ -153,14 +165,21 @@ class MixtralForCausalLM(torch.nn.Module):
                 p.fill_(1.0)
 
     @torch.no_grad()
-    def pack_decode_weights(self):
+    def pack_decode_weights(self, expert_fp8: bool = False):
         for layer in self.layers:
             layer.attn.qkv_packed = ops.pack_weight(layer.attn.qkv_proj.data)
             layer.attn.o_packed = ops.pack_weight(layer.attn.o_proj.data)
             for i in range(layer.moe.n_local):
-                layer.moe.gate_up_packed[i] = ops.pack_weight(
-                    layer.moe.gate_up[i].data)
-                layer.moe.down_packed[i] = ops.pack_weight(layer.moe.down[i].data)
+                if expert_fp8:
+                    layer.moe.gate_up_fp8[i] = ops.quantize_weight_fp8(
+                        layer.moe.gate_up[i].data)
+                    layer.moe.down_fp8[i] = ops.quantize_weight_fp8(
+                        layer.moe.down[i].data)
+                else:
+                    layer.moe.gate_up_packed[i] = ops.pack_weight(
+                        layer.moe.gate_up[i].data)
+                    layer.moe.down_packed[i] = ops.pack_weight(
+                        layer.moe.down[i].data)
         self.lm_head_packed = ops.pack_weight(self.lm_head.data)
 
     @torch.no_grad()

@@ -98,6 +98,18 @@ def rope_inplace(q, k, cos_sin, positions):
         reference.rope_inplace(q, k, cos_sin, positions)
 
 
+def rope_append(q, k, v, k_cache, v_cache, cos_sin, positions, slot_mapping):
+    """Fused rope_inplace(q,k) + kv_append(k,v). CPU falls back to the two
+    reference ops."""
+    mod = _dispatch("rope_append", q)
+    if mod:
+        mod.rope_append(q, k, v, k_cache, v_cache, cos_sin, positions,
+                        slot_mapping)
+    else:
+        reference.rope_inplace(q, k, cos_sin, positions)
+        reference.kv_append(k_cache, v_cache, k, v, slot_mapping)
+
+
 def kv_append(k_cache, v_cache, k, v, slot_mapping):
     mod = _dispatch("kv_append", k)
     if mod:
@@ -215,6 +227,40 @@ def linear(x: torch.Tensor, w: torch.Tensor,
             mod.skinny_gemm(out, x.contiguous(), w, ws, split)
         return out
     return torch.nn.functional.linear(x, w)
+
+
+def quantize_weight_fp8(w: torch.Tensor):
+    """Offline per-output-channel OCP e4m3 quantization of a [N, K] weight:
+    returns (fragment-linear packed bytes, f32 scales[N]). Serves the fp8
+    MFMA expert decode GEMM (BASELINE config 5)."""
+    N, K = w.shape
+    sw = w.float().abs().amax(dim=1).clamp(min=1e-8) / 448.0
+    scaled = (w.float() / sw[:, None]).clamp(-448.0, 448.0)
+    try:
+        w8 = scaled.to(torch.float8_e4m3fn).view(torch.uint8)
+    except (RuntimeError, TypeError):
+        w8 = scaled.cpu().to(torch.float8_e4m3fn).view(torch.uint8).to(w.device)
+    packed = (w8.view(N // 16, 16, K // 32, 4, 8)
+              .permute(0, 2, 3, 1, 4).contiguous().view(-1))
+    return packed, sw.float().contiguous()
+
+
+def linear_fp8(x: torch.Tensor, w_packed: torch.Tensor, sw: torch.Tensor,
+               N: int) -> torch.Tensor:
+    """x[M,K] bf16 @ fp8-quantized W^T: dynamic per-token activation quant
+    then the fp8 MFMA skinny GEMM (half the weight bytes of bf16)."""
+    M, K = x.shape
+    mod = _dispatch("skinny_gemm_fp8", x)
+    assert mod is not None, "linear_fp8 is GPU-only"
+    xc = x.contiguous()
+    x8 = torch.empty(M, K, dtype=torch.uint8, device=x.device)
+    sx = torch.empty(M, dtype=torch.float32, device=x.device)
+    mod.quant_fp8_rows(x8, sx, xc)
+    split = _skinny_split(N // 64, K)
+    ws = _skinny_ws(x.device, N, split)
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    mod.skinny_gemm_fp8(out, x8, sx, w_packed, sw, N, K, ws, split)
+    return out
 
 
 def gather_kv_pages(dst, k_cache, v_cache, page_ids):

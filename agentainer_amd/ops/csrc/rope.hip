@@ -60,3 +60,83 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
                      cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
                      n_q, n_kv, D, (long)q.stride(0), (long)k.stride(0));
 }
+
+// ---------------------------------------------------------------------------
+// Fused RoPE + KV-append: rotates q in place, rotates k in registers and
+// scatters k/v straight into the paged pools — one kernel instead of two,
+// and k never takes the rotate-write-reload round trip through HBM.
+// grid: (T, n_q + n_kv); blocks for h < n_q rotate q rows; blocks for
+// kv heads rotate k and append k+v.
+
+namespace {
+
+__global__ void rope_append_kernel(
+    short* __restrict__ q, short* __restrict__ k, const short* __restrict__ v,
+    short* __restrict__ k_cache, short* __restrict__ v_cache,
+    const float* __restrict__ cos_sin, const int* __restrict__ positions,
+    const long* __restrict__ slots, int n_q, int n_kv, int D, int PS,
+    long qs, long ks, long vs) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const int i = threadIdx.x;  // pair index < D/2
+  const int half = D / 2;
+  if (i >= half) return;
+  const int pos = positions[t];
+  const float c = cos_sin[(long)pos * D + i];
+  const float s = cos_sin[(long)pos * D + half + i];
+  if (h < n_q) {
+    short* base = q + (long)t * qs + (long)h * D;
+    const float x1 = bits2f(base[i]);
+    const float x2 = bits2f(base[half + i]);
+    base[i] = f2bits(x1 * c - x2 * s);
+    base[half + i] = f2bits(x1 * s + x2 * c);
+    return;
+  }
+  const int g = h - n_q;
+  const long slot = slots[t];
+  const short* kb = k + (long)t * ks + (long)g * D;
+  const float x1 = bits2f(kb[i]);
+  const float x2 = bits2f(kb[half + i]);
+  const short r1 = f2bits(x1 * c - x2 * s);
+  const short r2 = f2bits(x1 * s + x2 * c);
+  if (slot < 0) return;
+  const long page = slot / PS;
+  const int off = (int)(slot % PS);
+  const int D8 = D / 8;
+  // k_cache[page][g][d8][off][j]
+  long kbase = (((page * n_kv + g) * D8) * PS + off) * 8;
+  k_cache[kbase + (i / 8) * PS * 8 + i % 8] = r1;
+  k_cache[kbase + ((half + i) / 8) * PS * 8 + (half + i) % 8] = r2;
+  // v rows are not rotated: copy 2 elements per lane
+  const short* vb = v + (long)t * vs + (long)g * D;
+  long vbase = (((page * n_kv + g) * PS) + off) * D;
+  v_cache[vbase + i] = vb[i];
+  v_cache[vbase + half + i] = vb[half + i];
+}
+
+}  // namespace
+
+void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                 torch::Tensor k_cache, torch::Tensor v_cache,
+                 torch::Tensor cos_sin, torch::Tensor positions,
+                 torch::Tensor slot_mapping) {
+  TORCH_CHECK(cos_sin.is_contiguous() && cos_sin.scalar_type() == at::kFloat);
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(positions.scalar_type() == at::kInt);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  const int T = q.size(0), n_q = q.size(1), D = q.size(2);
+  const int n_kv = k.size(1), PS = k_cache.size(3);
+  TORCH_CHECK(D % 16 == 0 && D / 2 <= 1024);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D);
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == D);
+  TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == D);
+  if (T == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rope_append_kernel, dim3(T, n_q + n_kv), dim3(D / 2), 0,
+                     stream, (short*)q.data_ptr(), (short*)k.data_ptr(),
+                     (const short*)v.data_ptr(), (short*)k_cache.data_ptr(),
+                     (short*)v_cache.data_ptr(), cos_sin.data_ptr<float>(),
+                     positions.data_ptr<int>(), slot_mapping.data_ptr<long>(),
+                     n_q, n_kv, D, PS, (long)q.stride(0), (long)k.stride(0),
+                     (long)v.stride(0));
+}

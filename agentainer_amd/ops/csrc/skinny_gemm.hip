@@ -338,3 +338,203 @@ void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
 #undef SGP_DISPATCH
 #undef SGP_LAUNCH
 }
+
+// ---------------------------------------------------------------------------
+// fp8 (OCP e4m3) skinny GEMM — the Mixtral expert decode GEMM of
+// BASELINE.json config 5. Same structure as the bf16 packed kernel, but
+// operands are 8 fp8 bytes per lane (v_mfma_f32_16x16x32_fp8_fp8, i64
+// fragments) so the expert weight stream is HALF the bytes of bf16.
+// Scales: per-output-channel weight scales sw[n] (offline), per-token
+// activation scales sx[m] (quant_fp8_rows, dynamic) — applied in the
+// epilogue: out = acc * sx[row] * sw[col].
+
+#include <hip/hip_fp8.h>
+
+namespace {
+
+// per-row dynamic quantization: x bf16 [M,K] -> x8 e4m3 [M,K] + sx f32 [M]
+__global__ __launch_bounds__(256) void quant_fp8_rows_kernel(
+    unsigned char* __restrict__ x8, float* __restrict__ sx,
+    const short* __restrict__ x, int K) {
+  const int row = blockIdx.x;
+  const short* xp = x + (long)row * K;
+  float mx = 0.f;
+  for (int i = threadIdx.x * 8; i < K; i += blockDim.x * 8) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(xp + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) mx = fmaxf(mx, fabsf(bits2f(v[j])));
+  }
+  mx = wave_max(mx);
+  __shared__ float red[4];
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x / 64] = mx;
+  __syncthreads();
+  mx = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+  const float scale = fmaxf(mx, 1e-8f) / 448.f;  // e4m3 max normal = 448
+  const float inv = 1.0f / scale;
+  if (threadIdx.x == 0) sx[row] = scale;
+  for (int i = threadIdx.x * 8; i < K; i += blockDim.x * 8) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(xp + i);
+    unsigned char o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_fp8_e4m3 f8(bits2f(v[j]) * inv);
+      o[j] = f8.__x;
+    }
+    *reinterpret_cast<uint2*>(x8 + (long)row * K + i) =
+        *reinterpret_cast<uint2*>(o);
+  }
+}
+
+template <bool SPLIT>
+__global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
+    void* __restrict__ out,                 // bf16 [M,N] or f32 ws
+    const unsigned char* __restrict__ x8,   // [M,K] e4m3
+    const float* __restrict__ sx,           // [M]
+    const unsigned char* __restrict__ wp_packed,  // fragment-linear e4m3
+    const float* __restrict__ sw,           // [N]
+    int M, int N, int K, int k_per_split) {
+  const int n_tile = blockIdx.x * 4 + threadIdx.x / WAVE;
+  const int split = blockIdx.y;
+  const int lane = threadIdx.x % WAVE;
+  const int l16 = lane % 16;
+  const int lg = lane / 16;
+  const int k0 = split * k_per_split;
+  const int k1 = min(k0 + k_per_split, K);
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) acc[ms] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const unsigned char* wp =
+      wp_packed + ((long)n_tile * (K / 32) + k0 / 32) * 512 + lane * 8;
+
+  constexpr int KC = 256;
+  constexpr int XS = KC + 16;  // bytes; keeps 8-B frag reads conflict-light
+  __shared__ unsigned char x_lds[2][64 * XS];
+  const int s_row = threadIdx.x % 64;
+  const int s_col0 = (threadIdx.x / 64) * 8;  // + 32*i, i in 0..7 (bytes)
+  const unsigned char* s_xp = x8 + (long)min(s_row, M - 1) * K;
+  const bool s_alive = s_row < M;
+
+  uint2 st[8];
+  auto stage_load = [&](int k) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      st[i] = uint2{0, 0};
+      if (s_alive)
+        st[i] = *reinterpret_cast<const uint2*>(s_xp + k + s_col0 + 32 * i);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      *reinterpret_cast<uint2*>(&x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
+  };
+
+  long bw_cur[8], bw_nxt[8];
+  auto w_load = [&](long (&dst)[8], long woff) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      dst[i] = __builtin_nontemporal_load(
+          reinterpret_cast<const long*>(wp + woff + (long)i * 512));
+  };
+
+  stage_load(k0);
+  stage_write(0);
+  w_load(bw_cur, 0);
+  __syncthreads();
+
+  int buf = 0;
+  long woff = 8 * 512;
+  for (int k = k0; k < k1; k += KC) {
+    const bool has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_nxt, woff);
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      long a[4];
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        a[ms] = *reinterpret_cast<const long*>(
+            &x_lds[buf][(ms * 16 + l16) * XS + i * 32 + lg * 8]);
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms)
+        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            a[ms], bw_cur[i], acc[ms], 0, 0, 0);
+    }
+    if (has_next) stage_write(buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+    woff += 8 * 512;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) bw_cur[i] = bw_nxt[i];
+  }
+
+  const int col = n_tile * 16 + l16;
+  const float swc = sw[col];
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = ms * 16 + lg * 4 + r;
+      if (row >= M) continue;
+      const float v = acc[ms][r] * sx[row] * swc;
+      if (SPLIT) {
+        reinterpret_cast<float*>(out)[((long)split * M + row) * N + col] = v;
+      } else {
+        reinterpret_cast<short*>(out)[(long)row * N + col] = f2bits(v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void quant_fp8_rows(torch::Tensor x8, torch::Tensor sx, torch::Tensor x) {
+  TORCH_CHECK(x.is_contiguous() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x8.scalar_type() == at::kByte && sx.scalar_type() == at::kFloat);
+  const int M = x.size(0), K = x.size(1);
+  TORCH_CHECK(K % 8 == 0);
+  if (M == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(quant_fp8_rows_kernel, dim3(M), dim3(256), 0, stream,
+                     x8.data_ptr<unsigned char>(), sx.data_ptr<float>(),
+                     (const short*)x.data_ptr(), K);
+}
+
+void skinny_gemm_fp8(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
+                     torch::Tensor w_packed, torch::Tensor sw, long N, long K,
+                     torch::Tensor ws, long split) {
+  TORCH_CHECK(x8.is_contiguous() && w_packed.is_contiguous());
+  TORCH_CHECK(x8.scalar_type() == at::kByte &&
+              w_packed.scalar_type() == at::kByte);
+  TORCH_CHECK(sw.scalar_type() == at::kFloat && sx.scalar_type() == at::kFloat);
+  const int M = x8.size(0);
+  TORCH_CHECK(x8.size(1) == K);
+  TORCH_CHECK(M <= 64 && N % 64 == 0 && K % 256 == 0);
+  TORCH_CHECK(split >= 1 && (K % (256 * split)) == 0);
+  auto stream = at::hip::getCurrentHIPStream();
+  const int ntiles = (int)N / 64;
+  const int kps = (int)K / (int)split;
+  if (split == 1) {
+    hipLaunchKernelGGL((skinny_gemm_fp8_kernel<false>), dim3(ntiles, 1),
+                       dim3(256), 0, stream, out.data_ptr(),
+                       x8.data_ptr<unsigned char>(), sx.data_ptr<float>(),
+                       w_packed.data_ptr<unsigned char>(), sw.data_ptr<float>(),
+                       M, (int)N, (int)K, kps);
+  } else {
+    TORCH_CHECK(ws.numel() >= (long)split * M * N &&
+                ws.scalar_type() == at::kFloat);
+    hipLaunchKernelGGL((skinny_gemm_fp8_kernel<true>), dim3(ntiles, (int)split),
+                       dim3(256), 0, stream, ws.data_ptr(),
+                       x8.data_ptr<unsigned char>(), sx.data_ptr<float>(),
+                       w_packed.data_ptr<unsigned char>(), sw.data_ptr<float>(),
+                       M, (int)N, (int)K, kps);
+    const long mn = (long)M * N;
+    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 255) / 256),
+                       dim3(256), 0, stream, (short*)out.data_ptr(),
+                       ws.data_ptr<float>(), mn, split);
+  }
+}
