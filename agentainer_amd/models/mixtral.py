@@ -18,8 +18,12 @@ MoE execution model, chosen for the MI355X serving regime:
     and one all-reduce combines — the xGMI-friendly formulation of EP for
     replicated-activation serving. A token all-to-all dispatch (for
     sequence-sharded activations) is the round-2 upgrade path.
-  * fp8 MFMA expert GEMMs (config 5's decode dtype) are a planned upgrade
-    of ops.skinny_gemm (OCP e4m3 + MX scaling); experts run bf16 today.
+  * fp8 MFMA expert GEMMs (config 5's decode dtype): expert weights are
+    quantized offline to OCP e4m3 with per-output-channel scales,
+    activations dynamically per token (ops.quantize_weight_fp8 /
+    linear_fp8, v_mfma_f32_16x16x32_fp8_fp8) — half the weight bytes of
+    bf16 on the streaming-bound decode path. Enable with
+    engine.expert_fp8: true (or agent dtype fp8).
 """
 
 from __future__ import annotations
