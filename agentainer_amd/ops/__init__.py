@@ -173,9 +173,10 @@ SKINNY_NT = os.environ.get("AGENTAINER_SKINNY_NT", "0") == "1"
 
 
 def _skinny_split(ntiles: int, K: int) -> int:
-    # target >= 512 blocks (2 per CU) so chunk barriers overlap across blocks
+    # target ~256 blocks (1 per CU — measured optimum: 2 blocks/CU halve
+    # the per-block LDS and contend on the chunk barriers)
     split = 1
-    while split < 16 and ntiles * split < 512 and K % (256 * split * 2) == 0:
+    while split < 16 and ntiles * split < 256 and K % (256 * split * 2) == 0:
         split *= 2
     return split
 

@@ -245,19 +245,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     }
   };
 
-  stage_load(k0);
-  stage_write(0);
-  w_load(bw_cur, 0);
-  __syncthreads();
-
-  int buf = 0;
-  long woff = 8 * 512;  // shorts into this wave's packed stream (chunk t+1)
-  for (int k = k0; k < k1; k += KC) {
-    const bool has_next = (k + KC < k1);
-    if (has_next) {
-      stage_load(k + KC);
-      w_load(bw_nxt, woff);
-    }
+  auto compute = [&](bf16v8 (&bw)[8], int buf) {
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       bf16v8 a[4];
@@ -267,15 +255,45 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
             &x_lds[buf][(ms * 16 + l16) * XS + i * 32 + lg * 8]);
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms)
-        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bw_cur[i],
+        acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ms], bw[i],
                                                           acc[ms], 0, 0, 0);
     }
+  };
+  stage_load(k0);
+  stage_write(0);
+  w_load(bw_cur, 0);
+  __syncthreads();
+
+  // ping-pong 2x unroll: no bw register copies, so chunk t+1's weight
+  // loads never drain at a chunk boundary (a copy would force vmcnt(0))
+  int buf = 0;
+  long woff = 8 * 512;
+  int k = k0;
+  while (true) {
+    bool has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_nxt, woff);
+      woff += 8 * 512;
+    }
+    compute(bw_cur, buf);
     if (has_next) stage_write(buf ^ 1);
     __syncthreads();
     buf ^= 1;
-    woff += 8 * 512;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) bw_cur[i] = bw_nxt[i];
+    k += KC;
+    if (!has_next) break;
+    has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_cur, woff);
+      woff += 8 * 512;
+    }
+    compute(bw_nxt, buf);
+    if (has_next) stage_write(buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+    k += KC;
+    if (!has_next) break;
   }
 
 #pragma unroll
@@ -439,19 +457,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
           reinterpret_cast<const long*>(wp + woff + (long)i * 512));
   };
 
-  stage_load(k0);
-  stage_write(0);
-  w_load(bw_cur, 0);
-  __syncthreads();
-
-  int buf = 0;
-  long woff = 8 * 512;
-  for (int k = k0; k < k1; k += KC) {
-    const bool has_next = (k + KC < k1);
-    if (has_next) {
-      stage_load(k + KC);
-      w_load(bw_nxt, woff);
-    }
+  auto compute = [&](long (&bw)[8], int buf) {
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       long a[4];
@@ -462,14 +468,42 @@ __global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms)
         acc[ms] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            a[ms], bw_cur[i], acc[ms], 0, 0, 0);
+            a[ms], bw[i], acc[ms], 0, 0, 0);
     }
+  };
+  stage_load(k0);
+  stage_write(0);
+  w_load(bw_cur, 0);
+  __syncthreads();
+
+  int buf = 0;
+  long woff = 8 * 512;
+  int k = k0;
+  while (true) {
+    bool has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_nxt, woff);
+      woff += 8 * 512;
+    }
+    compute(bw_cur, buf);
     if (has_next) stage_write(buf ^ 1);
     __syncthreads();
     buf ^= 1;
-    woff += 8 * 512;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) bw_cur[i] = bw_nxt[i];
+    k += KC;
+    if (!has_next) break;
+    has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_cur, woff);
+      woff += 8 * 512;
+    }
+    compute(bw_nxt, buf);
+    if (has_next) stage_write(buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+    k += KC;
+    if (!has_next) break;
   }
 
   const int col = n_tile * 16 + l16;
