@@ -13,11 +13,13 @@ MoE execution model, chosen for the MI355X serving regime:
     the same HBM traffic as gather/scatter — while keeping the step free
     of host syncs (hipGraph-capturable) .
   * Expert parallelism (EP over RCCL/xGMI): experts are PARTITIONED
-    across ranks; activations are replicated (attention output is already
-    all-reduced), so each rank computes its local experts' contribution
-    and one all-reduce combines — the xGMI-friendly formulation of EP for
-    replicated-activation serving. A token all-to-all dispatch (for
-    sequence-sharded activations) is the round-2 upgrade path.
+    across ranks. Decode keeps activations replicated (the attention
+    all-reduce already paid for that), each rank computes its local
+    experts and one all-reduce combines — latency-optimal for tiny
+    batches on xGMI. Prefill-scale batches switch to TOKEN-SHUFFLE EP:
+    each rank routes its token shard, ships tokens to their experts'
+    owner ranks with all_to_all over xGMI, computes, ships results back
+    and all-gathers the output shards (forward_a2a).
   * fp8 MFMA expert GEMMs (config 5's decode dtype): expert weights are
     quantized offline to OCP e4m3 with per-output-channel scales,
     activations dynamically per token (ops.quantize_weight_fp8 /
@@ -73,8 +75,14 @@ class MixtralMoE(torch.nn.Module):
         self.gate_up_fp8: List[Optional[tuple]] = [None] * self.n_local
         self.down_fp8: List[Optional[tuple]] = [None] * self.n_local
 
+    # prefill batches at least this big use token-shuffle all-to-all EP
+    A2A_MIN_TOKENS = 128
+
     def forward(self, h: torch.Tensor, ep_group=None) -> torch.Tensor:
         T = h.size(0)
+        if (ep_group is not None and T >= self.A2A_MIN_TOKENS
+                and torch.distributed.get_world_size(ep_group) > 1):
+            return self.forward_a2a(h, ep_group)
         logits = F.linear(h, self.router).float()           # [T, E]
         probs = torch.softmax(logits, dim=-1)
         topv, topi = probs.topk(self.top_k, dim=-1)         # [T, k]
@@ -104,6 +112,106 @@ class MixtralMoE(torch.nn.Module):
         if ep_group is not None:
             torch.distributed.all_reduce(out, group=ep_group)
         return out
+
+    # ---------- token-shuffle EP (prefill) ----------
+
+    @staticmethod
+    def _exchange(send, group):
+        """Variable-size tensor all-to-all: send[d] goes to rank d; returns
+        what every rank sent here. NCCL/RCCL uses all_to_all; gloo (CPU CI)
+        falls back to all_gather_object."""
+        import torch.distributed as dist
+
+        world = dist.get_world_size(group)
+        me = dist.get_rank(group)
+        backend = dist.get_backend(group)
+        if backend == "nccl":
+            counts = torch.tensor([t.shape[0] for t in send], device="cuda")
+            all_counts = torch.zeros(world, world, dtype=counts.dtype,
+                                     device="cuda")
+            dist.all_gather_into_tensor(all_counts.view(-1), counts, group=group)
+            cols = send[0].shape[1]
+            recv = [torch.empty(int(all_counts[d, me]), cols,
+                                dtype=send[0].dtype, device="cuda")
+                    for d in range(world)]
+            dist.all_to_all(recv, list(send), group=group)
+            return recv
+        holder = [None] * world
+        dist.all_gather_object(holder, [t.cpu() for t in send], group=group)
+        return [holder[d][me].to(send[0].device) for d in range(world)]
+
+    def forward_a2a(self, h: torch.Tensor, ep_group) -> torch.Tensor:
+        """Token-shuffle expert parallelism: route my token shard, ship
+        each (token, expert) pair to the expert owner rank, compute there,
+        ship weighted results back, all-gather output shards."""
+        import torch.distributed as dist
+
+        world = dist.get_world_size(ep_group)
+        me = dist.get_rank(ep_group)
+        T, H = h.shape
+        # contiguous token shard for this rank
+        base, rem = divmod(T, world)
+        t0 = me * base + min(me, rem)
+        ts = base + (1 if me < rem else 0)
+        hs = h[t0:t0 + ts]
+        logits = F.linear(hs, self.router).float()
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+        # flatten (token, k) pairs and bucket by owner rank
+        rows = torch.arange(ts, device=h.device).repeat_interleave(self.top_k)
+        experts = topi.reshape(-1)
+        weights = topv.reshape(-1)
+        owner = experts // self.n_local
+        send, send_meta = [], []
+        for d in range(world):
+            sel = (owner == d).nonzero(as_tuple=True)[0]
+            send.append(hs[rows[sel]].contiguous())
+            meta = torch.stack([
+                rows[sel].float(),
+                (experts[sel] % self.n_local).float(),
+                weights[sel].float()], dim=1)
+            send_meta.append(meta.contiguous())
+        recv = self._exchange(send, ep_group)
+        recv_meta = self._exchange(send_meta, ep_group)
+        # compute local experts on the received tokens, weighted
+        results = []
+        for d in range(world):
+            xd, md = recv[d], recv_meta[d]
+            yd = torch.zeros(xd.shape[0], H, dtype=torch.float32,
+                             device=h.device)
+            le = md[:, 1].long()
+            for i in range(self.n_local):
+                sel = (le == i).nonzero(as_tuple=True)[0]
+                if sel.numel() == 0:
+                    continue
+                xi = xd[sel]
+                gu = ops.linear(xi, self.gate_up[i], None)
+                gate, up = gu[:, :self.inter], gu[:, self.inter:]
+                act = torch.empty(xi.shape[0], self.inter, dtype=gu.dtype,
+                                  device=gu.device)
+                ops.silu_mul(act, gate, up)
+                eo = ops.linear(act, self.down[i], None)
+                yd[sel] = md[sel, 2].unsqueeze(1) * eo.float()
+            results.append(yd.to(h.dtype).contiguous())
+        returned = self._exchange(results, ep_group)
+        out_shard = torch.zeros(ts, H, dtype=torch.float32, device=h.device)
+        for d in range(world):
+            md = send_meta[d]
+            if md.shape[0]:
+                out_shard.index_add_(0, md[:, 0].long(),
+                                     returned[d].float())
+        # all-gather the output shards back to the replicated layout
+        shards = [None] * world
+        sizes = [base + (1 if r < rem else 0) for r in range(world)]
+        out_shard = out_shard.to(h.dtype)
+        if dist.get_backend(ep_group) == "nccl" and len(set(sizes)) == 1:
+            full = torch.empty(T, H, dtype=h.dtype, device=h.device)
+            dist.all_gather_into_tensor(full.view(world, -1).view(-1),
+                                        out_shard.contiguous(), group=ep_group)
+            return full
+        dist.all_gather_object(shards, out_shard.cpu(), group=ep_group)
+        return torch.cat([s.to(h.device) for s in shards], dim=0)
 
 
 class MixtralLayer(torch.nn.Module):
