@@ -174,6 +174,33 @@ class KVCacheManager:
         with self._lock:
             self._seqs[seq_id].length += 1
 
+    def decode_batch_prepare(self, seq_ids) -> list:
+        """One-lock batch form of ensure_decode_page: guarantees each
+        sequence's next-token page exists; returns the slot indices."""
+        rows = []
+        with self._lock:
+            for seq_id in seq_ids:
+                s = self._seqs[seq_id]
+                page_idx = s.length // self.page_size
+                if page_idx >= self.max_pages_per_seq:
+                    raise OutOfPages("sequence exceeded max_pages_per_seq")
+                if page_idx >= len(s.pages):
+                    if not self._free:
+                        raise OutOfPages(
+                            f"KV pool exhausted ({self.n_pages} pages)")
+                    page = self._free.pop()
+                    s.pages.append(page)
+                    if self.is_gpu:
+                        self.dev_page_table[self._slot_of[seq_id],
+                                            page_idx] = page
+                rows.append(self._slot_of[seq_id] if self.is_gpu else -1)
+        return rows
+
+    def advance_many(self, seq_ids) -> None:
+        with self._lock:
+            for seq_id in seq_ids:
+                self._seqs[seq_id].length += 1
+
     def free_seq(self, seq_id: str) -> None:
         with self._lock:
             s = self._seqs.pop(seq_id, None)

@@ -48,7 +48,7 @@ class EngineDead(RuntimeError):
     pass
 
 
-@dataclass
+@dataclass(eq=False)
 class GenRequest:
     agent_id: str
     prompt_tokens: List[int]
@@ -120,6 +120,7 @@ class ModelInstance:
         self.refcount = 0
         self.waiting: "queue.Queue[GenRequest]" = queue.Queue()
         self.running: List[GenRequest] = []
+        self._running_set = set()
         self._bindings: Dict[str, AgentBinding] = {}
         self._lock = threading.RLock()
         self._stop = threading.Event()
@@ -450,10 +451,8 @@ class ModelInstance:
         kvm = self.kvm
         B = len(plan)
         bucket = min(self._bucket(B), max(self.max_decode_batch, 1))
-        row_ids = []
-        for seq_id, _tok in plan:
-            kvm.ensure_decode_page(seq_id)   # host half of the append
-            row_ids.append(kvm.slot(seq_id))
+        # host half of the append, one lock for the whole batch
+        row_ids = kvm.decode_batch_prepare([s for s, _t in plan])
         entry = self._get_graph(bucket)
         entry["rows_pin"][:B] = torch.tensor(row_ids, dtype=torch.long)
         entry["rows_pin"][B:] = self._pad_slot
@@ -467,19 +466,20 @@ class ModelInstance:
         else:
             logits = self._device_decode_fwd(entry["rows"], entry["ids"],
                                              entry["inc"])[:B]
-        for seq_id, _tok in plan:
-            kvm.advance_host(seq_id)
+        kvm.advance_many([s for s, _t in plan])
         return logits
 
     def _finish_or_run(self, r: GenRequest, tok: int):
         """Called with lock held, after appending tok."""
         finished = (tok == self.tokenizer.eos_id or
                     len(r.generated) >= r.max_new)
-        if r in self.running:
+        if r in self._running_set:
             if finished:
                 self.running.remove(r)
+                self._running_set.discard(r)
         elif not finished:
             self.running.append(r)
+            self._running_set.add(r)
         if finished:
             b = self._bindings.get(r.agent_id)
             if b is not None:
@@ -532,8 +532,9 @@ class ModelInstance:
             if b.active is not None and not b.active.done.is_set():
                 b.active.error = "agent detached"
                 b.active.done.set()
-                if b.active in self.running:
+                if b.active in self._running_set:
                     self.running.remove(b.active)
+                    self._running_set.discard(b.active)
             while True:
                 try:
                     r = b.queue.get_nowait()

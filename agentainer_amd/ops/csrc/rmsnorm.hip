@@ -19,53 +19,61 @@
 
 namespace {
 
-template <bool FUSED_ADD>
-__global__ void rmsnorm_kernel(short* __restrict__ out,
-                               const short* __restrict__ x,
-                               short* __restrict__ residual,
-                               const short* __restrict__ w, float eps, int H) {
+// Single-pass: the row lives in registers between the reduction and the
+// scale, so x (and the fused residual sum) is read from HBM exactly once.
+// 512 threads/block; VPT = ceil(H / 4096) bf16x8 vectors per thread.
+template <bool FUSED_ADD, int VPT>
+__global__ __launch_bounds__(512) void rmsnorm_kernel(
+    short* __restrict__ out, const short* __restrict__ x,
+    short* __restrict__ residual, const short* __restrict__ w, float eps,
+    int H) {
   const int row = blockIdx.x;
   const int tid = threadIdx.x;
   const long base = (long)row * H;
-  __shared__ float red[4];
+  __shared__ float red[8];
 
-  // pass 1: sum of squares (and the fused residual add, kept in registers
-  // only when H fits one pass; otherwise re-read)
+  bf16x8 v[VPT];
   float ss = 0.f;
-  for (int i = tid * 8; i < H; i += blockDim.x * 8) {
-    bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + base + i);
-    if (FUSED_ADD) {
-      bf16x8 rv = *reinterpret_cast<const bf16x8*>(residual + base + i);
+#pragma unroll
+  for (int r = 0; r < VPT; ++r) {
+    const int i = (tid + r * 512) * 8;
+    if (i < H) {
+      v[r] = *reinterpret_cast<const bf16x8*>(x + base + i);
+      if (FUSED_ADD) {
+        bf16x8 rv = *reinterpret_cast<const bf16x8*>(residual + base + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[r][j] = f2bits(bits2f(v[r][j]) + bits2f(rv[j]));
+        *reinterpret_cast<bf16x8*>(residual + base + i) = v[r];
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float s = bits2f(xv[j]) + bits2f(rv[j]);
-        xv[j] = f2bits(s);
+        const float f = bits2f(v[r][j]);
+        ss = fmaf(f, f, ss);
       }
-      *reinterpret_cast<bf16x8*>(residual + base + i) = xv;
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float f = bits2f(xv[j]);
-      ss = fmaf(f, f, ss);
+    } else {
+      v[r] = bf16x8{};
     }
   }
   ss = wave_sum(ss);
-  const int wid = tid / WAVE;
-  if ((tid & (WAVE - 1)) == 0) red[wid] = ss;
+  if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = ss;
   __syncthreads();
-  ss = red[0] + red[1] + red[2] + red[3];
+  ss = 0.f;
+#pragma unroll
+  for (int wv = 0; wv < 8; ++wv) ss += red[wv];
   const float rms = rsqrtf(ss / (float)H + eps);
 
-  // pass 2: scale+weight (reads the post-add residual when fused)
-  const short* src = FUSED_ADD ? residual : x;
-  for (int i = tid * 8; i < H; i += blockDim.x * 8) {
-    bf16x8 xv = *reinterpret_cast<const bf16x8*>(src + base + i);
-    bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i);
-    bf16x8 ov;
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      ov[j] = f2bits(bits2f(xv[j]) * rms * bits2f(wv[j]));
-    *reinterpret_cast<bf16x8*>(out + base + i) = ov;
+  for (int r = 0; r < VPT; ++r) {
+    const int i = (tid + r * 512) * 8;
+    if (i < H) {
+      bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i);
+      bf16x8 ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        ov[j] = f2bits(bits2f(v[r][j]) * rms * bits2f(wv[j]));
+      *reinterpret_cast<bf16x8*>(out + base + i) = ov;
+    }
   }
 }
 
@@ -78,9 +86,16 @@ void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
   TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
   const int T = x.numel() / H;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL((rmsnorm_kernel<false>), dim3(T), dim3(256), 0, stream,
-                     (short*)out.data_ptr(), (const short*)x.data_ptr(),
-                     nullptr, (const short*)w.data_ptr(), (float)eps, H);
+#define RN_LAUNCH(VPT)                                                         \
+  hipLaunchKernelGGL((rmsnorm_kernel<false, VPT>), dim3(T), dim3(512), 0,      \
+                     stream, (short*)out.data_ptr(),                           \
+                     (const short*)x.data_ptr(), nullptr,                      \
+                     (const short*)w.data_ptr(), (float)eps, H)
+  if (H <= 4096) RN_LAUNCH(1);
+  else if (H <= 8192) RN_LAUNCH(2);
+  else if (H <= 16384) RN_LAUNCH(4);
+  else TORCH_CHECK(false, "rmsnorm: hidden > 16384 unsupported");
+#undef RN_LAUNCH
 }
 
 void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
@@ -92,8 +107,14 @@ void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
   TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
   const int T = x.numel() / H;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL((rmsnorm_kernel<true>), dim3(T), dim3(256), 0, stream,
-                     (short*)out.data_ptr(), (const short*)x.data_ptr(),
-                     (short*)residual.data_ptr(), (const short*)w.data_ptr(),
-                     (float)eps, H);
+#define RNF_LAUNCH(VPT)                                                        \
+  hipLaunchKernelGGL((rmsnorm_kernel<true, VPT>), dim3(T), dim3(512), 0,       \
+                     stream, (short*)out.data_ptr(),                           \
+                     (const short*)x.data_ptr(), (short*)residual.data_ptr(),  \
+                     (const short*)w.data_ptr(), (float)eps, H)
+  if (H <= 4096) RNF_LAUNCH(1);
+  else if (H <= 8192) RNF_LAUNCH(2);
+  else if (H <= 16384) RNF_LAUNCH(4);
+  else TORCH_CHECK(false, "rmsnorm: hidden > 16384 unsupported");
+#undef RNF_LAUNCH
 }
