@@ -67,8 +67,8 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
   const int l16 = lane % 16;    // fragment col / A row
   const int lg = lane / 16;     // fragment k-group
 
-  __shared__ short k_lds[QBLK * D];            // swizzled [32][128]
-  __shared__ short vt_lds[D * VT_STRIDE];      // [128][32+8]
+  __shared__ short k_lds[2][QBLK * D];         // swizzled [32][128], dbuf
+  __shared__ short vt_lds[2][D * VT_STRIDE];   // [128][32+8], dbuf
   __shared__ short p_lds[4][16 * P_STRIDE];    // per wave [16 q][32+8 kv]
 
   const int* pt = page_table + (long)b * max_pages;
@@ -106,42 +106,68 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
   const int kv_end = max_qpos + 1;
   const int n_tiles = (kv_end + KVBLK - 1) / KVBLK;
 
+  // T14 split staging: each thread owns 2 K chunks + 2 V chunks of a tile;
+  // tile t+1's GLOBAL loads issue before tile t's MFMAs, the LDS writes land
+  // after them, ONE barrier per tile.
+  const int kc0 = threadIdx.x;            // K chunk ids (tok = c%32, d8 = c/32)
+  const int kc1 = threadIdx.x + 256;
+  bf16x8 st_k0, st_k1, st_v0, st_v1;
+  auto stage_load = [&](int t) {
+    const int kv0 = t * KVBLK;
+    auto loadk = [&](int c) -> bf16x8 {
+      const int tok = c % QBLK, d8 = c / QBLK;
+      const int gt = kv0 + tok;
+      bf16x8 vvv{};
+      if (gt < kv_end && gt < seq_len) {
+        const long pg = pt[gt / PS];
+        vvv = *reinterpret_cast<const bf16x8*>(
+            k_cache + ((((pg * n_kv + g) * (D / 8) + d8) * PS) + gt % PS) * 8);
+      }
+      return vvv;
+    };
+    auto loadv = [&](int c) -> bf16x8 {
+      const int tok = c / 16, d0 = (c % 16) * 8;
+      const int gt = kv0 + tok;
+      bf16x8 vvv{};
+      if (gt < kv_end && gt < seq_len) {
+        const long pg = pt[gt / PS];
+        vvv = *reinterpret_cast<const bf16x8*>(
+            v_cache + (((pg * n_kv + g) * PS) + gt % PS) * D + d0);
+      }
+      return vvv;
+    };
+    st_k0 = loadk(kc0);
+    st_k1 = loadk(kc1);
+    st_v0 = loadv(kc0);
+    st_v1 = loadv(kc1);
+  };
+  auto stage_write = [&](int buf) {
+    auto writek = [&](int c, bf16x8 vvv) {
+      const int tok = c % QBLK, d8 = c / QBLK;
+      const int byte0 = tok * (D * 2) + d8 * 16;
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(k_lds[buf]) + kswz(tok, byte0)) = vvv;
+    };
+    auto writev = [&](int c, bf16x8 vvv) {
+      const int tok = c / 16, d0 = (c % 16) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[buf][(d0 + j) * VT_STRIDE + tok] = vvv[j];
+    };
+    writek(kc0, st_k0);
+    writek(kc1, st_k1);
+    writev(kc0, st_v0);
+    writev(kc1, st_v1);
+  };
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  int buf = 0;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * KVBLK;
-    // ---- stage K tile (swizzled) + V^T tile, cooperatively (256 threads)
-    __syncthreads();
-    {
-      // K: 512 chunks of 8 shorts => 2 rounds; c -> (tok = c%32, d8 = c/32)
-      for (int c = threadIdx.x; c < QBLK * (D / 8); c += 256) {
-        const int tok = c % QBLK, d8 = c / QBLK;
-        const int gt = kv0 + tok;
-        bf16x8 vvv{};
-        if (gt < kv_end && gt < seq_len) {
-          const long pg = pt[gt / PS];
-          const short* kp = k_cache +
-              ((((pg * n_kv + g) * (D / 8) + d8) * PS) + gt % PS) * 8;
-          vvv = *reinterpret_cast<const bf16x8*>(kp);
-        }
-        const int byte0 = tok * (D * 2) + d8 * 16;
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(k_lds) + kswz(tok, byte0)) = vvv;
-      }
-      // V^T: c -> (tok = c/16, d0 = (c%16)*8); scalar transpose writes
-      for (int c = threadIdx.x; c < QBLK * (D / 8); c += 256) {
-        const int tok = c / 16, d0 = (c % 16) * 8;
-        const int gt = kv0 + tok;
-        bf16x8 vvv{};
-        if (gt < kv_end && gt < seq_len) {
-          const long pg = pt[gt / PS];
-          const short* vp = v_cache +
-              (((pg * n_kv + g) * PS) + gt % PS) * D + d0;
-          vvv = *reinterpret_cast<const bf16x8*>(vp);
-        }
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vt_lds[(d0 + j) * VT_STRIDE + tok] = vvv[j];
-      }
-    }
-    __syncthreads();
+    const bool has_next = (t + 1 < n_tiles);
+    if (has_next) stage_load(t + 1);  // next tile's HBM loads in flight
 
     // ---- K fragments for this tile: [2 ksub][4 dchunk]
     bf16v8 kf[2][4];
@@ -152,7 +178,7 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
       for (int dc = 0; dc < 4; ++dc) {
         const int byte0 = tok * (D * 2) + (dc * 32 + lg * 8) * 2;
         kf[ks][dc] = *reinterpret_cast<const bf16v8*>(
-            reinterpret_cast<const char*>(k_lds) + kswz(tok, byte0));
+            reinterpret_cast<const char*>(k_lds[buf]) + kswz(tok, byte0));
       }
     }
     // ---- V^T fragments: [8 dim tiles]
@@ -161,7 +187,7 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
     for (int dt = 0; dt < 8; ++dt) {
       const int dim = dt * 16 + l16;
       vf[dt] = *reinterpret_cast<const bf16v8*>(
-          &vt_lds[dim * VT_STRIDE + lg * 8]);
+          &vt_lds[buf][dim * VT_STRIDE + lg * 8]);
     }
 
 #pragma unroll
@@ -234,6 +260,9 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
         o[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             pf, vf[dt], o[qs][dt], 0, 0, 0);
     }
+    if (has_next) stage_write(buf ^ 1);  // writes land after the MFMAs
+    __syncthreads();
+    buf ^= 1;
   }
 
   // ---- epilogue: O /= l, write bf16
