@@ -91,21 +91,13 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       qv[h][r] = *reinterpret_cast<const bf16x8*>(qp + (r * 4 + ds) * 8);
   }
 
-  // PV lane split: tg = lane/8 walks 8 tokens, dg = lane%8 owns 16 dims —
-  // per instruction the wave reads 64 x 32 B = a fully contiguous 2 KB V
-  // run (the old lane-per-dim-pair layout issued 4 B loads at 1/4 the
-  // coalescing). o reduced across the 8 token groups at publish time.
-  const int tg = lane / 8;
-  const int dg = lane % 8;
-  float m[RATIO], l[RATIO];
-  float o[RATIO][16];
+  float m[RATIO], l[RATIO], o0[RATIO], o1[RATIO];
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
-    m[h] = -FLT_MAX; l[h] = 0.f;
-#pragma unroll
-    for (int j = 0; j < 16; ++j) o[h][j] = 0.f;
+    m[h] = -FLT_MAX; l[h] = 0.f; o0[h] = 0.f; o1[h] = 0.f;
   }
   const int* pt = page_table + (long)b * max_pages;
+  const int d0 = lane * 2;
 
   for (int c = c0; c < c1; ++c) {
     float s[4][RATIO];  // [16-token sub-pass][head]
@@ -166,51 +158,56 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       const float csum = wave_sum(psum);
       const float alpha = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
       l[h] = l[h] * alpha + csum;
-#pragma unroll
-      for (int j = 0; j < 16; ++j) o[h][j] *= alpha;
+      o0[h] *= alpha;
+      o1[h] *= alpha;
       m[h] = mn;
     }
     __builtin_amdgcn_s_waitcnt(0);
 
-    // PV: this lane accumulates dims [dg*16, dg*16+16) over its 8 tokens
-    // (tok = base + tg*8 + u); V rows read as 2 x 16 B per token — the
-    // wave covers contiguous 2 KB per u-iteration.
+    // PV: lane = dim pair; V row loaded once, feeds all RATIO heads.
+    // 8-token blocks: preload 8 independent rows, then FMA.
     const int base_tok = c * CHUNK;
+    int t = 0;
+    for (; t + 8 <= c_len; t += 8) {
+      float v0[8], v1[8];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int t = tg * 8 + u;
+      for (int u = 0; u < 8; ++u) {
+        const int gt = base_tok + t + u;
+        const long page = pt[gt / PS];
+        const short* vp = v_cache + (((long)page * n_kv + g) * PS + gt % PS) * D + d0;
+        v0[u] = bits2f(vp[0]);
+        v1[u] = bits2f(vp[1]);
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+#pragma unroll
+        for (int h = 0; h < RATIO; ++h) {
+          const float pw = p_lds[h][t + u];
+          o0[h] = fmaf(pw, v0[u], o0[h]);
+          o1[h] = fmaf(pw, v1[u], o1[h]);
+        }
+      }
+    }
+    for (; t < c_len; ++t) {
       const int gt = base_tok + t;
-      const int gt_c = min(gt, len - 1);
-      const long page = pt[gt_c / PS];
-      const short* vp = v_cache +
-          (((long)page * n_kv + g) * PS + gt_c % PS) * D + dg * 16;
-      bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vp);
-      bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vp + 8);
+      const long page = pt[gt / PS];
+      const short* vp = v_cache + (((long)page * n_kv + g) * PS + gt % PS) * D + d0;
+      const float v0 = bits2f(vp[0]), v1 = bits2f(vp[1]);
 #pragma unroll
       for (int h = 0; h < RATIO; ++h) {
-        const float pw = p_lds[h][t];  // 0 for tokens past len
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          o[h][j] = fmaf(pw, bits2f(v0[j]), o[h][j]);
-          o[h][8 + j] = fmaf(pw, bits2f(v1[j]), o[h][8 + j]);
-        }
+        const float pw = p_lds[h][t];
+        o0[h] = fmaf(pw, v0, o0[h]);
+        o1[h] = fmaf(pw, v1, o1[h]);
       }
     }
   }
 
-  // publish partials (unnormalized): reduce o across the 8 token groups
-  // (lanes differing in bits 3..5), then the tg==0 lanes own their dims
+  // publish partials (unnormalized)
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
     float* pp = my_part + h * PART_STRIDE;
-#pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      float v = o[h][j];
-      v += __shfl_xor(v, 8, WAVE);
-      v += __shfl_xor(v, 16, WAVE);
-      v += __shfl_xor(v, 32, WAVE);
-      if (tg == 0) pp[dg * 16 + j] = v;
-    }
+    pp[2 * lane] = o0[h];
+    pp[2 * lane + 1] = o1[h];
     if (lane == 0) { pp[128] = m[h]; pp[129] = l[h]; }
   }
 }
