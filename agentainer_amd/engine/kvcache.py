@@ -201,6 +201,25 @@ class KVCacheManager:
             for seq_id in seq_ids:
                 self._seqs[seq_id].length += 1
 
+    def rollback_many(self, seq_ids) -> None:
+        """Undo one speculative decode append per sequence (async decode:
+        a row that turned out finished had a garbage token appended by the
+        in-flight step). Device decrement is stream-ordered after that
+        step's graph, so it lands exactly once the append has happened."""
+        if not seq_ids:
+            return
+        with self._lock:
+            slots = []
+            for seq_id in seq_ids:
+                self._seqs[seq_id].length -= 1
+                if self.is_gpu:
+                    slots.append(self._slot_of[seq_id])
+        if self.is_gpu and slots:
+            idx = torch.tensor(slots, dtype=torch.long, device=self.device)
+            self.dev_seq_lens.index_add_(
+                0, idx, torch.full((len(slots),), -1, dtype=torch.int32,
+                                   device=self.device))
+
     def free_seq(self, seq_id: str) -> None:
         with self._lock:
             s = self._seqs.pop(seq_id, None)

@@ -146,6 +146,14 @@ class ModelInstance:
         self.is_gpu = device.startswith("cuda")
         self.use_graph = (self.is_gpu and self.tp_size == 1
                           and bool(engine_cfg.get("graph_capture", True)))
+        # async decode (speculative one-step lag): the next decode step is
+        # launched with device-fed tokens BEFORE the previous step's tokens
+        # reach the host, so per-step host bookkeeping overlaps GPU work.
+        # Rows that resolve as finished get their speculative append rolled
+        # back (kvm.rollback_many, stream-ordered).
+        self.async_decode = (self.use_graph
+                             and bool(engine_cfg.get("async_decode", True)))
+        self._spec: Optional[Dict[str, Any]] = None
         self._graphs: Dict[int, Dict[str, Any]] = {}
         self._pad_slot = -1
         if self.is_gpu:
@@ -181,6 +189,8 @@ class ModelInstance:
             self._thread.start()
 
     def stop(self):
+        if self.async_decode:
+            self.drain_async()
         self._stop.set()
         self._wake.set()
         if self._thread is not None:
@@ -222,6 +232,12 @@ class ModelInstance:
     def step(self) -> bool:
         """Admit + prefill, then one decode step. Returns True if work ran."""
         self._step_started = time.time()
+        if self.async_decode:
+            ran = self._step_async()
+            self.steps += 1 if ran else 0
+            self._step_started = None
+            self.last_step_t = time.time()
+            return ran
         admitted = self._admit()
         if admitted:
             self._prefill(admitted)
@@ -235,6 +251,96 @@ class ModelInstance:
         self._step_started = None
         self.last_step_t = time.time()
         return bool(admitted) or ran_decode
+
+    # ---------- async (speculative) decode step ----------
+
+    def _step_async(self) -> bool:
+        kvm = self.kvm
+        dev = self.device
+        # phase 1: LAUNCH the next decode for the current running set; old
+        # rows' input tokens come straight from the previous step's device
+        # sample buffer (their host values are not resolved yet)
+        with self._lock:
+            batch = [r for r in self.running if not r.done.is_set()]
+        launched = None
+        if batch:
+            bucket = min(self._bucket(len(batch)), max(self.max_decode_batch, 1))
+            batch = batch[:bucket]
+            B = len(batch)
+            seq_ids = [self._bindings[r.agent_id].seq_id for r in batch]
+            rows = kvm.decode_batch_prepare(seq_ids)
+            entry = self._get_graph(bucket)
+            entry["rows_pin"][:B] = torch.tensor(rows, dtype=torch.long)
+            entry["rows_pin"][B:] = self._pad_slot
+            entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
+            prev = self._spec
+            pos = {id(r): i for i, r in enumerate(prev["reqs"])} if prev else {}
+            idx = [pos.get(id(r), -1) for r in batch]
+            entry["ids_pin"][:B] = torch.tensor(
+                [r.generated[-1] if idx[i] < 0 else 0
+                 for i, r in enumerate(batch)], dtype=torch.long)
+            entry["ids"][:B].copy_(entry["ids_pin"][:B], non_blocking=True)
+            if prev is not None and any(x >= 0 for x in idx):
+                gidx = torch.tensor([max(x, 0) for x in idx], dtype=torch.long,
+                                    device=dev)
+                mask = torch.tensor([x >= 0 for x in idx], dtype=torch.bool,
+                                    device=dev)
+                gathered = prev["sampled"].index_select(0, gidx)
+                entry["ids"][:B].copy_(
+                    torch.where(mask, gathered, entry["ids"][:B]))
+            entry["graph"].replay()
+            logits = entry["logits"][:B]
+            sampled = self._sample_device(logits, batch)
+            kvm.advance_many(seq_ids)
+            self.decode_tokens += B
+            self.occupancy_acc += B / max(1, self.max_decode_batch)
+            launched = {"reqs": batch, "sampled": sampled, "invalid": set()}
+        prev = self._spec
+        self._spec = launched
+
+        # phase 2: RESOLVE the previous step's tokens on the host while the
+        # GPU runs the step launched above
+        self._resolve_spec(prev, launched)
+
+        # phase 3: admission + prefill (runs after the in-flight decode)
+        admitted = self._admit()
+        if admitted:
+            self._prefill(admitted)
+        return bool(batch) or bool(admitted) or prev is not None
+
+    def _resolve_spec(self, prev, launched) -> None:
+        if prev is None:
+            return
+        toks = prev["sampled"].tolist()
+        rollback_seqs = []
+        with self._lock:
+            lpos = ({id(r): i for i, r in enumerate(launched["reqs"])}
+                    if launched else {})
+            for i, r in enumerate(prev["reqs"]):
+                if i in prev["invalid"] or r.done.is_set():
+                    # invalidated row, or the request died (detach) mid-flight
+                    continue
+                t = int(toks[i])
+                r.generated.append(t)
+                self._finish_or_run(r, t)
+                if r.done.is_set():
+                    # the just-launched step speculatively appended a token
+                    # for this row: mark invalid + roll back that append
+                    j = lpos.get(id(r))
+                    if j is not None:
+                        launched["invalid"].add(j)
+                        b = self._bindings.get(r.agent_id)
+                        if b is not None:
+                            rollback_seqs.append(b.seq_id)
+        self.kvm.rollback_many(rollback_seqs)
+
+    def drain_async(self) -> None:
+        """Resolve the in-flight speculative step WITHOUT launching a new
+        one (called before detach/offload and at engine stop, so
+        checkpoints never capture an unresolved speculative token)."""
+        prev = self._spec
+        self._spec = None
+        self._resolve_spec(prev, None)
 
     def _admit(self) -> List[GenRequest]:
         out: List[GenRequest] = []
@@ -274,6 +380,10 @@ class ModelInstance:
         return out
 
     def _sample(self, logits: torch.Tensor, reqs: List[GenRequest]) -> List[int]:
+        return self._sample_device(logits, reqs).tolist()
+
+    def _sample_device(self, logits: torch.Tensor,
+                       reqs: List[GenRequest]) -> torch.Tensor:
         B = logits.size(0)
         out = torch.empty(B, dtype=torch.long, device=logits.device)
         greedy_rows = [i for i, r in enumerate(reqs) if r.temperature <= 0.0]
@@ -298,7 +408,7 @@ class ModelInstance:
                  for i in samp_rows], dtype=torch.int64, device=logits.device)
             ops.topp_sample(sub, lb[samp_rows].contiguous(), temps, tps, seeds)
             out[samp_rows] = sub
-        return out.tolist()
+        return out
 
     def _bcast(self, cmd):
         if self.tp_size > 1 and self.tp_rank == 0:
@@ -522,6 +632,8 @@ class ModelInstance:
             self.refcount += 1
 
     def unbind(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:
+        if self.async_decode:
+            self.drain_async()
         with self._lock:
             b = self._bindings.pop(agent_id, None)
             if b is None:
