@@ -101,13 +101,19 @@ class ModelInstance:
         if weights_path:
             self.model.load_safetensors(weights_path)
         if device.startswith("cuda"):
-            import inspect
-            if "expert_fp8" in inspect.signature(
-                    self.model.pack_decode_weights).parameters:
-                self.model.pack_decode_weights(
-                    expert_fp8=bool(engine_cfg.get("expert_fp8", False)))
-            else:
-                self.model.pack_decode_weights()
+            # packed decode copies double weight memory; skip when the
+            # model alone already claims a large share of HBM (70B TP=1)
+            param_bytes = sum(p.numel() * p.element_size()
+                              for p in self.model.parameters())
+            free, total = torch.cuda.mem_get_info()
+            if param_bytes * 2 < 0.6 * total:
+                import inspect
+                if "expert_fp8" in inspect.signature(
+                        self.model.pack_decode_weights).parameters:
+                    self.model.pack_decode_weights(
+                        expert_fp8=bool(engine_cfg.get("expert_fp8", False)))
+                else:
+                    self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))
         n_pages = self._pool_pages(cfg, page_size, device, engine_cfg,
                                    self.tp_size)

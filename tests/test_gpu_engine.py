@@ -211,3 +211,45 @@ def test_async_decode_staggered_finish(gpu_rt):
             assert r2.generated == reqs[i].generated, (
                 f"agent {i}: async {reqs[i].generated} != sync {r2.generated}")
         e2.shutdown()
+
+
+def test_threaded_engine_chat_gpu(tmp_path):
+    """The real serving path on MI355X: engine THREAD (not sync_mode),
+    chat() through the text layer, concurrent agents."""
+    import threading
+
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import LLMEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    cfg.data["engine"]["kv_pool_gb"] = 1.0
+    store = Store(str(tmp_path / "state"), sync="never")
+    engine = LLMEngine(store, cfg, device="cuda", state_root=str(tmp_path))
+    manager = Manager(store, engine, cfg)
+    agents = []
+    for i in range(4):
+        a = manager.deploy(name=f"th{i}", model="tiny-llama",
+                           sampling={"max_tokens": 8})
+        manager.start(a.id)
+        agents.append(a)
+    results = {}
+
+    def chat(a):
+        results[a.id] = engine.chat(a.id, "hello from thread")
+
+    threads = [threading.Thread(target=chat, args=(a,)) for a in agents]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert len(results) == 4
+    outs = {r["response"] for r in results.values()}
+    assert len(outs) == 1  # same prompt + greedy => identical
+    assert all(r["tokens"] == 8 for r in results.values())
+    # history written through the text layer
+    hist = store.lrange(f"agent:{agents[0].id}:conversations")
+    assert len(hist) == 1
+    engine.shutdown()
+    store.close()
