@@ -293,6 +293,29 @@ def create_app(rt: Runtime) -> FastAPI:
         rt.backups.delete(backup_id)
         return envelope(True, "backup deleted")
 
+    @app.get("/backups/{backup_id}/export")
+    async def export_backup(backup_id: str, _tok: str = Depends(require_auth)):
+        import tempfile
+
+        from fastapi.responses import FileResponse
+
+        out = tempfile.mktemp(suffix=".tar.gz")
+        rt.backups.export(backup_id, out)
+        return FileResponse(out, filename=f"{backup_id}.tar.gz",
+                            media_type="application/gzip")
+
+    @app.post("/backups/import")
+    async def import_backup(request: HttpRequest,
+                            _tok: str = Depends(require_auth)):
+        import tempfile
+
+        raw = await request.body()
+        tmp = tempfile.mktemp(suffix=".tar.gz")
+        with open(tmp, "wb") as f:
+            f.write(raw)
+        ids = rt.backups.import_(tmp)
+        return envelope(True, "", {"imported": ids})
+
     @app.get("/audit")
     async def audit_logs(user: str = "", action: str = "", resource: str = "",
                          limit: int = 200, _tok: str = Depends(require_auth)):

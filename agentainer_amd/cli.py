@@ -311,6 +311,33 @@ def backup_delete(client: Client, backup_id):
     click.echo(resp.get("message") or "ok")
 
 
+@backup.command("export")
+@click.argument("backup_id")
+@click.option("--output", "-o", required=True, type=click.Path())
+@pass_client
+def backup_export(client: Client, backup_id, output):
+    r = httpx.get(client.url + f"/backups/{backup_id}/export",
+                  headers={"Authorization": f"Bearer {client.token}"},
+                  timeout=120.0)
+    r.raise_for_status()
+    with open(output, "wb") as f:
+        f.write(r.content)
+    click.echo(f"exported {backup_id} -> {output}")
+
+
+@backup.command("import")
+@click.argument("bundle", type=click.Path(exists=True))
+@pass_client
+def backup_import(client: Client, bundle):
+    with open(bundle, "rb") as f:
+        raw = f.read()
+    r = httpx.post(client.url + "/backups/import", content=raw,
+                   headers={"Authorization": f"Bearer {client.token}"},
+                   timeout=120.0)
+    r.raise_for_status()
+    click.echo(json.dumps(r.json().get("data")))
+
+
 def main():
     cli()
 

@@ -95,3 +95,17 @@ def test_cli_unreachable_server():
     runner = CliRunner()
     res = runner.invoke(cli, ["--url", "http://127.0.0.1:9", "list"])
     assert res.exit_code == 2
+
+
+def test_cli_backup_export_import(run, server, tmp_path):
+    run("deploy", "echo", "--name", "exp-agent")
+    out = run("backup", "create", "--name", "expsnap")
+    import json as _json
+    bid = _json.loads(out)["id"]
+    bundle = str(tmp_path / "b.tar.gz")
+    out = run("backup", "export", bid, "-o", bundle)
+    assert "exported" in out
+    run("backup", "delete", bid)
+    out = run("backup", "import", bundle)
+    assert bid in out
+    assert bid in run("backup", "list")
