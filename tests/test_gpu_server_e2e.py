@@ -1,0 +1,72 @@
+"""Full-stack E2E on MI355X: real server process (uvicorn + threaded
+engine + llama3-8b), concurrent HTTP chat load, lifecycle + KV restore
+through the REST API."""
+
+import concurrent.futures
+import sys
+import time
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+sys.path.insert(0, ".")
+from test_crash_integration import Server, _free_port  # noqa: E402
+
+
+@pytest.mark.timeout(900)
+def test_server_llama8b_concurrent_chat(tmp_path):
+    root = str(tmp_path / "root")
+    srv = Server(root, _free_port(), device="cuda")
+    try:
+        srv.start(timeout=300)  # model load + pool allocation
+        agents = []
+        for i in range(8):
+            st, resp = srv.call("POST", "/agents", {
+                "name": f"prod-{i}", "model": "llama3-8b",
+                "auto_restart": True, "sampling": {"max_tokens": 32}})
+            assert st == 200, resp
+            agents.append(resp["data"]["id"])
+        for aid in agents:
+            assert srv.call("POST", f"/agents/{aid}/start", timeout=300)[0] == 200
+
+        def one_chat(args):
+            aid, i = args
+            st, out = srv.call("POST", f"/agent/{aid}/chat",
+                               body={"message": f"msg-{i}"}, auth=False,
+                               timeout=300)
+            assert st == 200, out
+            assert out["tokens"] == 32
+            return out["e2e_s"]
+
+        t0 = time.time()
+        jobs = [(aid, i) for i in range(5) for aid in agents]  # 40 chats
+        with concurrent.futures.ThreadPoolExecutor(max_workers=16) as ex:
+            lat = list(ex.map(one_chat, jobs))
+        wall = time.time() - t0
+        print(f"# 40 concurrent chats in {wall:.2f}s "
+              f"({len(jobs)/wall:.1f} req/s), mean e2e {sum(lat)/len(lat):.3f}s",
+              file=sys.stderr)
+
+        # multi-turn KV continuity + stop/resume via REST
+        aid = agents[0]
+        h1 = srv.call("GET", f"/agent/{aid}/history", auth=False)[1]["history"]
+        assert len(h1) == 5
+        assert srv.call("POST", f"/agents/{aid}/stop")[0] == 200
+        assert srv.call("POST", f"/agents/{aid}/resume")[0] == 200
+        st, out = srv.call("POST", f"/agent/{aid}/chat",
+                           body={"message": "after resume"}, auth=False,
+                           timeout=300)
+        assert st == 200 and out["tokens"] == 32
+
+        # engine metrics live
+        st, m = srv.call("GET", "/metrics/engine")
+        assert st == 200
+        assert m["data"]["models"]["llama3-8b"]["decode_tokens"] > 0
+        assert m["data"].get("hbm_total_bytes", 0) > 0
+    finally:
+        srv.terminate()
