@@ -34,10 +34,12 @@ def _free_port():
 
 
 class Server:
-    def __init__(self, root: str, port: int, device: str = "echo"):
+    def __init__(self, root: str, port: int, device: str = "echo",
+                 kv_pool_gb: float = 0.02):
         self.root = root
         self.port = port
         self.device = device
+        self.kv_pool_gb = kv_pool_gb  # 0 = engine auto-size (GPU runs)
         self.proc = None
         self.base = f"http://127.0.0.1:{port}"
 
@@ -46,7 +48,7 @@ class Server:
         env.update({
             "AGENTAINER_STORE_PATH": self.root,
             "AGENTAINER_SERVER_PORT": str(self.port),
-            "AGENTAINER_ENGINE_KV_POOL_GB": "0.02",
+            "AGENTAINER_ENGINE_KV_POOL_GB": str(self.kv_pool_gb),
             "AGENTAINER_FEATURES_REPLAY_INTERVAL_S": "0.2",
             "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
         })

@@ -21,7 +21,7 @@ from test_crash_integration import Server, _free_port  # noqa: E402
 @pytest.mark.timeout(900)
 def test_server_llama8b_concurrent_chat(tmp_path):
     root = str(tmp_path / "root")
-    srv = Server(root, _free_port(), device="cuda")
+    srv = Server(root, _free_port(), device="cuda", kv_pool_gb=16.0)
     try:
         srv.start(timeout=300)  # model load + pool allocation
         agents = []
