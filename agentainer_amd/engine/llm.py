@@ -129,6 +129,10 @@ class ModelInstance:
         self._running_set = set()
         self._bindings: Dict[str, AgentBinding] = {}
         self._lock = threading.RLock()
+        # serializes whole engine steps against lifecycle ops (bind/unbind
+        # drain the speculative step and mutate the KV pool — they must
+        # never interleave with a step on the engine thread)
+        self._step_mutex = threading.Lock()
         self._stop = threading.Event()
         self._wake = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -218,7 +222,8 @@ class ModelInstance:
     def _loop(self):
         while not self._stop.is_set():
             try:
-                did = self.step()
+                with self._step_mutex:
+                    did = self.step()
             except Exception:
                 traceback.print_exc()
                 did = False
@@ -628,6 +633,16 @@ class ModelInstance:
 
     def bind(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint],
              worker_has_ckpt: bool = False):
+        if not self.sync_mode:
+            self._step_mutex.acquire()
+        try:
+            self._bind_locked(agent, seq_id, ckpt, worker_has_ckpt)
+        finally:
+            if not self.sync_mode:
+                self._step_mutex.release()
+
+    def _bind_locked(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint],
+                     worker_has_ckpt: bool = False):
         with self._lock:
             self._bcast(("bind", self.name, seq_id,
                          ckpt is not None or worker_has_ckpt))
@@ -638,6 +653,15 @@ class ModelInstance:
             self.refcount += 1
 
     def unbind(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:
+        if not self.sync_mode:
+            self._step_mutex.acquire()
+        try:
+            return self._unbind_locked(agent_id, offload)
+        finally:
+            if not self.sync_mode:
+                self._step_mutex.release()
+
+    def _unbind_locked(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:
         if self.async_decode:
             self.drain_async()
         with self._lock:

@@ -52,10 +52,13 @@ class Server:
             "AGENTAINER_FEATURES_REPLAY_INTERVAL_S": "0.2",
             "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
         })
+        self.log_path = os.path.join(self.root, f"server-{self.port}.log")
+        os.makedirs(self.root, exist_ok=True)
+        self._log_f = open(self.log_path, "ab")
         self.proc = subprocess.Popen(
             [sys.executable, "-m", "agentainer_amd.cli", "server",
              "--engine-device", self.device],
-            env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+            env=env, stdout=self._log_f, stderr=subprocess.STDOUT,
             start_new_session=True)
         deadline = time.time() + timeout
         while time.time() < deadline:
@@ -69,6 +72,13 @@ class Server:
                 raise RuntimeError(f"server died rc={self.proc.returncode}")
             time.sleep(0.2)
         raise TimeoutError("server did not come up")
+
+    def log_tail(self, n=4000) -> str:
+        try:
+            with open(self.log_path, "rb") as f:
+                return f.read()[-n:].decode("utf-8", "replace")
+        except OSError:
+            return "(no log)"
 
     def kill9(self):
         """SIGKILL — the docker-kill analog. No flush, no goodbye."""

@@ -61,7 +61,7 @@ def test_server_llama8b_concurrent_chat(tmp_path):
         st, out = srv.call("POST", f"/agent/{aid}/chat",
                            body={"message": "after resume"}, auth=False,
                            timeout=300)
-        assert st == 200 and out["tokens"] == 32
+        assert st == 200 and out["tokens"] == 32, (st, out, srv.log_tail())
 
         # engine metrics live
         st, m = srv.call("GET", "/metrics/engine")
