@@ -80,8 +80,10 @@ class Runtime:
             max_retries=int(feats.get("max_retries", 3)),
         )
         self.persistence_enabled = bool(feats.get("request_persistence", True))
+        self._inflight: set = set()
         self.replay = ReplayWorker(self.requests, self.agents, self._dispatch_replay,
-                                   interval_s=float(feats.get("replay_interval_s", 5.0)))
+                                   interval_s=float(feats.get("replay_interval_s", 5.0)),
+                                   inflight=self._inflight)
         self.reconciler = Reconciler(self.agents, interval_s=10.0)
         self.health = HealthMonitor(
             self.store, self.agents,
@@ -152,6 +154,8 @@ class Runtime:
             }
 
         t0 = time.time()
+        if req is not None:
+            self._inflight.add(req.id)
         try:
             payload = self._serve(agent, method, path, body)
         except EngineUnavailable as exc:
@@ -165,6 +169,9 @@ class Runtime:
                 self.requests.mark_failed(agent_id, req.id, str(exc))
             return 500, {"success": False, "message": str(exc),
                          "data": {"request_id": req.id if req else None}}
+        finally:
+            if req is not None:
+                self._inflight.discard(req.id)
         e2e = time.time() - t0
         if req is not None:
             self.requests.store_response(agent_id, req.id, payload)

@@ -40,11 +40,18 @@ class EngineUnavailable(Exception):
 
 class ReplayWorker:
     def __init__(self, requests: RequestManager, agents: Manager,
-                 dispatch: DispatchFn, interval_s: float = 5.0):
+                 dispatch: DispatchFn, interval_s: float = 5.0,
+                 inflight: Optional[set] = None):
         self.requests = requests
         self.agents = agents
         self.dispatch = dispatch
         self.interval_s = interval_s
+        # ids currently being dispatched by the live proxy path: a request
+        # is only "orphaned" (replayable) if no dispatch holds it. In-memory
+        # on purpose — a crash empties it, so everything pending at death
+        # replays (at-least-once), while live slow generations are not
+        # double-dispatched.
+        self.inflight = inflight if inflight is not None else set()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
 
@@ -79,6 +86,8 @@ class ReplayWorker:
             for req in self.requests.pending(agent_id):
                 if req.retry_count >= req.max_retries:
                     continue
+                if req.id in self.inflight:
+                    continue  # live dispatch in progress — not orphaned
                 try:
                     resp = self.dispatch(agent_id, req, replay=True)
                 except EngineUnavailable:

@@ -55,7 +55,7 @@ def test_server_llama8b_concurrent_chat(tmp_path):
         # multi-turn KV continuity + stop/resume via REST
         aid = agents[0]
         h1 = srv.call("GET", f"/agent/{aid}/history", auth=False)[1]["history"]
-        assert len(h1) == 5
+        assert len(h1) == 5, (h1, srv.log_tail())  # no replay duplicates
         assert srv.call("POST", f"/agents/{aid}/stop")[0] == 200
         assert srv.call("POST", f"/agents/{aid}/resume")[0] == 200
         st, out = srv.call("POST", f"/agent/{aid}/chat",

@@ -124,3 +124,18 @@ def test_hot_path_completes_request(runtime):
     assert "echo" in payload["response"]
     done = runtime.requests.by_queue(a.id, "completed")
     assert len(done) == 1 and done[0].response == payload
+
+
+def test_replay_skips_inflight(runtime):
+    """A request mid-dispatch (slow generation) must NOT be re-dispatched
+    by the replay worker; after a crash the in-flight set is empty so it
+    replays (at-least-once preserved)."""
+    a = _deploy_started(runtime, name="slow")
+    req = runtime.requests.store_request(a.id, "POST", "/chat",
+                                         body={"message": "slow one"})
+    runtime._inflight.add(req.id)   # simulate live dispatch in progress
+    assert runtime.replay.tick() == 0
+    assert len(runtime.requests.pending(a.id)) == 1
+    runtime._inflight.discard(req.id)  # dispatch died (crash analog)
+    assert runtime.replay.tick() == 1
+    assert runtime.requests.pending(a.id) == []
