@@ -37,6 +37,7 @@ from typing import Any, Dict, Optional
 
 from fastapi import Depends, FastAPI, Header, Query, Request as HttpRequest
 from fastapi.responses import JSONResponse
+from starlette.concurrency import run_in_threadpool
 
 from ..registry import AgentNotFound
 from ..engine.base import ModelNotFound
@@ -111,8 +112,12 @@ def create_app(rt: Runtime) -> FastAPI:
                 body = {"message": raw.decode("utf-8", "replace")} if raw else {}
         replay = request.headers.get("x-agentainer-replay", "").lower() == "true"
         req_id = request.headers.get("x-agentainer-request-id") or None
-        status, payload = rt.agent_request(
-            agent_id, request.method, "/" + agent_path, body=body,
+        # blocking dispatch (a generation can take seconds) runs in the
+        # threadpool so concurrent requests reach the engine together —
+        # that concurrency IS what feeds the continuous-batching scheduler
+        status, payload = await run_in_threadpool(
+            rt.agent_request, agent_id, request.method, "/" + agent_path,
+            body=body,
             headers={"content-type": request.headers.get("content-type", "")},
             replay=replay, req_id=req_id)
         return JSONResponse(status_code=status, content=payload)
@@ -135,19 +140,20 @@ def create_app(rt: Runtime) -> FastAPI:
         if len(env) > 50:
             return JSONResponse(status_code=422,
                                 content=envelope(False, "too many env vars (max 50)"))
-        agent = rt.agents.deploy(
-            name=name, model=model,
-            dtype=body.get("dtype", "bf16"),
-            tp_degree=int(body.get("tp_degree", 1)),
-            kv_budget=int(body.get("kv_budget", 0)),
-            max_context=int(body.get("max_context", 8192)),
-            env=env,
-            auto_restart=bool(body.get("auto_restart", False)),
-            token=body.get("token"),
-            health_check=body.get("health_check"),
-            system_prompt=body.get("system_prompt", ""),
-            sampling=body.get("sampling") or {},
-        )
+        agent = await run_in_threadpool(
+            lambda: rt.agents.deploy(
+                name=name, model=model,
+                dtype=body.get("dtype", "bf16"),
+                tp_degree=int(body.get("tp_degree", 1)),
+                kv_budget=int(body.get("kv_budget", 0)),
+                max_context=int(body.get("max_context", 8192)),
+                env=env,
+                auto_restart=bool(body.get("auto_restart", False)),
+                token=body.get("token"),
+                health_check=body.get("health_check"),
+                system_prompt=body.get("system_prompt", ""),
+                sampling=body.get("sampling") or {},
+            ))
         cl = _client(request)
         rt.logger.audit("api", "deploy", agent.id, "success", **cl)
         return envelope(True, f"agent {agent.name} deployed", agent.to_dict())
@@ -167,7 +173,7 @@ def create_app(rt: Runtime) -> FastAPI:
                           _tok: str = Depends(require_auth)):
             fn = getattr(rt.agents, op)
             try:
-                agent = fn(agent_id)
+                agent = await run_in_threadpool(fn, agent_id)
             except AgentNotFound:
                 raise
             except Exception as exc:  # noqa: BLE001
@@ -203,8 +209,8 @@ def create_app(rt: Runtime) -> FastAPI:
         body = await request.json()
         path = body.get("path", "/chat")
         method = body.get("method", "POST")
-        status, payload = rt.agent_request(agent_id, method, path,
-                                           body=body.get("body", body))
+        status, payload = await run_in_threadpool(
+            rt.agent_request, agent_id, method, path, body=body.get("body", body))
         return JSONResponse(status_code=status,
                             content=envelope(status < 400, "", payload))
 
@@ -234,8 +240,9 @@ def create_app(rt: Runtime) -> FastAPI:
         if r is None:
             return JSONResponse(status_code=404,
                                 content=envelope(False, f"request {req_id} not found"))
-        status, payload = rt.agent_request(agent_id, r.method, r.path, body=r.body,
-                                           replay=True, req_id=req_id)
+        status, payload = await run_in_threadpool(
+            rt.agent_request, agent_id, r.method, r.path, body=r.body,
+            replay=True, req_id=req_id)
         if status == 200:
             rt.requests.store_response(agent_id, req_id, payload)
         return JSONResponse(status_code=status, content=envelope(status < 400, "", payload))
@@ -272,8 +279,9 @@ def create_app(rt: Runtime) -> FastAPI:
     @app.post("/backups")
     async def create_backup(request: HttpRequest, _tok: str = Depends(require_auth)):
         body = await request.json()
-        b = rt.backups.create(body.get("name", "backup"), body.get("description", ""),
-                              body.get("agent_ids"))
+        b = await run_in_threadpool(
+            rt.backups.create, body.get("name", "backup"),
+            body.get("description", ""), body.get("agent_ids"))
         rt.logger.audit("api", "backup.create", b["id"], "success", **_client(request))
         return envelope(True, "", {"id": b["id"], "n_agents": len(b["agents"])})
 
@@ -284,7 +292,7 @@ def create_app(rt: Runtime) -> FastAPI:
     @app.post("/backups/{backup_id}/restore")
     async def restore_backup(backup_id: str, request: HttpRequest,
                              _tok: str = Depends(require_auth)):
-        agents = rt.backups.restore(backup_id)
+        agents = await run_in_threadpool(rt.backups.restore, backup_id)
         rt.logger.audit("api", "backup.restore", backup_id, "success", **_client(request))
         return envelope(True, "", [a.to_dict() for a in agents])
 

@@ -222,3 +222,66 @@ def test_network_isolation(tmp_path):
         assert srv.call("POST", f"/agents/{aid}/stop", auth=False)[0] == 401
     finally:
         srv.terminate()
+
+
+@pytest.mark.timeout(300)
+def test_kill_mid_stream(tmp_path):
+    """BASELINE config 4's crash contract, literally: SIGKILL the server
+    while a generation is IN FLIGHT; after restart the unacked request is
+    still pending and replays to completion (greedy => the regenerated
+    response is the one the caller would have gotten)."""
+    import threading
+
+    root = str(tmp_path / "root")
+    port = _free_port()
+    srv = Server(root, port, device="cpu")
+    killed = {}
+    try:
+        srv.start(timeout=120)
+        st, resp = srv.call("POST", "/agents", {
+            "name": "midstream", "model": "tiny-llama", "auto_restart": True,
+            "sampling": {"max_tokens": 800}})  # ~seconds of CPU decode
+        aid = resp["data"]["id"]
+        assert srv.call("POST", f"/agents/{aid}/start")[0] == 200
+
+        def fire():
+            try:
+                srv.call("POST", f"/agent/{aid}/chat",
+                         body={"message": "cut me off"}, auth=False,
+                         timeout=120)
+            except Exception as exc:  # connection dies with the server
+                killed["client_error"] = type(exc).__name__
+
+        th = threading.Thread(target=fire)
+        th.start()
+        # wait until the request is admitted (visible as pending in the WAL)
+        deadline = time.time() + 30
+        rid = None
+        while time.time() < deadline and rid is None:
+            st, r = srv.call("GET", f"/agents/{aid}/requests")
+            pend = r.get("data", {}).get("pending", [])
+            if pend:
+                rid = pend[0]["id"]
+            time.sleep(0.05)
+        assert rid, 'request never became pending'
+        time.sleep(0.3)  # let decoding actually start
+        srv.kill9()      # mid-stream
+        th.join(timeout=30)
+    finally:
+        srv.terminate()
+
+    srv2 = Server(root, port, device="cpu")
+    try:
+        srv2.start(timeout=120)
+        deadline = time.time() + 120
+        rec = None
+        while time.time() < deadline:
+            st, r = srv2.call("GET", f"/agents/{aid}/requests/{rid}")
+            if st == 200 and r["data"]["status"] == "completed":
+                rec = r["data"]
+                break
+            time.sleep(0.5)
+        assert rec is not None, "mid-stream request did not replay"
+        assert rec["response"]["tokens"] == 800  # full regeneration
+    finally:
+        srv2.terminate()
