@@ -62,6 +62,7 @@ class GenRequest:
     enq_t: float = 0.0
     first_token_t: float = 0.0
     fin_t: float = 0.0
+    trace_id: str = ""  # WAL request id — end-to-end tracing
 
 
 @dataclass
@@ -868,6 +869,7 @@ class LLMEngine:
         agent = b.agent
         sampling = dict(agent.sampling or {})
         sampling.update(kwargs)
+        trace_id = str(sampling.pop("trace_id", ""))
         prompt = self._build_prompt(agent, message)
         req = GenRequest(
             agent_id=agent_id,
@@ -876,6 +878,7 @@ class LLMEngine:
             temperature=float(sampling.get("temperature", 0.0)),
             top_p=float(sampling.get("top_p", 1.0)),
             seed=int(sampling.get("seed", 0)),
+            trace_id=trace_id,
         )
         with inst._lock:
             b.queue.put(req)
@@ -896,6 +899,7 @@ class LLMEngine:
         return {
             "response": text,
             "model": model,
+            "trace_id": req.trace_id or None,
             "tokens": len(req.generated),
             "ttft_s": (req.first_token_t - req.enq_t) if req.first_token_t else None,
             "e2e_s": (req.fin_t - req.enq_t) if req.fin_t else None,

@@ -157,7 +157,8 @@ class Runtime:
         if req is not None:
             self._inflight.add(req.id)
         try:
-            payload = self._serve(agent, method, path, body)
+            payload = self._serve(agent, method, path, body,
+                                  trace_id=req.id if req else "")
         except EngineUnavailable as exc:
             # crash-capture: leave pending (server.go:597-605)
             return 503, {"success": False,
@@ -186,17 +187,21 @@ class Runtime:
         agent = self.agents.get(agent_id)
         if agent.status != RUNNING:
             raise EngineUnavailable(f"agent {agent_id} is {agent.status}")
-        return self._serve(agent, req.method, req.path, req.body)
+        return self._serve(agent, req.method, req.path, req.body,
+                           trace_id=req.id)
 
     # ---------- per-agent endpoint surface (gpt-agent app.py contract) ----------
 
-    def _serve(self, agent, method: str, path: str, body: Any) -> Dict[str, Any]:
+    def _serve(self, agent, method: str, path: str, body: Any,
+               trace_id: str = "") -> Dict[str, Any]:
         path = "/" + path.strip("/")
         body = body or {}
         if path == "/chat":
             message = body.get("message", "") if isinstance(body, dict) else str(body)
-            out = self.engine.chat(agent.id, message,
-                                   **(body.get("sampling", {}) if isinstance(body, dict) else {}))
+            kwargs = dict(body.get("sampling", {})) if isinstance(body, dict) else {}
+            if trace_id:
+                kwargs["trace_id"] = trace_id
+            out = self.engine.chat(agent.id, message, **kwargs)
             return out
         if path == "/health":
             ok = self.engine.health_probe(agent.id)
