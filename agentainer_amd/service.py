@@ -222,9 +222,14 @@ class Runtime:
 
     # ---------- crash recovery ----------
 
-    def recover(self) -> int:
+    def recover(self, block: bool = False) -> int:
         """Boot-time recovery: reconcile, restart auto-restart agents and
-        replay their pending WAL (the resume -> replay flow)."""
+        kick off replay of their pending WAL (the resume -> replay flow,
+        SURVEY.md §3.5). Replay runs in the background by default so a
+        server with long pending generations still starts serving
+        immediately (the replay worker owns the retry cadence)."""
+        import threading
+
         self.reconciler.sync_all()
         for agent in self.agents.list():
             if agent.auto_restart and agent.status != RUNNING:
@@ -233,4 +238,9 @@ class Runtime:
                 except Exception:
                     self.logger.error(f"auto-restart of {agent.id} failed",
                                       component="service", agent_id=agent.id)
-        return self.replay.tick()
+        if block:
+            return self.replay.tick()
+        if self.persistence_enabled:
+            threading.Thread(target=self.replay.tick, name="boot-replay",
+                             daemon=True).start()
+        return 0
