@@ -81,23 +81,35 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
   typedef __bf16 bf2 __attribute__((ext_vector_type(2)));
   // lane (ds, t16) owns dims {(4r+ds)*8 .. +8} for r in 0..3: for each r
   // the wave's 64 lanes cover 4 CONSECUTIVE d8 groups x 16 consecutive
-  // tokens = one contiguous 1 KB run per page (full coalescing)
-  bf16x8 qv[RATIO][4];
+  // tokens = one contiguous 1 KB run per page (full coalescing).
+  // q parks in LDS (RATIO x 256 B/wave) and is read per k-slice as a
+  // broadcast b128 — holding it in VGPRs cost 64 registers (an occupancy
+  // level at RATIO=4).
+  __shared__ short q_lds[RATIO][D];
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
     const short* qp = q + (long)b * q_ts + (long)(qh0 + h) * D;
 #pragma unroll
-    for (int r = 0; r < 4; ++r)
-      qv[h][r] = *reinterpret_cast<const bf16x8*>(qp + (r * 4 + ds) * 8);
+    for (int r = 0; r < D / WAVE; ++r)
+      q_lds[h][r * WAVE + lane] = qp[r * WAVE + lane];
   }
+  __builtin_amdgcn_s_waitcnt(0);
 
-  float m[RATIO], l[RATIO], o0[RATIO], o1[RATIO];
+  // PV lane split: pv_tg = lane/16 walks tokens (4 per u-step, adjacent
+  // rows => 1 KB contiguous per load instruction), pv_dg = lane%16 owns
+  // 8 dims (one 16 B load per row). o[RATIO][8] keeps the register count
+  // inside 3 waves/SIMD.
+  const int pv_tg = lane / 16;
+  const int pv_dg = lane % 16;
+  float m[RATIO], l[RATIO];
+  float o[RATIO][8];
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
-    m[h] = -FLT_MAX; l[h] = 0.f; o0[h] = 0.f; o1[h] = 0.f;
+    m[h] = -FLT_MAX; l[h] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[h][j] = 0.f;
   }
   const int* pt = page_table + (long)b * max_pages;
-  const int d0 = lane * 2;
 
   for (int c = c0; c < c1; ++c) {
     float s[4][RATIO];  // [16-token sub-pass][head]
@@ -120,14 +132,17 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const bf2* kp2 = reinterpret_cast<const bf2*>(&kv[r]);
+          // this lane's dims (4r+ds)*8..+8 of each head's q: broadcast
+          // LDS reads (t16 lanes share the address — conflict-free)
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
+          for (int h = 0; h < RATIO; ++h) {
+            const bf16x8 qv8 = *reinterpret_cast<const bf16x8*>(
+                &q_lds[h][(r * 4 + ds) * 8]);
+            const bf2* qp2 = reinterpret_cast<const bf2*>(&qv8);
 #pragma unroll
-            for (int h = 0; h < RATIO; ++h) {
-              const bf2* qp2 = reinterpret_cast<const bf2*>(&qv[h][r]);
+            for (int j = 0; j < 4; ++j)
               acc[h] = __builtin_amdgcn_fdot2_f32_bf16(kp2[j], qp2[j],
                                                        acc[h], false);
-            }
           }
         }
       }
@@ -158,56 +173,47 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       const float csum = wave_sum(psum);
       const float alpha = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
       l[h] = l[h] * alpha + csum;
-      o0[h] *= alpha;
-      o1[h] *= alpha;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o[h][j] *= alpha;
       m[h] = mn;
     }
     __builtin_amdgcn_s_waitcnt(0);
 
-    // PV: lane = dim pair; V row loaded once, feeds all RATIO heads.
-    // 8-token blocks: preload 8 independent rows, then FMA.
+    // PV: per u-step this lane reads V[tok = base + u*4 + pv_tg]
+    // [dims pv_dg*8 .. +8) — the wave covers 4 adjacent token rows x
+    // full 256-B width = 1 KB contiguous; p broadcasts from LDS.
     const int base_tok = c * CHUNK;
-    int t = 0;
-    for (; t + 8 <= c_len; t += 8) {
-      float v0[8], v1[8];
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int gt = base_tok + t + u;
-        const long page = pt[gt / PS];
-        const short* vp = v_cache + (((long)page * n_kv + g) * PS + gt % PS) * D + d0;
-        v0[u] = bits2f(vp[0]);
-        v1[u] = bits2f(vp[1]);
-      }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-#pragma unroll
-        for (int h = 0; h < RATIO; ++h) {
-          const float pw = p_lds[h][t + u];
-          o0[h] = fmaf(pw, v0[u], o0[h]);
-          o1[h] = fmaf(pw, v1[u], o1[h]);
-        }
-      }
-    }
-    for (; t < c_len; ++t) {
+#pragma unroll 4
+    for (int u = 0; u < 16; ++u) {
+      const int t = u * 4 + pv_tg;
       const int gt = base_tok + t;
-      const long page = pt[gt / PS];
-      const short* vp = v_cache + (((long)page * n_kv + g) * PS + gt % PS) * D + d0;
-      const float v0 = bits2f(vp[0]), v1 = bits2f(vp[1]);
+      const int gt_c = min(gt, len - 1);
+      const long page = pt[gt_c / PS];
+      const short* vp = v_cache +
+          (((long)page * n_kv + g) * PS + gt_c % PS) * D + pv_dg * 8;
+      bf16x8 vv = *reinterpret_cast<const bf16x8*>(vp);
 #pragma unroll
       for (int h = 0; h < RATIO; ++h) {
-        const float pw = p_lds[h][t];
-        o0[h] = fmaf(pw, v0, o0[h]);
-        o1[h] = fmaf(pw, v1, o1[h]);
+        const float pw = p_lds[h][t];  // 0 beyond len
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[h][j] = fmaf(pw, bits2f(vv[j]), o[h][j]);
       }
     }
   }
 
-  // publish partials (unnormalized)
+  // publish partials (unnormalized): reduce o across the 4 token groups
+  // (lane bits 4..5), then pv_tg==0 lanes own dims [pv_dg*8, pv_dg*8+8)
 #pragma unroll
   for (int h = 0; h < RATIO; ++h) {
     float* pp = my_part + h * PART_STRIDE;
-    pp[2 * lane] = o0[h];
-    pp[2 * lane + 1] = o1[h];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float v = o[h][j];
+      v += __shfl_xor(v, 16, WAVE);
+      v += __shfl_xor(v, 32, WAVE);
+      if (pv_tg == 0) pp[pv_dg * 8 + j] = v;
+    }
     if (lane == 0) { pp[128] = m[h]; pp[129] = l[h]; }
   }
 }
