@@ -71,6 +71,13 @@ class KVCacheManager:
         # write their garbage K/V there (see llm.ModelInstance._device_decode)
         self._free: List[int] = list(range(n_pages - 1, 0, -1))
         self._seqs: Dict[str, Sequence] = {}
+        # copy-on-write refcounts for SHARED pages (prefix sharing). A page
+        # absent from the map has exactly one owner; adopt_prefix bumps the
+        # count, release decrements and only returns a page to the pool when
+        # its last owner lets go. Shared pages are never written: adopters
+        # take only FULL pages, and appends always land at slot >= length,
+        # which is past the shared region by construction.
+        self._refs: Dict[int, int] = {}
         self._lock = threading.RLock()
         # shorts per page per layer: K plane + V plane
         self.page_shorts = 2 * n_kv * head_dim * page_size
@@ -220,11 +227,51 @@ class KVCacheManager:
                 0, idx, torch.full((len(slots),), -1, dtype=torch.int32,
                                    device=self.device))
 
+    def adopt_prefix(self, seq_id: str, src_seq_id: str, n_tokens: int) -> None:
+        """Share the first n_tokens (a whole number of pages) of src with an
+        EMPTY sequence — copy-on-write prefix sharing for agents with a
+        common system prompt. The adopter's page table points at the shared
+        pages; its subsequent appends land past the shared region (length
+        starts at n_tokens, a page boundary), so the shared pages stay
+        read-only. Refcounts keep them alive until the last owner frees."""
+        with self._lock:
+            src = self._seqs[src_seq_id]
+            dst = self._seqs[seq_id]
+            if dst.length != 0 or dst.pages:
+                raise ValueError(f"adopt_prefix: {seq_id} is not empty")
+            if n_tokens % self.page_size != 0 or n_tokens <= 0:
+                raise ValueError("adopt_prefix: n_tokens must be whole pages")
+            n_pg = n_tokens // self.page_size
+            if src.length < n_tokens or len(src.pages) < n_pg:
+                raise ValueError("adopt_prefix: source prefix too short")
+            shared = src.pages[:n_pg]
+            for p in shared:
+                self._refs[p] = self._refs.get(p, 1) + 1
+            dst.pages = list(shared)
+            dst.length = n_tokens
+            if self.is_gpu:
+                slot = self._slot_of[seq_id]
+                self.dev_page_table[slot, :n_pg] = torch.tensor(
+                    shared, dtype=torch.int32, device=self.device)
+                self.dev_seq_lens[slot] = n_tokens
+
+    def _release_pages(self, pages: List[int]) -> None:
+        """Return pages to the pool, honoring shared-page refcounts. Caller
+        holds the lock."""
+        for p in reversed(pages):
+            c = self._refs.get(p)
+            if c is None:
+                self._free.append(p)       # sole owner — really free
+            elif c <= 2:
+                del self._refs[p]          # one owner left; unshared again
+            else:
+                self._refs[p] = c - 1
+
     def free_seq(self, seq_id: str) -> None:
         with self._lock:
             s = self._seqs.pop(seq_id, None)
             if s:
-                self._free.extend(reversed(s.pages))
+                self._release_pages(s.pages)
             slot = self._slot_of.pop(seq_id, None)
             if slot is not None:
                 self.dev_seq_lens[slot] = -1
@@ -234,7 +281,7 @@ class KVCacheManager:
         """Drop a sequence's KV but keep it registered (context truncation)."""
         with self._lock:
             s = self._seqs[seq_id]
-            self._free.extend(reversed(s.pages))
+            self._release_pages(s.pages)
             s.pages = []
             s.length = 0
             if self.is_gpu:
@@ -248,11 +295,13 @@ class KVCacheManager:
             return need_pages <= len(self._free)
 
     def can_append_after_reset(self, seq_id: str, n_tokens: int) -> bool:
-        """Would n_tokens fit if this sequence's pages were freed first?"""
+        """Would n_tokens fit if this sequence's pages were freed first?
+        Shared (refcounted) pages don't return to the pool on reset."""
         with self._lock:
             s = self._seqs[seq_id]
             need_pages = -(-n_tokens // self.page_size)
-            return need_pages <= len(self._free) + len(s.pages)
+            freeable = sum(1 for p in s.pages if p not in self._refs)
+            return need_pages <= len(self._free) + freeable
 
     def append_slots(self, seq_id: str, n_tokens: int) -> List[int]:
         """Reserve slots for n_tokens new tokens; allocates pages as needed.

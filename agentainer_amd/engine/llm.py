@@ -75,6 +75,9 @@ class AgentBinding:
     tokens: int = 0
     queue: "queue.Queue[GenRequest]" = field(default_factory=queue.Queue)
     active: Optional[GenRequest] = None
+    # whole-page token prefix shared with other agents (system prompt),
+    # None when the system prompt is shorter than one KV page
+    prefix_tokens: Optional[List[int]] = None
 
 
 class ModelInstance:
@@ -125,6 +128,12 @@ class ModelInstance:
         self.max_batch_tokens = int(engine_cfg.get("max_batch_tokens", 8192))
         self.max_decode_batch = int(engine_cfg.get("max_decode_batch", 256))
         self.refcount = 0
+        # prefix sharing (copy-on-write system prompts): token-tuple ->
+        # shared-prefix sequence id; populated lazily by the first prefill
+        # that carries a registered prefix (plan-driven, so TP ranks stay
+        # in lockstep)
+        self.prefix_sharing = bool(engine_cfg.get("prefix_sharing", True))
+        self._prefixes: Dict[tuple, str] = {}
         self.waiting: "queue.Queue[GenRequest]" = queue.Queue()
         self.running: List[GenRequest] = []
         self._running_set = set()
@@ -428,11 +437,36 @@ class ModelInstance:
 
     def _prefill(self, reqs: List[GenRequest]):
         plan = []
+        n_extra = 0  # shared-prefix rows prepended to the plan (no GenRequest)
         for r in reqs:
             b = self._bindings[r.agent_id]
-            plan.append((b.seq_id, r.prompt_tokens, bool(getattr(r, "needs_reset", False))))
+            needs_reset = bool(getattr(r, "needs_reset", False))
+            tokens = r.prompt_tokens
+            adopt = None
+            pk = b.prefix_tokens
+            if (pk is not None and not needs_reset
+                    and self.kvm.seq_len(b.seq_id) == 0
+                    and len(tokens) > len(pk) and tokens[:len(pk)] == pk):
+                key = tuple(pk)
+                sid = self._prefixes.get(key)
+                if sid is None:
+                    # first user pays: prefill the shared-prefix sequence as
+                    # an extra plan row ahead of its adopter (its K/V append
+                    # kernel runs before any attention in the same forward)
+                    n_pg = len(pk) // self.kvm.page_size
+                    if self.kvm.free_pages >= n_pg + 8:
+                        sid = f"\x00pfx:{len(self._prefixes)}"
+                        self._prefixes[key] = sid
+                        plan.insert(n_extra, (sid, list(pk), False, None))
+                        n_extra += 1
+                if sid is not None:
+                    adopt = (sid, len(pk))
+                    tokens = tokens[len(pk):]
+            plan.append((b.seq_id, tokens, needs_reset, adopt))
         self._bcast(("prefill", self.name, plan))
         logits = self._prefill_exec(plan)
+        if n_extra:
+            logits = logits[n_extra:]
         toks = self._sample(logits, reqs)
         now = time.time()
         with self._lock:
@@ -448,9 +482,13 @@ class ModelInstance:
         positions: List[int] = []
         slots: List[int] = []
         q_starts, q_lens, seq_ids = [], [], []
-        for seq_id, tokens, needs_reset in plan:
+        for seq_id, tokens, needs_reset, adopt in plan:
+            if not self.kvm.has_seq(seq_id):
+                self.kvm.create_seq(seq_id)  # lazily-created shared-prefix seq
             if needs_reset:
                 self.kvm.reset_seq(seq_id)
+            if adopt is not None and self.kvm.seq_len(seq_id) == 0:
+                self.kvm.adopt_prefix(seq_id, adopt[0], adopt[1])
             prev = self.kvm.seq_len(seq_id)
             n = len(tokens)
             q_starts.append(len(ids))
@@ -648,9 +686,16 @@ class ModelInstance:
             self._bcast(("bind", self.name, seq_id,
                          ckpt is not None or worker_has_ckpt))
             self._bind_seq(seq_id, ckpt)
-            self._bindings[agent.id] = AgentBinding(agent=agent,
-                                                    model_name=self.name,
-                                                    seq_id=seq_id)
+            b = AgentBinding(agent=agent, model_name=self.name, seq_id=seq_id)
+            sp = getattr(agent, "system_prompt", "") or ""
+            if self.prefix_sharing and sp:
+                # the whole-page head of this agent's first-turn prompt
+                # (matches _build_prompt's "[system] ...\n" framing)
+                pk = self.tokenizer.encode(f"[system] {sp}\n")
+                n_full = (len(pk) // self.kvm.page_size) * self.kvm.page_size
+                if n_full >= self.kvm.page_size:
+                    b.prefix_tokens = pk[:n_full]
+            self._bindings[agent.id] = b
             self.refcount += 1
 
     def unbind(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:

@@ -208,15 +208,27 @@ def test_tp2_matches_full_model(tmp_path):
     assert len(want) == MAX_NEW
 
     result_file = os.path.join(tmpdir, "result.pt")
-    port = _free_port()
     ctx = mp.get_context("spawn")
-    procs = [ctx.Process(target=_tp_worker, args=(r, 2, port, tmpdir, result_file))
-             for r in range(2)]
-    for p in procs:
-        p.start()
-    for p in procs:
-        p.join(timeout=240)
-        assert p.exitcode == 0, f"worker exit {p.exitcode}"
+    # one retry with a fresh port: the gloo rendezvous can lose the port to
+    # another process between _free_port() and rank 0's bind. Token
+    # mismatches below stay strict — only infra failures are retried.
+    for attempt in range(2):
+        port = _free_port()
+        procs = [ctx.Process(target=_tp_worker,
+                             args=(r, 2, port, tmpdir, result_file))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        codes = []
+        for p in procs:
+            p.join(timeout=240)
+            codes.append(p.exitcode)
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+        if all(c == 0 for c in codes):
+            break
+        assert attempt == 0, f"worker exits {codes} (after retry)"
     res = torch.load(result_file, weights_only=True)
     assert res["err"] is None and res["err2"] is None
     assert res["tokens"] == want, f"TP tokens {res['tokens']} != full {want}"
