@@ -112,6 +112,39 @@ def test_kv_checkpoint_roundtrip_gpu(gpu_rt):
     assert out_a == out_ctl  # restored KV produces identical continuation
 
 
+def test_prefix_sharing_gpu(gpu_rt):
+    """COW prefix sharing on device: an adopter generating off SHARED KV
+    pages must match an agent that prefilled the same tokens itself (the
+    decode graph reads the shared pages through the device page table)."""
+    engine, manager, _ = gpu_rt
+    sp = "You answer tersely and never speculate about anything at all."
+    a1 = manager.deploy(name="p1", model="tiny-llama", system_prompt=sp)
+    manager.start(a1.id)
+    a2 = manager.deploy(name="p2", model="tiny-llama", system_prompt=sp)
+    manager.start(a2.id)
+    ctl = manager.deploy(name="pc", model="tiny-llama")  # no system prompt
+    manager.start(ctl.id)
+    inst = engine._instances["tiny-llama"]
+    pk = inst.binding(a1.id).prefix_tokens
+    assert pk and len(pk) % inst.kvm.page_size == 0
+    prompt = pk + list(range(3, 40))
+    out1 = _gen(engine, manager, a1, prompt)   # first payer (prefix prefill)
+    assert len(inst._prefixes) == 1
+    free_before = inst.kvm.free_pages
+    out2 = _gen(engine, manager, a2, prompt)   # adopter: shared pages
+    assert inst.kvm._refs, "no pages shared"
+    n_pfx_pages = len(pk) // inst.kvm.page_size
+    # the adopter allocated fewer pages than an unshared prefill would
+    assert free_before - inst.kvm.free_pages < n_pfx_pages
+    outc = _gen(engine, manager, ctl, prompt)  # unshared control
+    assert out1 == out2 == outc
+    # adopters survive the prefix's other owners detaching
+    manager.stop(a1.id)
+    out2b = _gen(engine, manager, a2, list(range(40, 60)))
+    outcb = _gen(engine, manager, ctl, list(range(40, 60)))
+    assert out2b == outcb
+
+
 def test_mixtral_generation_gpu(gpu_rt):
     """tiny-mixtral end-to-end on the HIP kernels + MoE routing."""
     engine, manager, _ = gpu_rt
