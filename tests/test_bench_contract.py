@@ -1,0 +1,67 @@
+"""Driver-contract test for bench.py: the round-end harness runs
+`python bench.py --gpus N --steps K --warmup W` (N>1 via
+torch.distributed.run), parses ONE JSON line from rank 0 and computes
+scaling efficiency from per-N values. These tests pin that contract on
+CPU (gloo world 2) so the 8-GPU scaling run can't be broken by a repo
+change that only CPU CI sees."""
+
+import json
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+REQUIRED = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+            "dtype", "data", "config"}
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _last_json_line(stdout: str) -> dict:
+    lines = [ln for ln in stdout.strip().splitlines() if ln.startswith("{")]
+    assert lines, f"no JSON line in output: {stdout[-2000:]}"
+    return json.loads(lines[-1])
+
+
+@pytest.mark.timeout(300)
+def test_bench_single_process_contract():
+    r = subprocess.run([sys.executable, "bench.py", "--steps", "10",
+                        "--warmup", "2"],
+                       capture_output=True, text=True, cwd=ROOT, timeout=280)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = _last_json_line(r.stdout)
+    assert REQUIRED.issubset(out.keys())
+    assert out["metric"] == "concurrent_agent_chat_req_per_s"
+    assert out["scaling"] == "weak"
+    assert out["higher_is_better"] is True
+    assert out["ms_per_step"] > 0
+    assert out["config"]["parallelism"] == "dp1"
+
+
+@pytest.mark.timeout(300)
+def test_bench_world2_gloo_contract():
+    """Exactly the driver's multi-GPU launch shape, on CPU/gloo."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "2",
+         "--steps", "8", "--warmup", "2"],
+        capture_output=True, text=True, cwd=ROOT, timeout=280, env=env)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+    out = _last_json_line(r.stdout)
+    assert out["config"]["parallelism"] == "dp2"
+    # whole-job aggregate: both ranks' completions are summed
+    assert out["config"]["global_batch"] == 2 * out["config"]["agents_per_gpu"]
+    assert out["value"] > 0
