@@ -32,11 +32,12 @@ validation: name<=64, model<=256, env<=50 entries (server.go:157-179).
 
 from __future__ import annotations
 
+import json
 import time
 from typing import Any, Dict, Optional
 
 from fastapi import Depends, FastAPI, Header, Query, Request as HttpRequest
-from fastapi.responses import JSONResponse
+from fastapi.responses import JSONResponse, StreamingResponse
 from starlette.concurrency import run_in_threadpool
 
 from ..registry import AgentNotFound
@@ -112,6 +113,28 @@ def create_app(rt: Runtime) -> FastAPI:
                 body = {"message": raw.decode("utf-8", "replace")} if raw else {}
         replay = request.headers.get("x-agentainer-replay", "").lower() == "true"
         req_id = request.headers.get("x-agentainer-request-id") or None
+        # SSE streaming on the chat path: opt in with {"stream": true},
+        # ?stream=1 or Accept: text/event-stream
+        wants_stream = (
+            agent_path.strip("/") == "chat" and request.method == "POST"
+            and not replay
+            and (bool(isinstance(body, dict) and body.get("stream"))
+                 or request.query_params.get("stream") in ("1", "true")
+                 or "text/event-stream" in request.headers.get("accept", "")))
+        if wants_stream:
+            if isinstance(body, dict):
+                body = {k: v for k, v in body.items() if k != "stream"}
+            status, result = await run_in_threadpool(
+                rt.agent_request_stream, agent_id, body,
+                {"content-type": request.headers.get("content-type", "")})
+            if status != 200:
+                return JSONResponse(status_code=status, content=result)
+
+            def sse():  # sync generator — starlette iterates in threadpool
+                for ev in result:
+                    yield f"data: {json.dumps(ev)}\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
         # blocking dispatch (a generation can take seconds) runs in the
         # threadpool so concurrent requests reach the engine together —
         # that concurrency IS what feeds the continuous-batching scheduler

@@ -194,9 +194,28 @@ def logs(client: Client, agent_id: str, limit: int):
 @cli.command()
 @click.argument("agent_id")
 @click.option("--message", "-m", required=True)
+@click.option("--stream", is_flag=True, help="stream tokens as they decode (SSE)")
 @pass_client
-def invoke(client: Client, agent_id: str, message: str):
+def invoke(client: Client, agent_id: str, message: str, stream: bool):
     """Send a chat message to an agent (authenticated dispatch)."""
+    if stream:
+        with httpx.stream("POST", f"{client.url}/agent/{agent_id}/chat",
+                          json={"message": message, "stream": True},
+                          timeout=120.0) as r:
+            if r.status_code != 200:
+                click.echo(f"error ({r.status_code})", err=True)
+                sys.exit(1)
+            for line in r.iter_lines():
+                if not line.startswith("data: "):
+                    continue
+                ev = json.loads(line[len("data: "):])
+                if ev.get("done"):
+                    click.echo("")  # newline after the streamed text
+                elif "error" in ev:
+                    click.echo(f"\nerror: {ev['error']}", err=True)
+                else:
+                    click.echo(ev.get("text", ""), nl=False)
+        return
     resp = client.call("POST", f"/agents/{agent_id}/invoke",
                        {"path": "/chat", "method": "POST",
                         "body": {"message": message}})

@@ -104,6 +104,14 @@ class EchoEngine:
         self.store.hset(f"agent:{agent_id}:metrics", "total_requests", st["requests"])
         return {"response": reply, "model": "echo", "context_turns": len(context)}
 
+    def chat_stream(self, agent_id: str, message: str, **kwargs: Any):
+        """SSE analog: word-chunked stream of the echo reply, then the
+        final payload event (same contract as LLMEngine.chat_stream)."""
+        payload = self.chat(agent_id, message, **kwargs)
+        for word in payload["response"].split(" "):
+            yield {"token": None, "text": word + " "}
+        yield {"done": True, **payload}
+
     # ---------- stats / fault injection ----------
 
     def stats(self) -> Dict[str, Any]:

@@ -63,6 +63,9 @@ class GenRequest:
     first_token_t: float = 0.0
     fin_t: float = 0.0
     trace_id: str = ""  # WAL request id — end-to-end tracing
+    # streaming: the scheduler puts each sampled token id here as it is
+    # produced, then None when the request finishes (SSE /chat path)
+    stream_q: Optional[Any] = None
 
 
 @dataclass
@@ -633,6 +636,10 @@ class ModelInstance:
         """Called with lock held, after appending tok."""
         finished = (tok == self.tokenizer.eos_id or
                     len(r.generated) >= r.max_new)
+        if r.stream_q is not None:
+            r.stream_q.put(tok)
+            if finished:
+                r.stream_q.put(None)
         if r in self._running_set:
             if finished:
                 self.running.remove(r)
@@ -897,14 +904,13 @@ class LLMEngine:
 
     # ---------- chat ----------
 
-    def chat(self, agent_id: str, message: str, **kwargs: Any) -> Dict[str, Any]:
+    def _submit_chat(self, agent_id: str, message: str, kwargs: Dict[str, Any],
+                     stream: bool = False):
+        """Shared head of chat/chat_stream: validate, build + enqueue the
+        GenRequest. Returns (inst, binding, req, timeout_s)."""
         from ..wal import EngineUnavailable
 
         model = self._agent_model.get(agent_id)
-        if model is None:
-            raise EngineUnavailable(f"agent {agent_id} is not attached")
-        if model == "echo":
-            return self._echo_chat(agent_id, message)
         inst = self._instances.get(model)
         b = inst.binding(agent_id) if inst else None
         if inst is None or b is None or not inst.alive():
@@ -925,14 +931,16 @@ class LLMEngine:
             seed=int(sampling.get("seed", 0)),
             trace_id=trace_id,
         )
+        if stream:
+            req.stream_q = queue.Queue()
         with inst._lock:
             b.queue.put(req)
             inst._pump_agent(b)
-        timeout = float(sampling.get("timeout_s", 120.0))
-        if not req.done.wait(timeout):
-            raise EngineUnavailable(f"generation timed out after {timeout}s")
-        if req.error:
-            raise EngineUnavailable(req.error)
+        return inst, b, req, float(sampling.get("timeout_s", 120.0))
+
+    def _chat_payload(self, agent_id: str, model: str, inst, b,
+                      req: GenRequest, message: str) -> Dict[str, Any]:
+        """Shared tail: history + metrics writes, response envelope."""
         text = inst.tokenizer.decode(req.generated)
         hist_key = f"agent:{agent_id}:conversations"
         self.store.rpush(hist_key, {"user": message, "assistant": text,
@@ -949,6 +957,68 @@ class LLMEngine:
             "ttft_s": (req.first_token_t - req.enq_t) if req.first_token_t else None,
             "e2e_s": (req.fin_t - req.enq_t) if req.fin_t else None,
         }
+
+    def chat(self, agent_id: str, message: str, **kwargs: Any) -> Dict[str, Any]:
+        from ..wal import EngineUnavailable
+
+        model = self._agent_model.get(agent_id)
+        if model is None:
+            raise EngineUnavailable(f"agent {agent_id} is not attached")
+        if model == "echo":
+            return self._echo_chat(agent_id, message)
+        inst, b, req, timeout = self._submit_chat(agent_id, message, kwargs)
+        if not req.done.wait(timeout):
+            raise EngineUnavailable(f"generation timed out after {timeout}s")
+        if req.error:
+            raise EngineUnavailable(req.error)
+        return self._chat_payload(agent_id, model, inst, b, req, message)
+
+    def chat_stream(self, agent_id: str, message: str, **kwargs: Any):
+        """Streaming chat: yields {"token", "text"} events as the scheduler
+        produces tokens, then one final {"done": True, **payload} event —
+        the SSE data frames of the /chat?stream path. Token->text deltas
+        re-decode the full id list each step so multi-byte UTF-8 sequences
+        only surface once complete."""
+        from ..wal import EngineUnavailable
+
+        model = self._agent_model.get(agent_id)
+        if model is None:
+            raise EngineUnavailable(f"agent {agent_id} is not attached")
+        if model == "echo":
+            payload = self._echo_chat(agent_id, message)
+            for word in payload["response"].split(" "):
+                yield {"token": None, "text": word + " "}
+            yield {"done": True, **payload}
+            return
+        import codecs
+
+        inst, b, req, timeout = self._submit_chat(agent_id, message, kwargs,
+                                                  stream=True)
+        deadline = time.time() + timeout
+        dec = codecs.getincrementaldecoder("utf-8")("replace")
+        while True:
+            try:
+                tok = req.stream_q.get(timeout=0.1)
+            except queue.Empty:
+                # fallback for completions that bypass _finish_or_run
+                # (detach/unbind drain paths set done+error directly)
+                if req.done.is_set() and req.stream_q.empty():
+                    break
+                if time.time() > deadline:
+                    raise EngineUnavailable(
+                        f"generation timed out after {timeout}s")
+                continue
+            if tok is None:
+                break
+            yield {"token": int(tok),
+                   "text": dec.decode(inst.tokenizer.id_bytes(int(tok)))}
+        if req.error:
+            raise EngineUnavailable(req.error)
+        tail = dec.decode(b"", True)  # flush a dangling partial sequence
+        if tail:
+            yield {"token": None, "text": tail}
+        yield {"done": True,
+               **self._chat_payload(agent_id, model, inst, b, req, message)}
 
     def _build_prompt(self, agent, message: str) -> str:
         """This turn's incremental prompt; earlier turns are already in the

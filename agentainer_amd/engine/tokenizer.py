@@ -29,3 +29,9 @@ class ByteTokenizer:
     def decode(self, ids: List[int]) -> str:
         bs = bytes(i - OFFSET for i in ids if OFFSET <= i < OFFSET + 256)
         return bs.decode("utf-8", "replace")
+
+    def id_bytes(self, i: int) -> bytes:
+        """Raw bytes of one token (b'' for specials) — feeds the streaming
+        path's incremental UTF-8 decoder so multi-byte sequences are only
+        emitted once complete."""
+        return bytes([i - OFFSET]) if OFFSET <= i < OFFSET + 256 else b""

@@ -111,31 +111,7 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
   }
   const int* pt = page_table + (long)b * max_pages;
 
-  // V staging for the chunk's first 32 tokens: async global->LDS DMA
-  // (no VGPRs held) issued ALONGSIDE the K loads, so the K and V HBM
-  // round trips overlap — at bench contexts (<=512) each wave owns one
-  // chunk and the old K->score->V chain paid the HBM latency twice.
-  // 8 KB/wave keeps the LDS-per-CU occupancy at the VGPR limit.
-  __shared__ short v_lds[32 * D];
-
   for (int c = c0; c < c1; ++c) {
-    {
-      const int tg4 = lane >> 4, dg16 = lane & 15;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int gt = c * CHUNK + i * 4 + tg4;
-        const int gt_c = min(gt, len - 1);
-        const long page = pt[gt_c / PS];
-        const short* vp = v_cache +
-            (((long)page * n_kv + g) * PS + gt_c % PS) * D + dg16 * 8;
-        // lane l deposits 16 B at v_lds + i*1024 + l*16: token-major
-        // [tok][D] with tok = i*4 + l/16, dims (l%16)*8..+8
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) void*)vp,
-            (__attribute__((address_space(3))) void*)&v_lds[i * 4 * D],
-            16, 0, 0);
-      }
-    }
     float s[4][RATIO];  // [16-token sub-pass][head]
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
@@ -206,23 +182,9 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
     // PV: per u-step this lane reads V[tok = base + u*4 + pv_tg]
     // [dims pv_dg*8 .. +8) — the wave covers 4 adjacent token rows x
     // full 256-B width = 1 KB contiguous; p broadcasts from LDS.
-    // Tokens 0..31 come from the prefetched LDS image (the s_waitcnt(0)
-    // above drained the DMA); 32..63 stream from HBM as before.
     const int base_tok = c * CHUNK;
 #pragma unroll 4
-    for (int u = 0; u < 8; ++u) {
-      const int t = u * 4 + pv_tg;
-      bf16x8 vv = *reinterpret_cast<const bf16x8*>(&v_lds[t * D + pv_dg * 8]);
-#pragma unroll
-      for (int h = 0; h < RATIO; ++h) {
-        const float pw = p_lds[h][t];  // 0 beyond len
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          o[h][j] = fmaf(pw, bits2f(vv[j]), o[h][j]);
-      }
-    }
-#pragma unroll 4
-    for (int u = 8; u < 16; ++u) {
+    for (int u = 0; u < 16; ++u) {
       const int t = u * 4 + pv_tg;
       const int gt = base_tok + t;
       const int gt_c = min(gt, len - 1);

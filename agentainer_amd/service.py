@@ -181,6 +181,67 @@ class Runtime:
                                      if isinstance(payload, dict) else 0)
         return 200, payload
 
+    def agent_request_stream(self, agent_id: str, body: Any = None,
+                             headers: Optional[Dict[str, str]] = None):
+        """Streaming variant of the /chat hot path (SSE). Returns
+        (status, payload) for non-streamable outcomes (404/202), else
+        (200, generator) — the generator yields event dicts and performs
+        the WAL ack + metrics write when the generation completes. A
+        stream cut by an engine crash leaves the WAL entry PENDING, so
+        the replay worker regenerates it (same at-least-once contract as
+        the blocking path)."""
+        agent = self.agents.try_get(agent_id)
+        if agent is None:
+            return 404, {"success": False, "message": f"agent {agent_id} not found"}
+        req: Optional[Request] = None
+        if self.persistence_enabled:
+            req = self.requests.store_request(agent_id, "POST", "/chat",
+                                              headers=headers, body=body)
+        if agent.status != RUNNING:
+            return 202, {
+                "success": True,
+                "message": "agent not running; request queued",
+                "data": {"request_id": req.id if req else None, "status": "pending"},
+            }
+        message = body.get("message", "") if isinstance(body, dict) else str(body)
+        kwargs = dict(body.get("sampling", {})) if isinstance(body, dict) else {}
+        if req is not None:
+            kwargs["trace_id"] = req.id
+
+        def run():
+            t0 = time.time()
+            if req is not None:
+                self._inflight.add(req.id)
+            final = None
+            try:
+                for ev in self.engine.chat_stream(agent.id, message, **kwargs):
+                    if ev.get("done"):
+                        final = ev
+                    yield ev
+            except EngineUnavailable as exc:
+                yield {"error": f"agent unavailable: {exc}",
+                       "request_id": req.id if req else None,
+                       "status": "pending"}
+                return
+            except Exception as exc:  # noqa: BLE001
+                if req is not None:
+                    self.requests.mark_failed(agent_id, req.id, str(exc))
+                yield {"error": str(exc),
+                       "request_id": req.id if req else None}
+                return
+            finally:
+                if req is not None:
+                    self._inflight.discard(req.id)
+            if final is not None:
+                payload = {k: v for k, v in final.items() if k != "done"}
+                if req is not None:
+                    self.requests.store_response(agent_id, req.id, payload)
+                self.metrics.observe_request(
+                    agent_id, time.time() - t0,
+                    tokens=payload.get("tokens", 0) or 0)
+
+        return 200, run()
+
     def _dispatch_replay(self, agent_id: str, req: Request, replay: bool = True) -> Any:
         """ReplayWorker dispatch: direct engine path, response stored by the
         worker (single store — quirk fix vs replay_worker.go:158)."""

@@ -160,3 +160,40 @@ def test_backup_endpoints(client):
     rr = client.post(f"/backups/{bid}/restore", headers=AUTH)
     assert rr.status_code == 200 and rr.json()["data"][0]["name"] == "b1-restored"
     assert client.delete(f"/backups/{bid}", headers=AUTH).status_code == 200
+
+
+def test_chat_sse_streaming(client):
+    """SSE streaming on the chat proxy: token events then a final done
+    event matching the blocking /chat contract; WAL-acked on completion."""
+    import json as _json
+
+    a = _deploy(client, name="sse1")
+    client.post(f"/agents/{a['id']}/start", headers=AUTH)
+    with client.stream("POST", f"/agent/{a['id']}/chat",
+                       json={"message": "stream me", "stream": True}) as r:
+        assert r.status_code == 200
+        assert r.headers["content-type"].startswith("text/event-stream")
+        events = []
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                events.append(_json.loads(line[len("data: "):]))
+    assert len(events) >= 2
+    final = events[-1]
+    assert final.get("done") is True
+    assert "stream me" in final["response"]
+    # chunk texts concatenate to the full response
+    text = "".join(e["text"] for e in events[:-1])
+    assert final["response"] in text or text.strip() == final["response"]
+    # the request was WAL-acked as completed
+    rq = client.get(f"/agents/{a['id']}/requests", headers=AUTH).json()["data"]
+    assert len(rq["completed"]) == 1
+    assert rq["completed"][0]["response"]["response"] == final["response"]
+    assert rq["pending"] == []
+
+
+def test_chat_sse_not_running_queues(client):
+    a = _deploy(client, name="sse2")  # never started
+    r = client.post(f"/agent/{a['id']}/chat?stream=1",
+                    json={"message": "queued"})
+    assert r.status_code == 202
+    assert r.json()["data"]["status"] == "pending"

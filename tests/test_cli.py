@@ -45,6 +45,9 @@ def test_cli_full_surface(run):
     out = run("invoke", aid, "-m", "ping from cli")
     assert "ping from cli" in out
 
+    out = run("invoke", aid, "-m", "streamed ping", "--stream")
+    assert "streamed ping" in out
+
     run("pause", aid)
     run("resume", aid)
     run("restart", aid)

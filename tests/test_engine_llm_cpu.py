@@ -177,3 +177,35 @@ def test_engine_stats_shape(llm_runtime):
     assert st["agents"][a.id]["kv_pages"] > 0
     assert "tiny-llama" in st["models"]
     assert st["models"]["tiny-llama"]["decode_tokens"] > 0
+
+
+def test_chat_stream_matches_blocking(llm_runtime):
+    """Token-streamed chat produces the same text as the blocking path
+    (greedy decode, fresh agents, same prompt)."""
+    rt = llm_runtime
+    a = _mk_agent(rt, name="st-a")
+    b = _mk_agent(rt, name="st-b")
+    blocking = rt.engine.chat(a.id, "stream parity")
+    events = list(rt.engine.chat_stream(b.id, "stream parity"))
+    final = events[-1]
+    assert final.get("done") is True
+    assert final["response"] == blocking["response"]
+    assert final["tokens"] == blocking["tokens"]
+    toks = [e for e in events[:-1] if e.get("token") is not None]
+    assert len(toks) == blocking["tokens"]
+    assert "".join(e["text"] for e in events[:-1]) == blocking["response"]
+    # history recorded for the streamed turn too
+    hist = rt.store.lrange(f"agent:{b.id}:conversations")
+    assert len(hist) == 1 and hist[0]["assistant"] == final["response"]
+
+
+def test_runtime_stream_wal_ack(llm_runtime):
+    rt = llm_runtime
+    a = _mk_agent(rt, name="st-wal")
+    status, gen = rt.agent_request_stream(a.id, {"message": "ack me"})
+    assert status == 200
+    events = list(gen)
+    assert events[-1].get("done") is True
+    done = rt.requests.by_queue(a.id, "completed")
+    assert len(done) == 1
+    assert done[0].response["response"] == events[-1]["response"]

@@ -68,5 +68,22 @@ def test_server_llama8b_concurrent_chat(tmp_path):
         assert st == 200
         assert m["data"]["models"]["llama3-8b"]["decode_tokens"] > 0
         assert m["data"].get("hbm_total_bytes", 0) > 0
+
+        # SSE streaming through the full GPU stack (graph-captured decode
+        # feeding per-token events): 32 token frames + final done frame
+        import json as _json
+
+        import httpx
+        events = []
+        with httpx.stream("POST", f"{srv.base}/agent/{aid}/chat",
+                          json={"message": "stream on gpu", "stream": True},
+                          timeout=300) as r:
+            assert r.status_code == 200
+            for line in r.iter_lines():
+                if line.startswith("data: "):
+                    events.append(_json.loads(line[len("data: "):]))
+        assert events[-1].get("done") is True, events[-1:]
+        assert events[-1]["tokens"] == 32
+        assert sum(1 for e in events[:-1] if e.get("token") is not None) == 32
     finally:
         srv.terminate()
