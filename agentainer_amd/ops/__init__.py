@@ -166,10 +166,16 @@ _ws_cache = {}
 _EMPTY_WS = {}
 SKINNY_DISABLED = os.environ.get("AGENTAINER_DISABLE_SKINNY", "0") == "1"
 SKINNY_FORCED = os.environ.get("AGENTAINER_FORCE_SKINNY", "0") == "1"
-# shapes where the hand-written skinny GEMM beat hipBLASLt on MI355X
-# (measured via tools/gemm_probe.py; see profiles/)
-_SKINNY_SHAPES = {(4096, 4096)}
+# shapes where the hand-written skinny GEMM beat hipBLASLt on MI355X,
+# with the (split_k, k_chunk) that won the tools/sg_sweep.py grid
+# (M=64; see profiles/README.md). Library keeps qkv/gate_up: at
+# N>=6144 its MT256 tiles already stream near the roofline.
+_SKINNY_SHAPES = {
+    (4096, 4096): (4, 128),    # o-proj: 14.2 us vs lib 18.6
+    (4096, 14336): (8, 128),   # down-proj: 30.3 us vs lib 34.4
+}
 SKINNY_NT = os.environ.get("AGENTAINER_SKINNY_NT", "0") == "1"
+SKINNY_KC = int(os.environ.get("AGENTAINER_SKINNY_KC", "0"))  # 0 = tuned
 
 
 def _skinny_split(ntiles: int, K: int) -> int:
@@ -214,16 +220,20 @@ def linear(x: torch.Tensor, w: torch.Tensor,
     hipBLASLt via F.linear."""
     M, K = x.shape
     N = w.size(0)
-    skinny_wins = (N, K) in _SKINNY_SHAPES or SKINNY_FORCED
+    tuned = _SKINNY_SHAPES.get((N, K))
     if (x.is_cuda and M <= 64 and N % 64 == 0 and K % 256 == 0
-            and skinny_wins and not SKINNY_DISABLED):
+            and (tuned is not None or SKINNY_FORCED) and not SKINNY_DISABLED):
         mod = _dispatch("skinny_gemm", x)
-        split = _skinny_split(N // 64, K)
+        split, kc = tuned if tuned else (_skinny_split(N // 64, K), 128)
+        if SKINNY_KC:
+            kc = SKINNY_KC
+        if K % (kc * split):
+            split = _skinny_split(N // 64, K)
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
         ws = _skinny_ws(x.device, N, split)
         if w_packed is not None:
             mod.skinny_gemm_packed(out, x.contiguous(), w_packed, N, K, ws,
-                                   split, SKINNY_NT)
+                                   split, SKINNY_NT, kc)
         else:
             mod.skinny_gemm(out, x.contiguous(), w, ws, split)
         return out

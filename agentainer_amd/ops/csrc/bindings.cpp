@@ -31,7 +31,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws, long split);
 void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
                         torch::Tensor w_packed, long N, long K,
-                        torch::Tensor ws, long split, bool nt);
+                        torch::Tensor ws, long split, bool nt, long kc);
 void quant_fp8_rows(torch::Tensor x8, torch::Tensor sx, torch::Tensor x);
 void skinny_gemm_fp8(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
                      torch::Tensor w_packed, torch::Tensor sw, long N, long K,

@@ -187,7 +187,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
 
 namespace {
 
-template <bool SPLIT, bool FULL, bool NT>
+template <bool SPLIT, bool FULL, bool NT, int KC>
 __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     void* __restrict__ out, const short* __restrict__ x,
     const short* __restrict__ wp_packed, int M, int N, int K,
@@ -208,7 +208,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   const short* wp = wp_packed + ((long)n_tile * (K / 32) + k0 / 32) * 512 +
                     lane * 8;
 
-  constexpr int KC = 256;
+  // KC=256: deep per-wave pipeline (8 W frags in flight) at 67.6 KB LDS
+  // (1-2 blocks/CU). KC=128: half the LDS and W registers so 2x the
+  // blocks/CU carry the HBM latency instead of per-wave depth.
+  constexpr int FR = KC / 32;  // W fragments / x vectors per chunk
   constexpr int XS = KC + 8;
   __shared__ short x_lds[2][64 * XS];
   const int s_row = threadIdx.x % 64;
@@ -216,10 +219,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   const short* s_xp = x + (long)(FULL ? s_row : min(s_row, M - 1)) * K;
   const bool s_alive = FULL || s_row < M;
 
-  bf16x8 st[8];
+  bf16x8 st[FR];
   auto stage_load = [&](int k) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < FR; ++i) {
       st[i] = bf16x8{};
       if (s_alive)
         st[i] = *reinterpret_cast<const bf16x8*>(s_xp + k + s_col0 + 32 * i);
@@ -227,7 +230,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
+    for (int i = 0; i < FR; ++i)
       *reinterpret_cast<bf16x8*>(
           &x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
   };
@@ -236,18 +239,18 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   // t+1's eight nt loads are issued while chunk t computes, so no chunk
   // pays the ~900-cycle HBM latency cold (the barrier would otherwise
   // serialize it at 1-2 blocks/CU).
-  bf16v8 bw_cur[8], bw_nxt[8];
-  auto w_load = [&](bf16v8 (&dst)[8], long woff) {
+  bf16v8 bw_cur[FR], bw_nxt[FR];
+  auto w_load = [&](bf16v8 (&dst)[FR], long woff) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < FR; ++i) {
       const bf16v8* p = reinterpret_cast<const bf16v8*>(wp + woff + (long)i * 512);
       dst[i] = NT ? __builtin_nontemporal_load(p) : *p;
     }
   };
 
-  auto compute = [&](bf16v8 (&bw)[8], int buf) {
+  auto compute = [&](bf16v8 (&bw)[FR], int buf) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < FR; ++i) {
       bf16v8 a[4];
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms)
@@ -267,14 +270,14 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   // ping-pong 2x unroll: no bw register copies, so chunk t+1's weight
   // loads never drain at a chunk boundary (a copy would force vmcnt(0))
   int buf = 0;
-  long woff = 8 * 512;
+  long woff = (long)FR * 512;
   int k = k0;
   while (true) {
     bool has_next = (k + KC < k1);
     if (has_next) {
       stage_load(k + KC);
       w_load(bw_nxt, woff);
-      woff += 8 * 512;
+      woff += (long)FR * 512;
     }
     compute(bw_cur, buf);
     if (has_next) stage_write(buf ^ 1);
@@ -286,7 +289,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     if (has_next) {
       stage_load(k + KC);
       w_load(bw_cur, woff);
-      woff += 8 * 512;
+      woff += (long)FR * 512;
     }
     compute(bw_nxt, buf);
     if (has_next) stage_write(buf ^ 1);
@@ -318,7 +321,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
 
 void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
                         torch::Tensor w_packed, long N, long K,
-                        torch::Tensor ws, long split, bool nt) {
+                        torch::Tensor ws, long split, bool nt, long kc) {
   TORCH_CHECK(x.is_contiguous() && w_packed.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               w_packed.scalar_type() == at::kBFloat16);
@@ -331,17 +334,24 @@ void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
   const int ntiles = (int)N / 64;
   const int kps = (int)K / (int)split;
   const bool full = (M == 64);
-#define SGP_LAUNCH(SPLIT_, FULL_, NT_, OUTP)                                   \
-  hipLaunchKernelGGL((skinny_gemm_packed_kernel<SPLIT_, FULL_, NT_>),          \
+  TORCH_CHECK(kc == 128 || kc == 256, "kc must be 128 or 256");
+  TORCH_CHECK((K % (kc * split)) == 0, "K must be kc*split aligned");
+#define SGP_LAUNCH(SPLIT_, FULL_, NT_, KC_, OUTP)                              \
+  hipLaunchKernelGGL((skinny_gemm_packed_kernel<SPLIT_, FULL_, NT_, KC_>),     \
                      dim3(ntiles, SPLIT_ ? (int)split : 1), dim3(256), 0,      \
                      stream, OUTP, (const short*)x.data_ptr(),                 \
                      (const short*)w_packed.data_ptr(), M, (int)N, (int)K, kps)
+#define SGP_KC(SPLIT_, FULL_, NT_, OUTP)                                       \
+  do {                                                                         \
+    if (kc == 128) SGP_LAUNCH(SPLIT_, FULL_, NT_, 128, OUTP);                  \
+    else SGP_LAUNCH(SPLIT_, FULL_, NT_, 256, OUTP);                            \
+  } while (0)
 #define SGP_DISPATCH(SPLIT_, OUTP)                                             \
   do {                                                                         \
-    if (full && nt) SGP_LAUNCH(SPLIT_, true, true, OUTP);                      \
-    else if (full) SGP_LAUNCH(SPLIT_, true, false, OUTP);                      \
-    else if (nt) SGP_LAUNCH(SPLIT_, false, true, OUTP);                        \
-    else SGP_LAUNCH(SPLIT_, false, false, OUTP);                               \
+    if (full && nt) SGP_KC(SPLIT_, true, true, OUTP);                          \
+    else if (full) SGP_KC(SPLIT_, true, false, OUTP);                          \
+    else if (nt) SGP_KC(SPLIT_, false, true, OUTP);                            \
+    else SGP_KC(SPLIT_, false, false, OUTP);                                   \
   } while (0)
   if (split == 1) {
     SGP_DISPATCH(false, out.data_ptr());

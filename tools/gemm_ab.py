@@ -25,7 +25,7 @@ for K, N in [(4096, 6144), (4096, 4096), (4096, 28672), (14336, 4096),
     split = O._skinny_split(N // 64, K)
     ws = O._skinny_ws(x.device, N, split)
     out = torch.empty(64, N, dtype=torch.bfloat16, device="cuda")
-    d_sk = t(lambda: mod.skinny_gemm_packed(out, x, wp, N, K, ws, split, False))
+    d_sk = t(lambda: mod.skinny_gemm_packed(out, x, wp, N, K, ws, split, False, 256))
     d_bl = t(lambda: F.linear(x, w))
     win = "skinny" if d_sk < d_bl else "blaslt"
     print(f"K={K:6d} N={N:6d} split={split}: skinny {d_sk*1e6:7.1f}us "

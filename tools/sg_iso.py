@@ -10,5 +10,5 @@ x = torch.randn(64, K, dtype=torch.bfloat16, device="cuda")
 out = torch.empty(64, N, dtype=torch.bfloat16, device="cuda")
 ws = O._skinny_ws(x.device, N, split)
 for _ in range(30):
-    mod.skinny_gemm_packed(out, x, wp, N, K, ws, split, False)
+    mod.skinny_gemm_packed(out, x, wp, N, K, ws, split, False, 256)
 torch.cuda.synchronize()

@@ -35,7 +35,7 @@ for K in [512, 1024, 2048, 4096, 8192]:
     x = torch.randn(64, K, dtype=torch.bfloat16, device="cuda")
     out = torch.empty(64, 4096, dtype=torch.bfloat16, device="cuda")
     ws = O._skinny_ws(x.device, 4096, 1)
-    d = t(lambda: mod.skinny_gemm_packed(out, x, wp, 4096, K, ws, 1, False))
+    d = t(lambda: mod.skinny_gemm_packed(out, x, wp, 4096, K, ws, 1, False, 256))
     print(f"  K={K:6d}: {d*1e6:8.1f} us  {4096*K*2/d/1e12:5.2f} TB/s")
 
 print("== skinny split sweep (N=4096, K=4096)")
@@ -47,7 +47,7 @@ for split in [1, 2, 4, 8, 16]:
     if 4096 % (256 * split):
         continue
     ws = O._skinny_ws(x.device, 4096, split)
-    d = t(lambda: mod.skinny_gemm_packed(out, x, wp, 4096, 4096, ws, split, False))
+    d = t(lambda: mod.skinny_gemm_packed(out, x, wp, 4096, 4096, ws, split, False, 256))
     print(f"  split={split:2d} ({4096//64*split:4d} blocks): {d*1e6:8.1f} us  "
           f"{4096*4096*2/d/1e12:5.2f} TB/s")
 
@@ -57,7 +57,7 @@ wps = [pack_weight(torch.randn(4096, 4096, dtype=torch.bfloat16, device="cuda"))
        for _ in range(16)]
 i = [0]
 def rot():
-    mod.skinny_gemm_packed(out, x, wps[i[0] % 16], 4096, 4096, ws, 8, False)
+    mod.skinny_gemm_packed(out, x, wps[i[0] % 16], 4096, 4096, ws, 8, False, 256)
     i[0] += 1
 d = t(rot)
 print(f"  rotating 16 buffers: {d*1e6:8.1f} us  {4096*4096*2/d/1e12:5.2f} TB/s")
