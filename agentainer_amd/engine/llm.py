@@ -92,7 +92,13 @@ class ModelInstance:
         self.name = name
         self.cfg = cfg
         self.device = device
-        self.tokenizer = ByteTokenizer(cfg.vocab_size)
+        tok_json = (os.path.join(weights_path, "tokenizer.json")
+                    if weights_path else None)
+        if tok_json and os.path.exists(tok_json):
+            from .tokenizer import HFTokenizer
+            self.tokenizer = HFTokenizer.from_checkpoint(weights_path)
+        else:
+            self.tokenizer = ByteTokenizer(cfg.vocab_size)
         # tensor parallelism: engine-level degree must match the launched
         # world size; rank 0 schedules, plans broadcast (parallel.dist)
         self.tp_size = int(engine_cfg.get("tp_degree", 1))
@@ -793,6 +799,14 @@ class LLMEngine:
                                  self.engine_cfg)
         if model in MIXTRAL_CONFIGS:
             return make_mixtral_instance(model, self.device, self.engine_cfg)
+        path = os.path.expanduser(model)
+        if os.path.isdir(path) and os.path.exists(
+                os.path.join(path, "config.json")):
+            # weights-path deploy: HF checkpoint directory (config.json +
+            # *.safetensors [+ tokenizer.json]) — the "image" analog
+            from ..models.llama import config_from_hf
+            return ModelInstance(model, config_from_hf(path), self.device,
+                                 self.engine_cfg, weights_path=path)
         raise ModelNotFound(f"cannot load {model!r}")
 
     def _get_instance(self, model: str) -> ModelInstance:
@@ -990,12 +1004,10 @@ class LLMEngine:
                 yield {"token": None, "text": word + " "}
             yield {"done": True, **payload}
             return
-        import codecs
-
         inst, b, req, timeout = self._submit_chat(agent_id, message, kwargs,
                                                   stream=True)
         deadline = time.time() + timeout
-        dec = codecs.getincrementaldecoder("utf-8")("replace")
+        dec = inst.tokenizer.stream_decoder()
         while True:
             try:
                 tok = req.stream_q.get(timeout=0.1)
@@ -1010,11 +1022,10 @@ class LLMEngine:
                 continue
             if tok is None:
                 break
-            yield {"token": int(tok),
-                   "text": dec.decode(inst.tokenizer.id_bytes(int(tok)))}
+            yield {"token": int(tok), "text": dec.feed(int(tok))}
         if req.error:
             raise EngineUnavailable(req.error)
-        tail = dec.decode(b"", True)  # flush a dangling partial sequence
+        tail = dec.flush()  # a dangling partial byte sequence
         if tail:
             yield {"token": None, "text": tail}
         yield {"done": True,

@@ -270,6 +270,42 @@ class LlamaForCausalLM(torch.nn.Module):
             layer.post_ln.copy_(t(pfx + "post_attention_layernorm.weight"))
 
 
+def config_from_hf(path: str) -> LlamaConfig:
+    """Build a LlamaConfig from an HF checkpoint directory's config.json
+    (weights-path deploys — the reference's "image" analog is a local
+    model directory, SURVEY.md §2.1 docker-client row)."""
+    import json
+    import os
+
+    with open(os.path.join(path, "config.json")) as f:
+        hf = json.load(f)
+    mt = hf.get("model_type", "")
+    archs = hf.get("architectures", [])
+    if mt not in ("llama", "") and not any("Llama" in a for a in archs):
+        raise ValueError(f"unsupported model_type {mt!r} at {path} "
+                         "(llama-family only)")
+    n_heads = int(hf["num_attention_heads"])
+    hidden = int(hf["hidden_size"])
+    head_dim = int(hf.get("head_dim") or hidden // n_heads)
+    if head_dim != 128:
+        raise ValueError(f"head_dim {head_dim} unsupported (kernels are "
+                         "specialized for 128)")
+    return LlamaConfig(
+        name=path,
+        vocab_size=int(hf["vocab_size"]),
+        hidden_size=hidden,
+        n_layers=int(hf["num_hidden_layers"]),
+        n_heads=n_heads,
+        n_kv_heads=int(hf.get("num_key_value_heads") or n_heads),
+        head_dim=head_dim,
+        intermediate_size=int(hf["intermediate_size"]),
+        rope_theta=float(hf.get("rope_theta", 500000.0)),
+        max_position=int(hf.get("max_position_embeddings", 8192)),
+        norm_eps=float(hf.get("rms_norm_eps", 1e-5)),
+        tie_embeddings=bool(hf.get("tie_word_embeddings", False)),
+    )
+
+
 # ---------------- model registry ----------------
 
 LLAMA_CONFIGS: Dict[str, LlamaConfig] = {
