@@ -66,6 +66,10 @@ class GenRequest:
     # streaming: the scheduler puts each sampled token id here as it is
     # produced, then None when the request finishes (SSE /chat path)
     stream_q: Optional[Any] = None
+    # chunked prefill: tokens of the prompt already written to KV, and the
+    # length of the slice admitted for the CURRENT step (set by _admit)
+    prefill_pos: int = 0
+    slice_len: int = 0
 
 
 @dataclass
@@ -144,6 +148,8 @@ class ModelInstance:
         self.prefix_sharing = bool(engine_cfg.get("prefix_sharing", True))
         self._prefixes: Dict[tuple, str] = {}
         self.waiting: "queue.Queue[GenRequest]" = queue.Queue()
+        # long prompts mid-chunked-prefill: continue before new admissions
+        self._chunking: List[GenRequest] = []
         self.running: List[GenRequest] = []
         self._running_set = set()
         self._bindings: Dict[str, AgentBinding] = {}
@@ -375,6 +381,13 @@ class ModelInstance:
     def _admit(self) -> List[GenRequest]:
         out: List[GenRequest] = []
         budget = self.max_batch_tokens
+        # chunked-prefill continuations first: their KV room is already
+        # reserved and decode is waiting on them
+        while self._chunking and budget > 0:
+            r = self._chunking.pop(0)
+            r.slice_len = min(len(r.prompt_tokens) - r.prefill_pos, budget)
+            budget -= r.slice_len
+            out.append(r)
         while budget > 0:
             try:
                 req = self.waiting.get_nowait()
@@ -386,14 +399,20 @@ class ModelInstance:
                 req.done.set()
                 continue
             need = len(req.prompt_tokens)
-            if need > self.max_batch_tokens:
-                req.error = f"prompt too long ({need} tokens)"
+            cap = self.kvm.max_pages_per_seq * self.kvm.page_size
+            if need + req.max_new > cap:
+                req.error = f"prompt too long ({need} tokens; cap {cap})"
                 req.done.set()
                 continue
             if need > budget and out:
                 # put back; try next step
                 self.waiting.put(req)
                 break
+            # a prompt above the whole step budget is admitted CHUNKED:
+            # this step prefills the first max_batch_tokens of it, the
+            # remainder continues next step (decode in between — long
+            # prompts no longer stall the whole batch)
+            req.slice_len = min(need, budget)
             # KV room: prompt + generation
             req.needs_reset = False
             if not self.kvm.can_append(b.seq_id, need + req.max_new):
@@ -406,7 +425,7 @@ class ModelInstance:
                     req.done.set()
                     continue
             out.append(req)
-            budget -= need
+            budget -= req.slice_len
         return out
 
     def _sample(self, logits: torch.Tensor, reqs: List[GenRequest]) -> List[int]:
@@ -447,13 +466,17 @@ class ModelInstance:
     def _prefill(self, reqs: List[GenRequest]):
         plan = []
         n_extra = 0  # shared-prefix rows prepended to the plan (no GenRequest)
+        final = []   # reqs whose slice completes the prompt -> sample
         for r in reqs:
             b = self._bindings[r.agent_id]
-            needs_reset = bool(getattr(r, "needs_reset", False))
-            tokens = r.prompt_tokens
+            first = r.prefill_pos == 0
+            needs_reset = first and bool(getattr(r, "needs_reset", False))
+            sl = r.slice_len or (len(r.prompt_tokens) - r.prefill_pos)
+            tokens = r.prompt_tokens[r.prefill_pos:r.prefill_pos + sl]
+            is_final = (r.prefill_pos + sl) == len(r.prompt_tokens)
             adopt = None
             pk = b.prefix_tokens
-            if (pk is not None and not needs_reset
+            if (pk is not None and first and is_final and not needs_reset
                     and self.kvm.seq_len(b.seq_id) == 0
                     and len(tokens) > len(pk) and tokens[:len(pk)] == pk):
                 key = tuple(pk)
@@ -472,14 +495,25 @@ class ModelInstance:
                     adopt = (sid, len(pk))
                     tokens = tokens[len(pk):]
             plan.append((b.seq_id, tokens, needs_reset, adopt))
+            r.prefill_pos += sl
+            final.append(is_final)
         self._bcast(("prefill", self.name, plan))
         logits = self._prefill_exec(plan)
-        if n_extra:
-            logits = logits[n_extra:]
-        toks = self._sample(logits, reqs)
+        logits = logits[n_extra:]
+        f_rows = [i for i, f in enumerate(final) if f]
+        f_reqs = [r for r, f in zip(reqs, final) if f]
+        with self._lock:
+            for r, f in zip(reqs, final):
+                if not f:
+                    self._chunking.append(r)  # next slice next step
+        if not f_reqs:
+            return
+        if len(f_rows) < len(reqs):
+            logits = logits[f_rows]
+        toks = self._sample(logits, f_reqs)
         now = time.time()
         with self._lock:
-            for r, t in zip(reqs, toks):
+            for r, t in zip(f_reqs, toks):
                 r.generated.append(int(t))
                 r.first_token_t = now
                 self._finish_or_run(r, int(t))
@@ -736,6 +770,8 @@ class ModelInstance:
                 if b.active in self._running_set:
                     self.running.remove(b.active)
                     self._running_set.discard(b.active)
+                if b.active in self._chunking:
+                    self._chunking.remove(b.active)  # mid-chunked-prefill
             while True:
                 try:
                     r = b.queue.get_nowait()

@@ -209,3 +209,42 @@ def test_runtime_stream_wal_ack(llm_runtime):
     done = rt.requests.by_queue(a.id, "completed")
     assert len(done) == 1
     assert done[0].response["response"] == events[-1]["response"]
+
+
+def test_chunked_prefill_matches_unchunked(tmp_path):
+    """A prompt longer than max_batch_tokens prefills across several
+    engine steps (decode interleaves); the generated tokens must equal a
+    single-shot prefill of the same prompt with the same weights."""
+    from agentainer_amd.config import load_config
+    from agentainer_amd.store import Store
+
+    outs = {}
+    for tag, mbt in (("big", 8192), ("small", 48)):
+        cfg = load_config(path="/nonexistent.yaml", env={})
+        root = str(tmp_path / tag)
+        cfg.data["store"]["path"] = root
+        cfg.data["engine"]["kv_pool_gb"] = 0.01
+        cfg.data["engine"]["max_batch_tokens"] = mbt
+        s = Store(root + "/state", sync="interval")
+        torch.manual_seed(11)
+        eng = LLMEngine(s, cfg, device="cpu", state_root=root)
+        rt = Runtime(cfg, engine=eng, store=s, state_root=root)
+        try:
+            a = rt.agents.deploy(name="long", model="tiny-llama",
+                                 sampling={"max_tokens": 5})
+            rt.agents.start(a.id)
+            long_msg = "alpha beta gamma " * 12  # ~200 byte tokens
+            st, p = rt.agent_request(a.id, "POST", "/chat",
+                                     body={"message": long_msg})
+            assert st == 200, p
+            assert p["tokens"] == 5
+            outs[tag] = p["response"]
+            inst = rt.engine._instances["tiny-llama"]
+            # the whole prompt landed in KV exactly once
+            assert inst.kvm.seq_len(a.id) >= len(long_msg)
+            if tag == "small":
+                assert inst.prefill_tokens > mbt  # really ran multiple slices
+                assert not inst._chunking
+        finally:
+            rt.shutdown()
+    assert outs["big"] == outs["small"]
