@@ -145,6 +145,34 @@ def test_prefix_sharing_gpu(gpu_rt):
     assert out2b == outcb
 
 
+def test_chunked_prefill_gpu(tmp_path):
+    """Chunked prefill under the graph-captured/async decode scheduler:
+    a long prompt sliced across steps matches the single-shot result."""
+    outs = {}
+    for tag, mbt in (("big", 8192), ("small", 64)):
+        cfg = load_config(path="/nonexistent.yaml", env={})
+        root = str(tmp_path / tag)
+        cfg.data["store"]["path"] = root
+        cfg.data["engine"]["sync_mode"] = True
+        cfg.data["engine"]["kv_pool_gb"] = 1.0
+        cfg.data["engine"]["max_batch_tokens"] = mbt
+        store = Store(root + "/state", sync="never")
+        torch.manual_seed(5)
+        engine = LLMEngine(store, cfg, device="cuda", state_root=root)
+        manager = Manager(store, engine, cfg)
+        try:
+            a = manager.deploy(name="long", model="tiny-llama")
+            manager.start(a.id)
+            prompt = [3 + (i * 37) % 250 for i in range(300)]
+            outs[tag] = _gen(engine, manager, a, prompt, max_new=6)
+            inst = engine._instances["tiny-llama"]
+            assert inst.kvm.seq_len(a.id) >= 300
+            assert not inst._chunking
+        finally:
+            engine.shutdown()
+    assert outs["big"] == outs["small"] and len(outs["big"]) == 6
+
+
 def test_mixtral_generation_gpu(gpu_rt):
     """tiny-mixtral end-to-end on the HIP kernels + MoE routing."""
     engine, manager, _ = gpu_rt
