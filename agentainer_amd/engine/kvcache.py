@@ -59,6 +59,9 @@ class KVCacheManager:
         self.page_size = page_size
         self.n_pages = n_pages
         self.device = device
+        # dtype: bf16 (default) or float8_e4m3fn — fp8 halves KV bytes per
+        # token (2x the agents per GPU) with OCP e4m3 storage, vLLM-style
+        # scale-free semantics (e4m3 range +-448 covers post-norm K/V)
         self.dtype = dtype
         self.k_caches: List[torch.Tensor] = []
         self.v_caches: List[torch.Tensor] = []

@@ -132,12 +132,19 @@ class ModelInstance:
                 else:
                     self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))
+        kv_dtype_s = str(engine_cfg.get("kv_dtype", "bf16"))
+        if kv_dtype_s in ("fp8", "fp8_e4m3", "e4m3"):
+            kv_dtype = torch.float8_e4m3fn  # halves KV bytes: 2x the agents
+        elif kv_dtype_s in ("bf16", "bfloat16"):
+            kv_dtype = torch.bfloat16
+        else:
+            raise ValueError(f"unsupported kv_dtype {kv_dtype_s!r}")
         n_pages = self._pool_pages(cfg, page_size, device, engine_cfg,
                                    self.tp_size)
         self.kvm = KVCacheManager(cfg.n_layers,
                                   max(cfg.n_kv_heads // self.tp_size, 1),
                                   cfg.head_dim, page_size, n_pages,
-                                  device=device)
+                                  device=device, dtype=kv_dtype)
         self.max_batch_tokens = int(engine_cfg.get("max_batch_tokens", 8192))
         self.max_decode_batch = int(engine_cfg.get("max_decode_batch", 256))
         self.refcount = 0
@@ -201,7 +208,8 @@ class ModelInstance:
     @staticmethod
     def _pool_pages(cfg: LlamaConfig, page_size: int, device: str,
                     engine_cfg: Dict[str, Any], tp_size: int = 1) -> int:
-        bytes_per_page = cfg.kv_bytes_per_token() * page_size // max(tp_size, 1)
+        eb = 1 if str(engine_cfg.get("kv_dtype", "bf16")).startswith(("fp8", "e4m3")) else 2
+        bytes_per_page = cfg.kv_bytes_per_token(eb) * page_size // max(tp_size, 1)
         pool_gb = float(engine_cfg.get("kv_pool_gb", 0.0))
         if device.startswith("cuda") and pool_gb <= 0:
             free, _total = torch.cuda.mem_get_info()

@@ -56,6 +56,22 @@ __device__ __forceinline__ short f2bits(float f) {
   return (short)(rounded >> 16);
 }
 
+// OCP e4m3 <-> f32 (fp8 KV cache, scale-free storage). On gfx950 these
+// lower to v_cvt_pk ops.
+#include <hip/hip_fp8.h>
+typedef unsigned char u8x8 __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ unsigned char f2fp8(float x) {
+  __hip_fp8_e4m3 v(x);
+  return (unsigned char)v.__x;
+}
+
+__device__ __forceinline__ float fp82f(unsigned char b) {
+  __hip_fp8_e4m3 v;
+  v.__x = b;
+  return (float)v;
+}
+
 // Wave-wide f32 reductions (64 lanes).
 __device__ __forceinline__ float wave_sum(float v) {
 #pragma unroll

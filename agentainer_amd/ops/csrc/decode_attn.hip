@@ -36,15 +36,19 @@ constexpr int CHUNK = 64;
 constexpr int SPLITS = 8;
 constexpr int PART_STRIDE = 132;  // 128 o + m + l (+2 pad)
 
-template <int RATIO, int D>
+// CT = short (bf16 pages) or unsigned char (fp8 e4m3 pages, scale-free):
+// the fp8 path reads HALF the KV bytes and pays VALU converts instead of
+// v_dot2 — a good trade everywhere the kernel is HBM- or latency-bound.
+template <int RATIO, int D, typename CT>
 __global__ __launch_bounds__(64) void decode_partial_kernel(
     float* __restrict__ part,           // [B, n_kv, SPLITS, RATIO, PART_STRIDE]
     const short* __restrict__ q,        // [B, n_q, D]
-    const short* __restrict__ k_cache,  // [P, n_kv, D/8, PS, 8]
-    const short* __restrict__ v_cache,  // [P, n_kv, PS, D]
+    const CT* __restrict__ k_cache,     // [P, n_kv, D/8, PS, 8]
+    const CT* __restrict__ v_cache,     // [P, n_kv, PS, D]
     const int* __restrict__ page_table, // [B, max_pages]
     const int* __restrict__ seq_lens,   // [B]
     float scale, int n_q, int n_kv, int PS, int max_pages, long q_ts) {
+  constexpr bool FP8 = sizeof(CT) == 1;
   const int b = blockIdx.x;
   const int g = blockIdx.y;
   const int split = blockIdx.z;
@@ -122,27 +126,49 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       if (tok < len) {
         const long page = pt[tok / PS];
         // this lane's 4 d8 groups: d8 = r*4 + ds (coalesced per r)
-        const short* kp = k_cache +
+        const CT* kp = k_cache +
             ((long)page * n_kv + g) * D8 * PS * 8 + (tok % PS) * 8;
-        bf16x8 kv[4];
+        if constexpr (FP8) {
+          u8x8 kv[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          kv[r] = *reinterpret_cast<const bf16x8*>(
-              kp + (long)(r * 4 + ds) * PS * 8);
+          for (int r = 0; r < 4; ++r)
+            kv[r] = *reinterpret_cast<const u8x8*>(
+                kp + (long)(r * 4 + ds) * PS * 8);
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const bf2* kp2 = reinterpret_cast<const bf2*>(&kv[r]);
-          // this lane's dims (4r+ds)*8..+8 of each head's q: broadcast
-          // LDS reads (t16 lanes share the address — conflict-free)
+          for (int r = 0; r < 4; ++r) {
+            float kf[8];
 #pragma unroll
-          for (int h = 0; h < RATIO; ++h) {
-            const bf16x8 qv8 = *reinterpret_cast<const bf16x8*>(
-                &q_lds[h][(r * 4 + ds) * 8]);
-            const bf2* qp2 = reinterpret_cast<const bf2*>(&qv8);
+            for (int j = 0; j < 8; ++j) kf[j] = fp82f(kv[r][j]);
 #pragma unroll
-            for (int j = 0; j < 4; ++j)
-              acc[h] = __builtin_amdgcn_fdot2_f32_bf16(kp2[j], qp2[j],
-                                                       acc[h], false);
+            for (int h = 0; h < RATIO; ++h) {
+              const bf16x8 qv8 = *reinterpret_cast<const bf16x8*>(
+                  &q_lds[h][(r * 4 + ds) * 8]);
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                acc[h] = fmaf(bits2f(qv8[j]), kf[j], acc[h]);
+            }
+          }
+        } else {
+          bf16x8 kv[4];
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            kv[r] = *reinterpret_cast<const bf16x8*>(
+                kp + (long)(r * 4 + ds) * PS * 8);
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const bf2* kp2 = reinterpret_cast<const bf2*>(&kv[r]);
+            // this lane's dims (4r+ds)*8..+8 of each head's q: broadcast
+            // LDS reads (t16 lanes share the address — conflict-free)
+#pragma unroll
+            for (int h = 0; h < RATIO; ++h) {
+              const bf16x8 qv8 = *reinterpret_cast<const bf16x8*>(
+                  &q_lds[h][(r * 4 + ds) * 8]);
+              const bf2* qp2 = reinterpret_cast<const bf2*>(&qv8);
+#pragma unroll
+              for (int j = 0; j < 4; ++j)
+                acc[h] = __builtin_amdgcn_fdot2_f32_bf16(kp2[j], qp2[j],
+                                                         acc[h], false);
+            }
           }
         }
       }
@@ -189,15 +215,24 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       const int gt = base_tok + t;
       const int gt_c = min(gt, len - 1);
       const long page = pt[gt_c / PS];
-      const short* vp = v_cache +
+      const CT* vp = v_cache +
           (((long)page * n_kv + g) * PS + gt_c % PS) * D + pv_dg * 8;
-      bf16x8 vv = *reinterpret_cast<const bf16x8*>(vp);
+      float vf[8];
+      if constexpr (FP8) {
+        const u8x8 vv = *reinterpret_cast<const u8x8*>(vp);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vf[j] = fp82f(vv[j]);
+      } else {
+        const bf16x8 vv = *reinterpret_cast<const bf16x8*>(vp);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vf[j] = bits2f(vv[j]);
+      }
 #pragma unroll
       for (int h = 0; h < RATIO; ++h) {
         const float pw = p_lds[h][t];  // 0 beyond len
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          o[h][j] = fmaf(pw, bits2f(vv[j]), o[h][j]);
+          o[h][j] = fmaf(pw, vf[j], o[h][j]);
       }
     }
   }
@@ -277,15 +312,21 @@ void paged_decode_attention_ws(torch::Tensor out, torch::Tensor q,
   if (B == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid1(B, n_kv, SPLITS);
-#define LAUNCH_RATIO(R)                                                        \
-  hipLaunchKernelGGL((decode_partial_kernel<R, 128>), grid1, dim3(64), 0,      \
+  const bool fp8 = k_cache.scalar_type() == at::kFloat8_e4m3fn;
+#define LAUNCH_RATIO_CT(R, CT)                                                 \
+  hipLaunchKernelGGL((decode_partial_kernel<R, 128, CT>), grid1, dim3(64), 0,  \
                      stream, workspace.data_ptr<float>(),                      \
                      (const short*)q.data_ptr(),                               \
-                     (const short*)k_cache.data_ptr(),                         \
-                     (const short*)v_cache.data_ptr(),                         \
+                     (const CT*)k_cache.data_ptr(),                            \
+                     (const CT*)v_cache.data_ptr(),                            \
                      page_table.data_ptr<int>(), seq_lens.data_ptr<int>(),     \
                      (float)scale, n_q, n_kv, PS, max_pages,                    \
                      (long)q.stride(0))
+#define LAUNCH_RATIO(R)                                                        \
+  do {                                                                         \
+    if (fp8) LAUNCH_RATIO_CT(R, unsigned char);                                \
+    else LAUNCH_RATIO_CT(R, short);                                            \
+  } while (0)
   switch (ratio) {
     case 1: LAUNCH_RATIO(1); break;
     case 2: LAUNCH_RATIO(2); break;
@@ -296,6 +337,7 @@ void paged_decode_attention_ws(torch::Tensor out, torch::Tensor q,
                   ratio);
   }
 #undef LAUNCH_RATIO
+#undef LAUNCH_RATIO_CT
   hipLaunchKernelGGL((decode_combine_kernel<128>), dim3(B, n_q), dim3(64), 0,
                      stream, (short*)out.data_ptr(),
                      workspace.data_ptr<float>(), seq_lens.data_ptr<int>(),

@@ -22,8 +22,25 @@ namespace {
 
 // grid: (T); block: 256. Each thread moves 16 B; loops over n_kv * D/8
 // chunks for K and n_kv * D/8 chunks for V.
-__global__ void kv_append_kernel(short* __restrict__ k_cache,
-                                 short* __restrict__ v_cache,
+// CT = short (bf16 pool) or unsigned char (fp8 e4m3 pool)
+template <typename CT>
+__device__ __forceinline__ void store8(CT* dst, bf16x8 src);
+template <>
+__device__ __forceinline__ void store8<short>(short* dst, bf16x8 src) {
+  *reinterpret_cast<bf16x8*>(dst) = src;
+}
+template <>
+__device__ __forceinline__ void store8<unsigned char>(unsigned char* dst,
+                                                      bf16x8 src) {
+  u8x8 b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) b[j] = f2fp8(bits2f(src[j]));
+  *reinterpret_cast<u8x8*>(dst) = b;
+}
+
+template <typename CT>
+__global__ void kv_append_kernel(CT* __restrict__ k_cache,
+                                 CT* __restrict__ v_cache,
                                  const short* __restrict__ k,
                                  const short* __restrict__ v,
                                  const long* __restrict__ slots, int n_kv,
@@ -40,11 +57,11 @@ __global__ void kv_append_kernel(short* __restrict__ k_cache,
     bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + (long)t * kts + (long)h * D + d8 * 8);
     // k_cache[page][h][d8][off][0..8]
     long kidx = ((((page * n_kv + h) * D8 + d8) * PS) + off) * 8;
-    *reinterpret_cast<bf16x8*>(k_cache + kidx) = kv8;
+    store8<CT>(k_cache + kidx, kv8);
     bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + (long)t * vts + (long)h * D + d8 * 8);
     // v_cache[page][h][off][d8*8..]
     long vidx = (((page * n_kv + h) * PS) + off) * D + d8 * 8;
-    *reinterpret_cast<bf16x8*>(v_cache + vidx) = vv8;
+    store8<CT>(v_cache + vidx, vv8);
   }
 }
 
@@ -54,11 +71,10 @@ __global__ void kv_append_kernel(short* __restrict__ k_cache,
 __global__ void page_copy_kernel(short* __restrict__ k_cache,
                                  short* __restrict__ v_cache,
                                  short* __restrict__ buf,
-                                 const int* __restrict__ page_ids, int n_kv,
-                                 int D, int PS, bool to_buf) {
+                                 const int* __restrict__ page_ids, long plane,
+                                 bool to_buf) {
   const int p = blockIdx.x;              // index into page_ids
   const long page = page_ids[p];
-  const long plane = (long)n_kv * D * PS;  // shorts per plane
   short* kp = k_cache + page * plane;
   short* vp = v_cache + page * plane;
   short* bk = buf + (long)p * 2 * plane;
@@ -78,7 +94,8 @@ __global__ void page_copy_kernel(short* __restrict__ k_cache,
 
 void kv_append(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k,
                torch::Tensor v, torch::Tensor slot_mapping) {
-  TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16 ||
+              k_cache.scalar_type() == at::kFloat8_e4m3fn);
   TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
   const int T = k.size(0), n_kv = k.size(1), D = k.size(2);
   const int PS = k_cache.size(3);
@@ -89,7 +106,16 @@ void kv_append(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k,
   TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == D);
   if (T == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(kv_append_kernel, dim3(T), dim3(256), 0, stream,
+  if (k_cache.scalar_type() == at::kFloat8_e4m3fn) {
+    hipLaunchKernelGGL(kv_append_kernel<unsigned char>, dim3(T), dim3(256), 0,
+                       stream, (unsigned char*)k_cache.data_ptr(),
+                       (unsigned char*)v_cache.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       slot_mapping.data_ptr<long>(), n_kv, D, PS,
+                       (long)k.stride(0), (long)v.stride(0));
+    return;
+  }
+  hipLaunchKernelGGL(kv_append_kernel<short>, dim3(T), dim3(256), 0, stream,
                      (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
                      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                      slot_mapping.data_ptr<long>(), n_kv, D, PS,
@@ -104,11 +130,15 @@ static void page_copy(torch::Tensor k_cache, torch::Tensor v_cache,
   if (n == 0) return;
   TORCH_CHECK(buf.numel() >= (long)n * 2 * n_kv * D * PS,
               "page buffer too small");
+  TORCH_CHECK(buf.element_size() == k_cache.element_size());
+  // raw byte copy expressed in shorts: plane bytes are even for both bf16
+  // (2 B/elem) and fp8 (1 B/elem, D=128 keeps it 16 B aligned)
+  const long plane = (long)n_kv * D * PS * k_cache.element_size() / 2;
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(page_copy_kernel, dim3(n), dim3(256), 0, stream,
                      (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
-                     (short*)buf.data_ptr(), page_ids.data_ptr<int>(), n_kv, D,
-                     PS, to_buf);
+                     (short*)buf.data_ptr(), page_ids.data_ptr<int>(), plane,
+                     to_buf);
 }
 
 void gather_kv_pages(torch::Tensor dst, torch::Tensor k_cache,

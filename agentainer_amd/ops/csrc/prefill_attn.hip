@@ -43,11 +43,28 @@ __device__ __forceinline__ int kswz(int tok, int byte_off) {
   return byte_off ^ ((tok & 7) << 4);
 }
 
+// CT = short (bf16 pages) or unsigned char (fp8 e4m3 pages): the fp8
+// path dequantizes to bf16 during LDS staging, so the MFMA pipeline and
+// both LDS images are byte-identical to the bf16 build.
+template <typename CT>
+__device__ __forceinline__ bf16x8 load8_cache(const CT* p) {
+  if constexpr (sizeof(CT) == 1) {
+    const u8x8 b = *reinterpret_cast<const u8x8*>(p);
+    bf16x8 r;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) r[j] = f2bits(fp82f(b[j]));
+    return r;
+  } else {
+    return *reinterpret_cast<const bf16x8*>(p);
+  }
+}
+
+template <typename CT>
 __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
     short* __restrict__ out,            // [Tq, n_q, D]
     const short* __restrict__ q,        // [Tq, n_q, D]
-    const short* __restrict__ k_cache,  // [P, n_kv, D/8, PS, 8]
-    const short* __restrict__ v_cache,  // [P, n_kv, PS, D]
+    const CT* __restrict__ k_cache,     // [P, n_kv, D/8, PS, 8]
+    const CT* __restrict__ v_cache,     // [P, n_kv, PS, D]
     const int* __restrict__ page_table, // [B, max_pages]
     const int* __restrict__ seq_lens,   // [B] total context length
     const int* __restrict__ q_starts,   // [B] row offset into q
@@ -120,7 +137,7 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
       bf16x8 vvv{};
       if (gt < kv_end && gt < seq_len) {
         const long pg = pt[gt / PS];
-        vvv = *reinterpret_cast<const bf16x8*>(
+        vvv = load8_cache<CT>(
             k_cache + ((((pg * n_kv + g) * (D / 8) + d8) * PS) + gt % PS) * 8);
       }
       return vvv;
@@ -131,7 +148,7 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
       bf16x8 vvv{};
       if (gt < kv_end && gt < seq_len) {
         const long pg = pt[gt / PS];
-        vvv = *reinterpret_cast<const bf16x8*>(
+        vvv = load8_cache<CT>(
             v_cache + (((pg * n_kv + g) * PS) + gt % PS) * D + d0);
       }
       return vvv;
@@ -308,13 +325,21 @@ void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
   if (max_qlen == 0) return;
   const int qtiles = (max_qlen + QBLK - 1) / QBLK;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(prefill_attn_kernel, dim3(qtiles, B, n_q / 4), dim3(256),
-                     0, stream, (short*)out.data_ptr(),
-                     (const short*)q.data_ptr(),
-                     (const short*)k_cache.data_ptr(),
-                     (const short*)v_cache.data_ptr(),
-                     page_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                     query_starts.data_ptr<int>(), query_lens.data_ptr<int>(),
-                     (float)scale, n_q, n_kv, PS, max_pages,
-                     (long)q.stride(0));
+#define PF_LAUNCH(CT)                                                          \
+  hipLaunchKernelGGL(prefill_attn_kernel<CT>, dim3(qtiles, B, n_q / 4),        \
+                     dim3(256), 0, stream, (short*)out.data_ptr(),             \
+                     (const short*)q.data_ptr(),                               \
+                     (const CT*)k_cache.data_ptr(),                            \
+                     (const CT*)v_cache.data_ptr(),                            \
+                     page_table.data_ptr<int>(), seq_lens.data_ptr<int>(),     \
+                     query_starts.data_ptr<int>(), query_lens.data_ptr<int>(), \
+                     (float)scale, n_q, n_kv, PS, max_pages,                    \
+                     (long)q.stride(0))
+  if (k_cache.scalar_type() == at::kFloat8_e4m3fn) {
+    PF_LAUNCH(unsigned char);
+  } else {
+    TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16);
+    PF_LAUNCH(short);
+  }
+#undef PF_LAUNCH
 }

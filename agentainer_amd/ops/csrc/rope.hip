@@ -70,9 +70,20 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
 
 namespace {
 
+// CT = short (bf16 cache) or unsigned char (fp8 e4m3 cache)
+template <typename CT>
+__device__ __forceinline__ CT f2cache(float x);
+template <>
+__device__ __forceinline__ short f2cache<short>(float x) { return f2bits(x); }
+template <>
+__device__ __forceinline__ unsigned char f2cache<unsigned char>(float x) {
+  return f2fp8(x);
+}
+
+template <typename CT>
 __global__ void rope_append_kernel(
     short* __restrict__ q, short* __restrict__ k, const short* __restrict__ v,
-    short* __restrict__ k_cache, short* __restrict__ v_cache,
+    CT* __restrict__ k_cache, CT* __restrict__ v_cache,
     const float* __restrict__ cos_sin, const int* __restrict__ positions,
     const long* __restrict__ slots, int n_q, int n_kv, int D, int PS,
     long qs, long ks, long vs) {
@@ -97,8 +108,8 @@ __global__ void rope_append_kernel(
   const short* kb = k + (long)t * ks + (long)g * D;
   const float x1 = bits2f(kb[i]);
   const float x2 = bits2f(kb[half + i]);
-  const short r1 = f2bits(x1 * c - x2 * s);
-  const short r2 = f2bits(x1 * s + x2 * c);
+  const CT r1 = f2cache<CT>(x1 * c - x2 * s);
+  const CT r2 = f2cache<CT>(x1 * s + x2 * c);
   if (slot < 0) return;
   const long page = slot / PS;
   const int off = (int)(slot % PS);
@@ -110,8 +121,8 @@ __global__ void rope_append_kernel(
   // v rows are not rotated: copy 2 elements per lane
   const short* vb = v + (long)t * vs + (long)g * D;
   long vbase = (((page * n_kv + g) * PS) + off) * D;
-  v_cache[vbase + i] = vb[i];
-  v_cache[vbase + half + i] = vb[half + i];
+  v_cache[vbase + i] = f2cache<CT>(bits2f(vb[i]));
+  v_cache[vbase + half + i] = f2cache<CT>(bits2f(vb[half + i]));
 }
 
 }  // namespace
@@ -132,7 +143,22 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == D);
   if (T == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(rope_append_kernel, dim3(T, n_q + n_kv), dim3(D / 2), 0,
+  if (k_cache.scalar_type() == at::kFloat8_e4m3fn) {
+    hipLaunchKernelGGL(rope_append_kernel<unsigned char>,
+                       dim3(T, n_q + n_kv), dim3(D / 2), 0, stream,
+                       (short*)q.data_ptr(), (short*)k.data_ptr(),
+                       (const short*)v.data_ptr(),
+                       (unsigned char*)k_cache.data_ptr(),
+                       (unsigned char*)v_cache.data_ptr(),
+                       cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
+                       slot_mapping.data_ptr<long>(), n_q, n_kv, D, PS,
+                       (long)q.stride(0), (long)k.stride(0),
+                       (long)v.stride(0));
+    return;
+  }
+  TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16);
+  hipLaunchKernelGGL(rope_append_kernel<short>, dim3(T, n_q + n_kv),
+                     dim3(D / 2), 0,
                      stream, (short*)q.data_ptr(), (short*)k.data_ptr(),
                      (const short*)v.data_ptr(), (short*)k_cache.data_ptr(),
                      (short*)v_cache.data_ptr(), cos_sin.data_ptr<float>(),

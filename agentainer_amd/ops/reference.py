@@ -70,9 +70,11 @@ def kv_append(k_cache: torch.Tensor, v_cache: torch.Tensor, k: torch.Tensor,
         if slot < 0:
             continue
         page, off = slot // PS, slot % PS
-        # k_cache [P, n_kv, D/8, PS, 8]
-        k_cache[page, :, :, off, :] = k[t].reshape(n_kv, D // 8, 8)
-        v_cache[page, :, off, :] = v[t]
+        # k_cache [P, n_kv, D/8, PS, 8]; .to() casts bf16 -> fp8 when the
+        # pool is e4m3 (index-assignment alone won't cast to float8)
+        k_cache[page, :, :, off, :] = (
+            k[t].reshape(n_kv, D // 8, 8).to(k_cache.dtype))
+        v_cache[page, :, off, :] = v[t].to(v_cache.dtype)
 
 
 def _gather_kv(k_cache, v_cache, page_table_row, length, g):

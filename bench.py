@@ -55,6 +55,8 @@ def parse_args():
     p.add_argument("--max-batch-tokens", type=int, default=16384)
     p.add_argument("--expert-fp8", action="store_true",
                    help="fp8 MFMA expert decode GEMMs (Mixtral, config 5)")
+    p.add_argument("--kv-fp8", action="store_true",
+                   help="OCP e4m3 KV cache (half the KV bytes, 2x agents)")
     return p.parse_args()
 
 
@@ -139,6 +141,8 @@ def main():
     cfg.data["engine"]["max_batch_tokens"] = args.max_batch_tokens
     cfg.data["engine"]["max_decode_batch"] = max(args.agents, 1)
     cfg.data["engine"]["expert_fp8"] = bool(args.expert_fp8)
+    if args.kv_fp8:
+        cfg.data["engine"]["kv_dtype"] = "fp8"
     store = Store(os.path.join(tmp, "state"), sync="interval")
     engine = LLMEngine(store, cfg, device=device, state_root=tmp)
     manager = Manager(store, engine, cfg)
@@ -215,7 +219,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no number
-            "dtype": "bf16+fp8exp" if args.expert_fp8 else "bf16",
+            "dtype": ("bf16+fp8exp" if args.expert_fp8 else
+                      "bf16+fp8kv" if args.kv_fp8 else "bf16"),
             "data": "synthetic",
             "config": {
                 "model": model_name,

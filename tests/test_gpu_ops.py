@@ -367,3 +367,144 @@ def test_mixtral_fp8_generation(tmp_path):
     # (random-init logits are near-uniform, so greedy picks legitimately
     # diverge under quantization and then compound)
     assert len(toks_fp8) == 8 and len(toks_bf16) == 8
+
+
+# ---------------------------------------------------------------- fp8 KV
+
+def _gpu_caches_fp8(P, n_kv, D, PS):
+    k = torch.zeros(P, n_kv, D // 8, PS, 8, dtype=torch.float8_e4m3fn,
+                    device=DEV)
+    v = torch.zeros(P, n_kv, PS, D, dtype=torch.float8_e4m3fn, device=DEV)
+    return k, v
+
+
+def test_kv_append_fp8_matches_reference():
+    torch.manual_seed(13)
+    P, n_kv, D, PS, T = 8, 8, 128, 16, 20
+    kc, vc = _gpu_caches_fp8(P, n_kv, D, PS)
+    k = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    slots = torch.arange(T, dtype=torch.long, device=DEV) * 3 % (P * PS)
+    ops.kv_append(kc, vc, k, v, slots)
+    kr = torch.zeros_like(kc, device="cpu")
+    vr = torch.zeros_like(vc, device="cpu")
+    ref.kv_append(kr, vr, k.cpu(), v.cpu(), slots.cpu())
+    # HIP __hip_fp8 and torch's float8 cast both round-to-nearest-even;
+    # compare dequantized with a 1-ulp allowance for saturation edges
+    dk = (kc.float().cpu() - kr.float()).abs().max().item()
+    dv = (vc.float().cpu() - vr.float()).abs().max().item()
+    assert dk < 0.07 and dv < 0.07, (dk, dv)
+
+
+def test_decode_attention_fp8_matches_reference():
+    torch.manual_seed(14)
+    lens = [1, 16, 130, 300]
+    n_q, n_kv, D, PS = 32, 8, 128, 16
+    B = len(lens)
+    max_pages = max(-(-l // PS) for l in lens)
+    P = B * max_pages + 1
+    kc, vc = _gpu_caches_fp8(P, n_kv, D, PS)
+    page_table = torch.zeros(B, max_pages, dtype=torch.int32)
+    next_page = 1
+    for b in range(B):
+        for i in range(-(-lens[b] // PS)):
+            page_table[b, i] = next_page
+            next_page += 1
+    for b in range(B):
+        _fill_seq(kc, vc, page_table[b], lens[b], n_kv, D, PS, seed=200 + b)
+    q = torch.randn(B, n_q, D, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty_like(q)
+    scale = 1.0 / math.sqrt(D)
+    ops.paged_decode_attention(out, q, kc, vc, page_table.to(DEV),
+                               torch.tensor(lens, dtype=torch.int32, device=DEV),
+                               scale)
+    want = torch.empty(B, n_q, D, dtype=torch.bfloat16)
+    ref.paged_decode_attention(want, q.cpu(), kc.cpu(), vc.cpu(), page_table,
+                               torch.tensor(lens, dtype=torch.int32), scale)
+    diff = (out.float().cpu() - want.float()).abs().max().item()
+    assert diff < 0.05, f"fp8 decode attn max abs diff {diff}"
+
+
+def test_prefill_attention_fp8_matches_reference():
+    torch.manual_seed(15)
+    n_q, n_kv, D, PS = 32, 8, 128, 16
+    ctx, new = 48, 64
+    total = ctx + new
+    max_pages = -(-total // PS)
+    kc, vc = _gpu_caches_fp8(max_pages + 2, n_kv, D, PS)
+    page_table = torch.arange(1, max_pages + 1, dtype=torch.int32).unsqueeze(0)
+    _fill_seq(kc, vc, page_table[0], total, n_kv, D, PS, seed=77)
+    q = torch.randn(new, n_q, D, dtype=torch.bfloat16, device=DEV)
+    out = torch.zeros_like(q)
+    scale = 1.0 / math.sqrt(D)
+    ops.paged_prefill_attention(
+        out, q, kc, vc, page_table.to(DEV),
+        torch.tensor([total], dtype=torch.int32, device=DEV),
+        torch.tensor([0], dtype=torch.int32, device=DEV),
+        torch.tensor([new], dtype=torch.int32, device=DEV), scale)
+    want = torch.zeros(new, n_q, D, dtype=torch.bfloat16)
+    ref.paged_prefill_attention(
+        want, q.cpu(), kc.cpu(), vc.cpu(), page_table,
+        torch.tensor([total], dtype=torch.int32),
+        torch.tensor([0], dtype=torch.int32),
+        torch.tensor([new], dtype=torch.int32), scale)
+    diff = (out.float().cpu() - want.float()).abs().max().item()
+    assert diff < 0.05, f"fp8 prefill attn max abs diff {diff}"
+
+
+def test_rope_append_fp8_matches_reference():
+    torch.manual_seed(16)
+    P, n_q, n_kv, D, PS, T = 6, 32, 8, 128, 16, 12
+    kc, vc = _gpu_caches_fp8(P, n_kv, D, PS)
+    cos_sin = ref.make_cos_sin_table(256, D, device=DEV)
+    q = torch.randn(T, n_q, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    pos = torch.arange(T, dtype=torch.int32, device=DEV)
+    slots = (torch.arange(T, dtype=torch.long, device=DEV) * 5) % (P * PS)
+    q_ref, k_ref = q.cpu().clone(), k.cpu().clone()
+    ops.rope_append(q, k, v, kc, vc, cos_sin, pos, slots)
+    ref.rope_inplace(q_ref, k_ref, cos_sin.cpu(), pos.cpu())
+    kr = torch.zeros_like(kc, device="cpu")
+    vr = torch.zeros_like(vc, device="cpu")
+    ref.kv_append(kr, vr, k_ref, v.cpu(), slots.cpu())
+    assert (q.float().cpu() - q_ref.float()).abs().max().item() < 2e-2
+    dk = (kc.float().cpu() - kr.float()).abs().max().item()
+    dv = (vc.float().cpu() - vr.float()).abs().max().item()
+    assert dk < 0.3 and dv < 0.07, (dk, dv)  # rope'd k: e4m3 quantum at |x|~4
+
+
+def test_engine_generation_fp8_kv(tmp_path):
+    """End-to-end fp8-KV generation on device + stop/resume exactness."""
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import LLMEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+    from test_gpu_engine import _gen
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    root = str(tmp_path)
+    cfg.data["store"]["path"] = root
+    cfg.data["engine"]["sync_mode"] = True
+    cfg.data["engine"]["kv_pool_gb"] = 1.0
+    cfg.data["engine"]["kv_dtype"] = "fp8"
+    store = Store(root + "/state", sync="never")
+    engine = LLMEngine(store, cfg, device="cuda", state_root=root)
+    manager = Manager(store, engine, cfg)
+    try:
+        a = manager.deploy(name="f8a", model="tiny-llama")
+        manager.start(a.id)
+        b = manager.deploy(name="f8b", model="tiny-llama")
+        manager.start(b.id)
+        inst = engine._instances["tiny-llama"]
+        assert inst.kvm.dtype == torch.float8_e4m3fn
+        prompt = list(range(3, 40))
+        out_a = _gen(engine, manager, a, prompt)
+        out_b = _gen(engine, manager, b, prompt)
+        assert out_a == out_b and len(out_a) == 8
+        manager.stop(a.id)   # offload fp8 pages to pinned host
+        manager.resume(a.id)
+        p2 = list(range(50, 70))
+        assert _gen(engine, manager, a, p2) == _gen(engine, manager, b, p2)
+    finally:
+        engine.shutdown()
