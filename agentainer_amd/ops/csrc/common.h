@@ -72,6 +72,23 @@ __device__ __forceinline__ float fp82f(unsigned char b) {
   return (float)v;
 }
 
+// Fast path: 8 e4m3 bytes -> 8 f32 in 4 v_cvt_pk_f32_fp8 (the scalar
+// __hip_fp8 float operator lowers to a software sequence on ROCm 7.2,
+// which made the first fp8 decode-attention build VALU-bound).
+typedef float f32x2_t __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ void fp8x8_to_f32(const u8x8 v, float* out) {
+  int w0, w1;
+  __builtin_memcpy(&w0, &v, 4);
+  __builtin_memcpy(&w1, reinterpret_cast<const char*>(&v) + 4, 4);
+  const f32x2_t p0 = __builtin_amdgcn_cvt_pk_f32_fp8(w0, false);
+  const f32x2_t p1 = __builtin_amdgcn_cvt_pk_f32_fp8(w0, true);
+  const f32x2_t p2 = __builtin_amdgcn_cvt_pk_f32_fp8(w1, false);
+  const f32x2_t p3 = __builtin_amdgcn_cvt_pk_f32_fp8(w1, true);
+  out[0] = p0.x; out[1] = p0.y; out[2] = p1.x; out[3] = p1.y;
+  out[4] = p2.x; out[5] = p2.y; out[6] = p3.x; out[7] = p3.y;
+}
+
 // Wave-wide f32 reductions (64 lanes).
 __device__ __forceinline__ float wave_sum(float v) {
 #pragma unroll

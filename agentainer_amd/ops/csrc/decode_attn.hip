@@ -137,8 +137,7 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             float kf[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j) kf[j] = fp82f(kv[r][j]);
+            fp8x8_to_f32(kv[r], kf);
 #pragma unroll
             for (int h = 0; h < RATIO; ++h) {
               const bf16x8 qv8 = *reinterpret_cast<const bf16x8*>(
@@ -220,8 +219,7 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       float vf[8];
       if constexpr (FP8) {
         const u8x8 vv = *reinterpret_cast<const u8x8*>(vp);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vf[j] = fp82f(vv[j]);
+        fp8x8_to_f32(vv, vf);
       } else {
         const bf16x8 vv = *reinterpret_cast<const bf16x8*>(vp);
 #pragma unroll

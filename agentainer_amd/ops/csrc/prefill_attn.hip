@@ -50,9 +50,11 @@ template <typename CT>
 __device__ __forceinline__ bf16x8 load8_cache(const CT* p) {
   if constexpr (sizeof(CT) == 1) {
     const u8x8 b = *reinterpret_cast<const u8x8*>(p);
+    float f[8];
+    fp8x8_to_f32(b, f);
     bf16x8 r;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) r[j] = f2bits(fp82f(b[j]));
+    for (int j = 0; j < 8; ++j) r[j] = f2bits(f[j]);
     return r;
   } else {
     return *reinterpret_cast<const bf16x8*>(p);
