@@ -129,7 +129,8 @@ class KVCacheManager:
             s = self._seqs.get(seq_id)
             if not s:
                 return 0
-            return len(s.pages) * self.page_shorts * self.n_layers * 2
+            return (len(s.pages) * self.page_shorts * self.n_layers *
+                    self.k_caches[0].element_size())
 
     # ---------- allocation ----------
 
