@@ -99,6 +99,37 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
   }
   __builtin_amdgcn_s_waitcnt(0);
 
+  // fp8 MFMA score phase operands: q quantized per head to e4m3 in LDS
+  // (B operand, row stride padded to dodge 16-way bank conflicts) plus
+  // the per-head scales. Scores come from v_mfma_f32_16x16x32_fp8_fp8 —
+  // CDNA4 has no VALU fp8 dot, and the cvt+fma fallback was VALU-bound
+  // (225 us vs bf16's 204 at ctx 4096 despite half the bytes).
+  constexpr int Q8S = D + 8;  // bytes per padded q8 row
+  __shared__ unsigned char q8_lds[16 * Q8S];
+  __shared__ float qsc_lds[16];
+  if constexpr (sizeof(CT) == 1) {
+#pragma unroll
+    for (int h = 0; h < 16; ++h) {
+      float sc = 0.f;
+      if (h < RATIO) {
+        float mx = fmaxf(fabsf(bits2f(q_lds[h][2 * lane])),
+                         fabsf(bits2f(q_lds[h][2 * lane + 1])));
+        mx = wave_max(mx);
+        sc = fmaxf(mx, 1e-8f) / 448.f;
+        const float inv = 1.f / sc;
+        q8_lds[h * Q8S + 2 * lane] =
+            f2fp8(bits2f(q_lds[h][2 * lane]) * inv);
+        q8_lds[h * Q8S + 2 * lane + 1] =
+            f2fp8(bits2f(q_lds[h][2 * lane + 1]) * inv);
+      } else {
+        q8_lds[h * Q8S + 2 * lane] = 0;
+        q8_lds[h * Q8S + 2 * lane + 1] = 0;
+      }
+      if (lane == 0) qsc_lds[h] = sc;
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+  }
+
   // PV lane split: pv_tg = lane/16 walks tokens (4 per u-step, adjacent
   // rows => 1 KB contiguous per load instruction), pv_dg = lane%16 owns
   // 8 dims (one 16 B load per row). o[RATIO][8] keeps the register count
@@ -116,6 +147,85 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
   const int* pt = page_table + (long)b * max_pages;
 
   for (int c = c0; c < c1; ++c) {
+    if constexpr (FP8) {
+      // MFMA score phase: S^T[16 tok x 16 head-col] per 16-token group.
+      // A = K rows (lane: tok = l%16, k-dims (l/16)*8..+8 per 32-dim
+      // chunk — 16 lanes x 8 B = one contiguous page-row span), B = q8
+      // from LDS. C lane l holds tokens g*16 + (l/16)*4 + r of head
+      // l%16. Per-head chunk stats cross lanes through small LDS
+      // broadcast arrays; the m/l/alpha bookkeeping stays statically
+      // indexed (a dynamic m[myh] would spill the accumulators).
+      __shared__ float cm_b[16], cs_b[16], mn_b[16];
+      const int myh = lane % 16;
+      float sv[16];  // this lane's 16 token scores for head `myh`
+#pragma unroll
+      for (int gt = 0; gt < 4; ++gt) {
+        const int tok = c * CHUNK + gt * 16 + (lane % 16);
+        const int tok_c = min(tok, len - 1);
+        const long page = pt[tok_c / PS];
+        const CT* kp = k_cache +
+            ((long)page * n_kv + g) * D8 * PS * 8 + (tok_c % PS) * 8;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int dc = 0; dc < 4; ++dc) {
+          long a, bq;
+          __builtin_memcpy(&a, kp + (long)(dc * 4 + lane / 16) * PS * 8, 8);
+          __builtin_memcpy(&bq, &q8_lds[myh * Q8S + dc * 32 + (lane / 16) * 8],
+                           8);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, bq, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) sv[gt * 4 + r] = acc[r];
+      }
+      // mask + scale, per-head chunk max (lanes sharing l%16: xor 16/32)
+      const float qs = qsc_lds[myh] * scale;
+      float cmax = -FLT_MAX;
+#pragma unroll
+      for (int gt = 0; gt < 4; ++gt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int tok = c * CHUNK + gt * 16 + (lane / 16) * 4 + r;
+          const float v = (tok < len) ? sv[gt * 4 + r] * qs : -FLT_MAX;
+          sv[gt * 4 + r] = v;
+          cmax = fmaxf(cmax, v);
+        }
+      cmax = fmaxf(cmax, __shfl_xor(cmax, 16, WAVE));
+      cmax = fmaxf(cmax, __shfl_xor(cmax, 32, WAVE));
+      if (lane < 16) cm_b[myh] = cmax;
+      __builtin_amdgcn_s_waitcnt(0);
+      // statically-indexed running-max update; new maxima broadcast back
+      float alpha_s[RATIO];
+#pragma unroll
+      for (int h = 0; h < RATIO; ++h) {
+        const float mn = fmaxf(m[h], cm_b[h]);
+        alpha_s[h] = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o[h][j] *= alpha_s[h];
+        m[h] = mn;
+        if (lane == 0) mn_b[h] = mn;
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+      // p = exp(s - mn) -> p_lds; per-head chunk sum via the same xor net
+      const float mn_my = mn_b[myh];
+      float csum = 0.f;
+#pragma unroll
+      for (int gt = 0; gt < 4; ++gt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int t = gt * 16 + (lane / 16) * 4 + r;
+          const float pv = (sv[gt * 4 + r] == -FLT_MAX)
+                               ? 0.f
+                               : __expf(sv[gt * 4 + r] - mn_my);
+          if (myh < RATIO) p_lds[myh][t] = pv;
+          csum += pv;
+        }
+      csum += __shfl_xor(csum, 16, WAVE);
+      csum += __shfl_xor(csum, 32, WAVE);
+      if (lane < 16) cs_b[myh] = csum;
+      __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+      for (int h = 0; h < RATIO; ++h) l[h] = l[h] * alpha_s[h] + cs_b[h];
+    } else {
     float s[4][RATIO];  // [16-token sub-pass][head]
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
@@ -128,26 +238,7 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
         // this lane's 4 d8 groups: d8 = r*4 + ds (coalesced per r)
         const CT* kp = k_cache +
             ((long)page * n_kv + g) * D8 * PS * 8 + (tok % PS) * 8;
-        if constexpr (FP8) {
-          u8x8 kv[4];
-#pragma unroll
-          for (int r = 0; r < 4; ++r)
-            kv[r] = *reinterpret_cast<const u8x8*>(
-                kp + (long)(r * 4 + ds) * PS * 8);
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            float kf[8];
-            fp8x8_to_f32(kv[r], kf);
-#pragma unroll
-            for (int h = 0; h < RATIO; ++h) {
-              const bf16x8 qv8 = *reinterpret_cast<const bf16x8*>(
-                  &q_lds[h][(r * 4 + ds) * 8]);
-#pragma unroll
-              for (int j = 0; j < 8; ++j)
-                acc[h] = fmaf(bits2f(qv8[j]), kf[j], acc[h]);
-            }
-          }
-        } else {
+        {
           bf16x8 kv[4];
 #pragma unroll
           for (int r = 0; r < 4; ++r)
@@ -180,7 +271,6 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       }
     }
     // per-head online softmax update over the whole 64-token chunk
-    const int c_len = min(CHUNK, len - c * CHUNK);
 #pragma unroll
     for (int h = 0; h < RATIO; ++h) {
       float cm = fmaxf(fmaxf(s[0][h], s[1][h]), fmaxf(s[2][h], s[3][h]));
@@ -203,6 +293,7 @@ __global__ __launch_bounds__(64) void decode_partial_kernel(
       m[h] = mn;
     }
     __builtin_amdgcn_s_waitcnt(0);
+    }
 
     // PV: per u-step this lane reads V[tok = base + u*4 + pv_tg]
     // [dims pv_dg*8 .. +8) — the wave covers 4 adjacent token rows x
