@@ -407,7 +407,8 @@ class ModelInstance:
                 req.done.set()
                 continue
             need = len(req.prompt_tokens)
-            cap = self.kvm.max_pages_per_seq * self.kvm.page_size
+            cap = min(self.kvm.max_pages_per_seq * self.kvm.page_size,
+                      self.cfg.max_position)
             if need + req.max_new > cap:
                 req.error = f"prompt too long ({need} tokens; cap {cap})"
                 req.done.set()
@@ -421,12 +422,17 @@ class ModelInstance:
             # remainder continues next step (decode in between — long
             # prompts no longer stall the whole batch)
             req.slice_len = min(need, budget)
-            # KV room: prompt + generation
-            req.needs_reset = False
-            if not self.kvm.can_append(b.seq_id, need + req.max_new):
+            # KV room: prompt + generation. Context truncation fires on
+            # pool pressure AND on the model's rope/position horizon —
+            # a conversation crossing max_position would otherwise step
+            # the rope table out of bounds
+            req.needs_reset = (
+                self.kvm.seq_len(b.seq_id) + need + req.max_new
+                > self.cfg.max_position
+                or not self.kvm.can_append(b.seq_id, need + req.max_new))
+            if req.needs_reset:
                 # truncate conversation: reset KV (applied in the prefill
                 # plan so TP workers mirror it), re-prefill just this turn
-                req.needs_reset = True
                 if not self.kvm.can_append_after_reset(b.seq_id,
                                                        need + req.max_new):
                     req.error = "KV pool exhausted"

@@ -248,3 +248,40 @@ def test_chunked_prefill_matches_unchunked(tmp_path):
         finally:
             rt.shutdown()
     assert outs["big"] == outs["small"]
+
+
+def test_chunked_prefill_streams_and_truncates(tmp_path):
+    """Chunked prefill composes with SSE streaming (first token after the
+    last slice) and with context-truncation resets."""
+    from agentainer_amd.config import load_config
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    root = str(tmp_path / "r")
+    cfg.data["store"]["path"] = root
+    cfg.data["engine"]["kv_pool_gb"] = 0.01
+    cfg.data["engine"]["max_batch_tokens"] = 40
+    s = Store(root + "/state", sync="interval")
+    eng = LLMEngine(s, cfg, device="cpu", state_root=root)
+    rt = Runtime(cfg, engine=eng, store=s, state_root=root)
+    try:
+        a = rt.agents.deploy(name="lng", model="tiny-llama",
+                             sampling={"max_tokens": 4})
+        rt.agents.start(a.id)
+        msg = "delta echo foxtrot " * 8  # ~150 tokens >> 40 budget
+        events = list(rt.engine.chat_stream(a.id, msg))
+        assert events[-1]["done"] is True and events[-1]["tokens"] == 4
+        toks = [e for e in events[:-1] if e.get("token") is not None]
+        assert len(toks) == 4
+        inst = rt.engine._instances["tiny-llama"]
+        assert not inst._chunking
+        # keep chatting until the pool forces a truncation reset, chunked
+        for i in range(30):
+            st, p = rt.agent_request(a.id, "POST", "/chat",
+                                     body={"message": msg})
+            if st != 200:
+                break
+            assert p["tokens"] == 4
+        assert st == 200  # kept serving through resets
+    finally:
+        rt.shutdown()
