@@ -39,6 +39,7 @@ class LlamaConfig:
     max_position: int = 8192
     norm_eps: float = 1e-5
     tie_embeddings: bool = True
+    qkv_bias: bool = False  # Qwen2-family checkpoints carry q/k/v biases
 
     @property
     def q_size(self):
@@ -83,12 +84,17 @@ class LlamaAttention(torch.nn.Module):
             torch.empty(q_out + 2 * kv_out, cfg.hidden_size, dtype=torch.bfloat16))
         self.o_proj = torch.nn.Parameter(
             torch.empty(cfg.hidden_size, q_out, dtype=torch.bfloat16))
+        self.qkv_bias = (torch.nn.Parameter(
+            torch.zeros(q_out + 2 * kv_out, dtype=torch.bfloat16))
+            if cfg.qkv_bias else None)
         self.qkv_packed = None
         self.o_packed = None
 
     def forward(self, h, k_cache, v_cache, md: AttnMetadata, cos_sin, tp_group=None):
         T = h.size(0)
         qkv = ops.linear(h, self.qkv_proj, self.qkv_packed)
+        if self.qkv_bias is not None:
+            qkv = qkv + self.qkv_bias
         q_sz = self.n_heads * self.head_dim
         kv_sz = self.n_kv * self.head_dim
         # strided views into the fused GEMM output — the HIP kernels take
@@ -261,6 +267,11 @@ class LlamaForCausalLM(torch.nn.Module):
             k = t(pfx + "self_attn.k_proj.weight")
             v = t(pfx + "self_attn.v_proj.weight")
             layer.attn.qkv_proj.copy_(torch.cat([q, k, v], dim=0))
+            if layer.attn.qkv_bias is not None:
+                layer.attn.qkv_bias.copy_(torch.cat(
+                    [t(pfx + "self_attn.q_proj.bias"),
+                     t(pfx + "self_attn.k_proj.bias"),
+                     t(pfx + "self_attn.v_proj.bias")], dim=0))
             layer.attn.o_proj.copy_(t(pfx + "self_attn.o_proj.weight"))
             g = t(pfx + "mlp.gate_proj.weight")
             u = t(pfx + "mlp.up_proj.weight")
@@ -281,9 +292,11 @@ def config_from_hf(path: str) -> LlamaConfig:
         hf = json.load(f)
     mt = hf.get("model_type", "")
     archs = hf.get("architectures", [])
-    if mt not in ("llama", "") and not any("Llama" in a for a in archs):
+    known = mt in ("llama", "qwen2", "") or any(
+        ("Llama" in a or "Qwen2" in a) for a in archs)
+    if not known:
         raise ValueError(f"unsupported model_type {mt!r} at {path} "
-                         "(llama-family only)")
+                         "(llama/qwen2 families only)")
     n_heads = int(hf["num_attention_heads"])
     hidden = int(hf["hidden_size"])
     head_dim = int(hf.get("head_dim") or hidden // n_heads)
@@ -303,6 +316,8 @@ def config_from_hf(path: str) -> LlamaConfig:
         max_position=int(hf.get("max_position_embeddings", 8192)),
         norm_eps=float(hf.get("rms_norm_eps", 1e-5)),
         tie_embeddings=bool(hf.get("tie_word_embeddings", False)),
+        qkv_bias=(mt == "qwen2" or any("Qwen2" in a for a in archs)
+                  or bool(hf.get("attention_bias", False))),
     )
 
 

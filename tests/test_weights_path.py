@@ -132,3 +132,67 @@ def test_deploy_unknown_model_rejected(tmp_path):
             rt.agents.deploy(name="bad", model="no-such-model-xyz")
     finally:
         rt.shutdown()
+
+
+def test_deploy_qwen2_checkpoint(tmp_path):
+    """Qwen2-family weights-path deploy: qkv biases loaded and applied."""
+    from safetensors.torch import save_file
+
+    d = str(tmp_path / "qwen")
+    os.makedirs(d, exist_ok=True)
+    cfg = {
+        "model_type": "qwen2", "architectures": ["Qwen2ForCausalLM"],
+        "vocab_size": 512, "hidden_size": 512, "num_hidden_layers": 2,
+        "num_attention_heads": 4, "num_key_value_heads": 1, "head_dim": 128,
+        "intermediate_size": 1024, "rope_theta": 1e6,
+        "max_position_embeddings": 4096, "rms_norm_eps": 1e-6,
+        "tie_word_embeddings": True, "eos_token_id": 0,
+    }
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    lc = config_from_hf(d)
+    assert lc.qkv_bias is True and lc.rope_theta == 1e6
+    torch.manual_seed(9)
+    m = LlamaForCausalLM(lc, device="cpu", seed=9)
+    sd = {"model.embed_tokens.weight": m.embed.data.clone(),
+          "model.norm.weight": m.final_ln.data.clone()}
+    q_sz, kv_sz = lc.q_size, lc.kv_size
+    for i, layer in enumerate(m.layers):
+        pfx = f"model.layers.{i}."
+        qkv = layer.attn.qkv_proj.data
+        sd[pfx + "self_attn.q_proj.weight"] = qkv[:q_sz].clone()
+        sd[pfx + "self_attn.k_proj.weight"] = qkv[q_sz:q_sz + kv_sz].clone()
+        sd[pfx + "self_attn.v_proj.weight"] = qkv[q_sz + kv_sz:].clone()
+        bias = torch.randn(q_sz + 2 * kv_sz, dtype=torch.bfloat16) * 0.02
+        sd[pfx + "self_attn.q_proj.bias"] = bias[:q_sz].clone()
+        sd[pfx + "self_attn.k_proj.bias"] = bias[q_sz:q_sz + kv_sz].clone()
+        sd[pfx + "self_attn.v_proj.bias"] = bias[q_sz + kv_sz:].clone()
+        sd[pfx + "self_attn.o_proj.weight"] = layer.attn.o_proj.data.clone()
+        gu = layer.mlp.gate_up.data
+        sd[pfx + "mlp.gate_proj.weight"] = gu[:lc.intermediate_size].clone()
+        sd[pfx + "mlp.up_proj.weight"] = gu[lc.intermediate_size:].clone()
+        sd[pfx + "mlp.down_proj.weight"] = layer.mlp.down.data.clone()
+        sd[pfx + "input_layernorm.weight"] = layer.input_ln.data.clone()
+        sd[pfx + "post_attention_layernorm.weight"] = layer.post_ln.data.clone()
+    save_file(sd, os.path.join(d, "model.safetensors"))
+
+    cfg2 = load_config(path="/nonexistent.yaml", env={})
+    root = str(tmp_path / "root")
+    cfg2.data["store"]["path"] = root
+    cfg2.data["engine"]["kv_pool_gb"] = 0.01
+    s = Store(root + "/state", sync="interval")
+    rt = Runtime(cfg2, engine=LLMEngine(s, cfg2, device="cpu", state_root=root),
+                 store=s, state_root=root)
+    try:
+        a = rt.agents.deploy(name="qwen-agent", model=d,
+                             sampling={"max_tokens": 5})
+        rt.agents.start(a.id)
+        inst = rt.engine._instances[d]
+        # biases survived the load (nonzero)
+        b0 = inst.model.layers[0].attn.qkv_bias
+        assert b0 is not None and b0.abs().sum() > 0
+        st, p = rt.agent_request(a.id, "POST", "/chat",
+                                 body={"message": "ni hao"})
+        assert st == 200 and 0 < p["tokens"] <= 5
+    finally:
+        rt.shutdown()
