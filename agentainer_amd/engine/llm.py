@@ -998,6 +998,12 @@ class LLMEngine:
         if stream:
             req.stream_q = queue.Queue()
         with inst._lock:
+            # the binding may have been unbound between our lookup and
+            # taking the lock — enqueueing then would orphan the request
+            # in a dead queue until the caller's timeout (unbind drains
+            # under this same lock, so this check closes the race)
+            if inst._bindings.get(agent_id) is not b:
+                raise EngineUnavailable(f"agent {agent_id} detached")
             b.queue.put(req)
             inst._pump_agent(b)
         return inst, b, req, float(sampling.get("timeout_s", 120.0))
