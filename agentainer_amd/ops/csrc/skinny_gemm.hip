@@ -131,11 +131,28 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 
 __global__ __launch_bounds__(256) void splitk_combine_kernel(
     short* __restrict__ out, const float* __restrict__ ws, long mn, int S) {
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= mn) return;
-  float acc = 0.f;
-  for (int s = 0; s < S; ++s) acc += ws[(long)s * mn + i];
-  out[i] = f2bits(acc);
+  // 4 elements per thread, 16-B loads per plane: the scalar version ran
+  // latency-bound at ~0.8 TB/s and cost 4.8 us inside the decode graph
+  const long i4 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i4 >= mn) return;
+  if (i4 + 4 <= mn) {
+    f32x4 acc = *reinterpret_cast<const f32x4*>(ws + i4);
+    for (int s = 1; s < S; ++s) {
+      const f32x4 v = *reinterpret_cast<const f32x4*>(ws + (long)s * mn + i4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] += v[j];
+    }
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = f2bits(acc[j]);
+    *reinterpret_cast<bf16x4*>(out + i4) = o;
+  } else {
+    for (long i = i4; i < mn; ++i) {
+      float acc = 0.f;
+      for (int s = 0; s < S; ++s) acc += ws[(long)s * mn + i];
+      out[i] = f2bits(acc);
+    }
+  }
 }
 
 }  // namespace
@@ -171,7 +188,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
     else SG_LAUNCH(true, false, ws.data_ptr());
     const long mn = (long)M * N;
     hipLaunchKernelGGL(splitk_combine_kernel,
-                       dim3((mn + 255) / 256), dim3(256), 0, stream,
+                       dim3((mn + 1023) / 1024), dim3(256), 0, stream,
                        (short*)out.data_ptr(), ws.data_ptr<float>(), mn, split);
   }
 #undef SG_LAUNCH
@@ -359,7 +376,7 @@ void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
     TORCH_CHECK(ws.numel() >= (long)split * M * N && ws.scalar_type() == at::kFloat);
     SGP_DISPATCH(true, ws.data_ptr());
     const long mn = (long)M * N;
-    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 255) / 256),
+    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 1023) / 1024),
                        dim3(256), 0, stream, (short*)out.data_ptr(),
                        ws.data_ptr<float>(), mn, split);
   }
@@ -577,7 +594,7 @@ void skinny_gemm_fp8(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
                        w_packed.data_ptr<unsigned char>(), sw.data_ptr<float>(),
                        M, (int)N, (int)K, kps);
     const long mn = (long)M * N;
-    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 255) / 256),
+    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 1023) / 1024),
                        dim3(256), 0, stream, (short*)out.data_ptr(),
                        ws.data_ptr<float>(), mn, split);
   }
