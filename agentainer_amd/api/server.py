@@ -217,7 +217,9 @@ def create_app(rt: Runtime) -> FastAPI:
     @app.delete("/agents/{agent_id}")
     async def remove(agent_id: str, request: HttpRequest,
                      _tok: str = Depends(require_auth)):
-        rt.agents.remove(agent_id, request_manager=rt.requests)
+        # detach + KV teardown can block on the engine step mutex
+        await run_in_threadpool(rt.agents.remove, agent_id,
+                                request_manager=rt.requests)
         rt.logger.audit("api", "remove", agent_id, "success", **_client(request))
         return envelope(True, "agent removed")
 

@@ -141,10 +141,15 @@ class ModelInstance:
             raise ValueError(f"unsupported kv_dtype {kv_dtype_s!r}")
         n_pages = self._pool_pages(cfg, page_size, device, engine_cfg,
                                    self.tp_size)
+        # sequence-slot budget: device page-table/length mirrors are tiny
+        # (2048 slots x 512 pages x 4 B = 4 MB), so default well above the
+        # largest measured concurrency (1k+ agents per GPU with fp8 KV)
+        max_seqs = int(engine_cfg.get("max_seqs", 2048))
         self.kvm = KVCacheManager(cfg.n_layers,
                                   max(cfg.n_kv_heads // self.tp_size, 1),
                                   cfg.head_dim, page_size, n_pages,
-                                  device=device, dtype=kv_dtype)
+                                  device=device, dtype=kv_dtype,
+                                  max_slots=max_seqs)
         self.max_batch_tokens = int(engine_cfg.get("max_batch_tokens", 8192))
         self.max_decode_batch = int(engine_cfg.get("max_decode_batch", 256))
         self.refcount = 0
