@@ -202,6 +202,15 @@ class ModelInstance:
                              and bool(engine_cfg.get("async_decode", True)))
         self._spec: Optional[Dict[str, Any]] = None
         self._graphs: Dict[int, Dict[str, Any]] = {}
+        # prefill runs on its own HIP stream so a prefill batch overlaps
+        # the in-flight decode graph replay (prefill is MFMA-bound, decode
+        # is HBM-bound — they share the chip well). Safe by construction:
+        # a sequence is either prefilling or decoding, never both, so the
+        # two streams touch disjoint pages/page-table rows; the only
+        # cross-stream readers are speculatively-launched rows that are
+        # already invalidated (their outputs are discarded).
+        self._prefill_stream = (torch.cuda.Stream()
+                                if self.is_gpu else None)
         self._pad_slot = -1
         if self.is_gpu:
             # reserve one sequence slot as the graph's pad row target
@@ -351,10 +360,16 @@ class ModelInstance:
         # GPU runs the step launched above
         self._resolve_spec(prev, launched)
 
-        # phase 3: admission + prefill (runs after the in-flight decode)
+        # phase 3: admission + prefill, overlapping the decode replay
+        # launched in phase 1 (side stream; fully host-synced by the
+        # prefill sampling before this returns)
         admitted = self._admit()
         if admitted:
-            self._prefill(admitted)
+            if self._prefill_stream is not None:
+                with torch.cuda.stream(self._prefill_stream):
+                    self._prefill(admitted)
+            else:
+                self._prefill(admitted)
         return bool(batch) or bool(admitted) or prev is not None
 
     def _resolve_spec(self, prev, launched) -> None:
