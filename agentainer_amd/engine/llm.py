@@ -931,6 +931,19 @@ class LLMEngine:
             return True
         return False
 
+    def purge_agent(self, agent_id: str) -> None:
+        """Drop every engine-side trace of a REMOVED agent: in-memory and
+        on-disk KV checkpoints (detach keeps them on purpose — a crashed
+        or stopped agent must be resumable; a removed one must not leak
+        checkpoint files into state_root/kv_ckpt)."""
+        self._ckpts.pop(agent_id, None)
+        p = self._ckpt_path(agent_id)
+        if os.path.exists(p):
+            try:
+                os.remove(p)
+            except OSError:
+                pass
+
     def pause(self, agent_id: str) -> None:
         b = self._binding(agent_id)
         if b is not None:

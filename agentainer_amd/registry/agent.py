@@ -222,6 +222,8 @@ class Manager:
             agent = self.get(agent_id)
             if self.engine.is_attached(agent.id):
                 self.engine.detach(agent.id, offload_kv=False)
+            if hasattr(self.engine, "purge_agent"):
+                self.engine.purge_agent(agent.id)  # drop KV checkpoints
             if request_manager is not None:
                 request_manager.purge_agent(agent.id)
             self.store.delete(f"agent:{agent.id}")

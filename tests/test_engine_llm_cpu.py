@@ -285,3 +285,43 @@ def test_chunked_prefill_streams_and_truncates(tmp_path):
         assert st == 200  # kept serving through resets
     finally:
         rt.shutdown()
+
+
+def test_no_page_or_slot_leaks_across_lifecycle(llm_runtime):
+    """Pages and sequence slots return to the pool across repeated
+    chat -> stop -> resume -> remove cycles (leak regression: refcounted
+    prefix pages + checkpoint restore paths both allocate)."""
+    rt = llm_runtime
+    seed = rt.agents.deploy(name="leak-seed", model="tiny-llama",
+                            system_prompt="always reply in fewer than ten words",
+                            sampling={"max_tokens": 4})
+    rt.agents.start(seed.id)
+    rt.agent_request(seed.id, "POST", "/chat", body={"message": "warm"})
+    inst = rt.engine._instances["tiny-llama"]
+    free0 = inst.kvm.free_pages
+    slots0 = len(inst.kvm._free_slots)
+    for i in range(3):
+        a = rt.agents.deploy(name=f"leak-{i}", model="tiny-llama",
+                             system_prompt="always reply in fewer than ten words",
+                             sampling={"max_tokens": 4})
+        rt.agents.start(a.id)
+        rt.agent_request(a.id, "POST", "/chat", body={"message": f"x{i}"})
+        rt.agents.stop(a.id)     # offload to checkpoint
+        rt.agents.resume(a.id)   # restore (fresh pages)
+        rt.agent_request(a.id, "POST", "/chat", body={"message": f"y{i}"})
+        rt.agents.remove(a.id, request_manager=rt.requests)
+    assert inst.kvm.free_pages == free0, (free0, inst.kvm.free_pages)
+    assert len(inst.kvm._free_slots) == slots0
+    assert not any(s.startswith("agent-leak") for s in inst.kvm._seqs)
+
+
+def test_remove_purges_disk_checkpoint(llm_runtime, tmp_path):
+    rt = llm_runtime
+    import glob as _glob
+    a = _mk_agent(rt, name="purge-me")
+    rt.agent_request(a.id, "POST", "/chat", body={"message": "hi"})
+    rt.agents.stop(a.id)  # creates a disk checkpoint
+    ckpts = _glob.glob(f"{rt.engine.state_root}/kv_ckpt/{a.id}*.pt")
+    assert ckpts, "stop should persist a checkpoint"
+    rt.agents.remove(a.id, request_manager=rt.requests)
+    assert not _glob.glob(f"{rt.engine.state_root}/kv_ckpt/{a.id}*.pt")
