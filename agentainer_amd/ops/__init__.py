@@ -176,6 +176,11 @@ _SKINNY_SHAPES = {
 }
 SKINNY_NT = os.environ.get("AGENTAINER_SKINNY_NT", "0") == "1"
 SKINNY_KC = int(os.environ.get("AGENTAINER_SKINNY_KC", "0"))  # 0 = tuned
+# A/B hook: extra tuned shapes as "N:K:split:kc,..." (perf experiments)
+for _spec in os.environ.get("AGENTAINER_SKINNY_SHAPES", "").split(","):
+    if _spec.strip():
+        _n, _k, _s, _c = (int(v) for v in _spec.split(":"))
+        _SKINNY_SHAPES[(_n, _k)] = (_s, _c)
 
 
 def _skinny_split(ntiles: int, K: int) -> int:
