@@ -87,3 +87,54 @@ def test_server_llama8b_concurrent_chat(tmp_path):
         assert sum(1 for e in events[:-1] if e.get("token") is not None) == 32
     finally:
         srv.terminate()
+
+
+@pytest.mark.timeout(600)
+def test_kill_and_replay_gpu(tmp_path):
+    """BASELINE config 4's crash contract on DEVICE: SIGKILL the server
+    while requests are queued against a GPU-resident model; restart on
+    the same state root; auto-restart + WAL replay regenerate every
+    pending request through the HIP decode path."""
+    root = str(tmp_path / "root")
+    port = _free_port()
+    srv = Server(root, port, device="cuda", kv_pool_gb=2.0)
+    try:
+        srv.start(timeout=300)
+        st, resp = srv.call("POST", "/agents", {
+            "name": "gpu-crash", "model": "tiny-llama", "auto_restart": True,
+            "sampling": {"max_tokens": 8}})
+        assert st == 200, resp
+        aid = resp["data"]["id"]
+        assert srv.call("POST", f"/agents/{aid}/start", timeout=300)[0] == 200
+        st, warm = srv.call("POST", f"/agent/{aid}/chat",
+                            body={"message": "warm"}, auth=False, timeout=300)
+        assert st == 200 and warm["tokens"] == 8
+        assert srv.call("POST", f"/agents/{aid}/stop")[0] == 200
+        rids = []
+        for i in range(3):
+            st, q = srv.call("POST", f"/agent/{aid}/chat",
+                             body={"message": f"q-{i}"}, auth=False)
+            assert st == 202
+            rids.append(q["data"]["request_id"])
+        srv.kill9()
+    finally:
+        srv.terminate()
+
+    srv2 = Server(root, port, device="cuda", kv_pool_gb=2.0)
+    try:
+        srv2.start(timeout=300)
+        deadline = time.time() + 240
+        done = {}
+        while time.time() < deadline and len(done) < len(rids):
+            for rid in rids:
+                if rid in done:
+                    continue
+                st, r = srv2.call("GET", f"/agents/{aid}/requests/{rid}")
+                if st == 200 and r["data"]["status"] == "completed":
+                    done[rid] = r["data"]["response"]
+            time.sleep(0.5)
+        assert len(done) == len(rids), f"{len(done)}/{len(rids)} replayed"
+        for rid in rids:
+            assert done[rid]["tokens"] == 8
+    finally:
+        srv2.terminate()
