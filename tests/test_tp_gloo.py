@@ -179,9 +179,35 @@ def _tp_worker(rank, world, port, tmpdir, result_file):
         inst.step()
         if req2.done.is_set():
             break
+    # COW prefix sharing under TP: adoption is plan-driven, so both ranks
+    # must adopt the same pages in lockstep; a divergence would corrupt
+    # the all-reduced activations and show up as a token mismatch
+    sp = "shared tp system prompt driving prefix adoption across ranks"
+    pf_toks = []
+    pags = []
+    for name in ("tp-p1", "tp-p2"):
+        ag = man.deploy(name=name, model=TP_MODEL, system_prompt=sp,
+                        sampling={"max_tokens": MAX_NEW})
+        man.start(ag.id)
+        pags.append(ag)
+    pk = inst.binding(pags[0].id).prefix_tokens or []
+    for ag in pags:
+        rq = GenRequest(agent_id=ag.id, prompt_tokens=list(pk) + PROMPT,
+                        max_new=MAX_NEW, temperature=0.0, top_p=1.0, seed=0)
+        bb = inst.binding(ag.id)
+        with inst._lock:
+            bb.queue.put(rq)
+            inst._pump_agent(bb)
+        for _ in range(MAX_NEW + 6):
+            inst.step()
+            if rq.done.is_set():
+                break
+        pf_toks.append(list(rq.generated))
     torch.save({"tokens": req.generated, "err": req.error,
                 "tokens2": req2.generated, "err2": req2.error,
-                "len_after": inst.kvm.seq_len(a.id)}, result_file)
+                "len_after": inst.kvm.seq_len(a.id),
+                "pfx_tokens": pf_toks, "pfx_len": len(pk),
+                "pfx_refs": bool(inst.kvm._refs)}, result_file)
     eng.shutdown()  # broadcasts shutdown to the worker
     dist.barrier()
 
@@ -234,3 +260,8 @@ def test_tp2_matches_full_model(tmp_path):
     assert res["tokens"] == want, f"TP tokens {res['tokens']} != full {want}"
     assert len(res["tokens2"]) == MAX_NEW  # post-restore turn completed
     assert res["len_after"] > len(PROMPT) + MAX_NEW  # KV restored + extended
+    # prefix sharing stayed in lockstep across ranks
+    assert res["pfx_len"] >= 16  # at least one whole page shared
+    assert res["pfx_refs"] is True
+    assert (res["pfx_tokens"][0] == res["pfx_tokens"][1]
+            and len(res["pfx_tokens"][0]) == MAX_NEW), res["pfx_tokens"]
