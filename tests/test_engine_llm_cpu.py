@@ -325,3 +325,25 @@ def test_remove_purges_disk_checkpoint(llm_runtime, tmp_path):
     assert ckpts, "stop should persist a checkpoint"
     rt.agents.remove(a.id, request_manager=rt.requests)
     assert not _glob.glob(f"{rt.engine.state_root}/kv_ckpt/{a.id}*.pt")
+
+
+def test_multi_model_cohosting(llm_runtime):
+    """Two model FAMILIES share one engine/GPU: agents on tiny-llama and
+    tiny-mixtral serve interleaved, each instance with its own KV pool
+    and scheduler; lifecycle ops on one family don't disturb the other."""
+    rt = llm_runtime
+    a = _mk_agent(rt, name="llama-side")
+    m = rt.agents.deploy(name="moe-side", model="tiny-mixtral",
+                         sampling={"max_tokens": 4})
+    rt.agents.start(m.id)
+    r1 = rt.agent_request(a.id, "POST", "/chat", body={"message": "one"})
+    r2 = rt.agent_request(m.id, "POST", "/chat", body={"message": "one"})
+    assert r1[0] == 200 and r2[0] == 200
+    assert set(rt.engine._instances) == {"tiny-llama", "tiny-mixtral"}
+    # stop the llama agent; the mixtral agent keeps serving
+    rt.agents.stop(a.id)
+    r3 = rt.agent_request(m.id, "POST", "/chat", body={"message": "two"})
+    assert r3[0] == 200
+    # engine stats report both models
+    st = rt.engine.stats()
+    assert {"tiny-llama", "tiny-mixtral"} <= set(st["models"])
