@@ -1,0 +1,101 @@
+"""Verify the hypothesized operand layout of
+mfma_scale_f32_16x16x128_f8f6f4 for fp8 (fmt 0) and fp4 (fmt 4).
+Run on a GPU box: python tools/mx_probe.py
+"""
+import ctypes
+import os
+import subprocess
+import sys
+
+import torch
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+SO = os.path.join(HERE, "_mx_probe.so")
+
+FP4_VALS = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+            -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0]  # e2m1
+
+
+def build():
+    src = os.path.join(HERE, "mx_probe.hip")
+    if os.path.exists(SO) and os.path.getmtime(SO) > os.path.getmtime(src):
+        return
+    subprocess.check_call(["/opt/rocm/bin/hipcc", "-O2",
+                           "--offload-arch=gfx950", "-shared", "-fPIC",
+                           src, "-o", SO])
+
+
+def pack4(vec):
+    out = torch.zeros(32, dtype=torch.uint8)
+    codes = [FP4_VALS.index(round(v, 1)) for v in vec.tolist()]
+    for i in range(16):
+        out[i] = codes[2 * i] | (codes[2 * i + 1] << 4)
+    return out
+
+
+def run_case(lib, A, B, sa, sb, fmt):
+    a_frag = torch.zeros(64, 32, dtype=torch.uint8)
+    b_frag = torch.zeros(64, 32, dtype=torch.uint8)
+    for lane in range(64):
+        row, k0 = lane % 16, (lane // 16) * 32
+        av = A[row, k0:k0 + 32]
+        bv = B[k0:k0 + 32, row].contiguous()
+        if fmt == 0:
+            a_frag[lane] = av.to(torch.float8_e4m3fn).view(torch.uint8)
+            b_frag[lane] = bv.to(torch.float8_e4m3fn).view(torch.uint8)
+        else:
+            a_frag[lane, :16] = pack4(av)
+            b_frag[lane, :16] = pack4(bv)
+    dev = "cuda"
+    c_d = torch.zeros(16, 16, dtype=torch.float32, device=dev)
+    a_d, b_d = a_frag.to(dev), b_frag.to(dev)
+    sa_d, sb_d = sa.to(dev).int(), sb.to(dev).int()
+    lib.mx_probe_launch(
+        ctypes.c_void_p(c_d.data_ptr()), ctypes.c_void_p(a_d.data_ptr()),
+        ctypes.c_void_p(b_d.data_ptr()), ctypes.c_void_p(sa_d.data_ptr()),
+        ctypes.c_void_p(sb_d.data_ptr()), ctypes.c_int(fmt))
+    return c_d.cpu()
+
+
+def main():
+    build()
+    lib = ctypes.CDLL(SO)
+    torch.manual_seed(0)
+    # ---- fp8, unit scales (e8m0 127 = 2^0)
+    A = (torch.randn(16, 128) * 0.5)
+    B = (torch.randn(128, 16) * 0.5)
+    Aq = A.to(torch.float8_e4m3fn).float()
+    Bq = B.to(torch.float8_e4m3fn).float()
+    ones = torch.full((64,), 127, dtype=torch.int32)
+    C = run_case(lib, Aq, Bq, ones, ones, 0)
+    ref = Aq @ Bq
+    err = (C - ref).abs().max().item()
+    print(f"fp8 unit-scale: max abs err {err:.4f} "
+          f"{'PASS' if err < 1e-2 else 'FAIL'}")
+    # ---- fp8, per-block scales: scale byte should multiply the lane block
+    sa = torch.tensor([127 + (l // 16) for l in range(64)], dtype=torch.int32)
+    C2 = run_case(lib, Aq, Bq, sa, ones, 0)
+    # hypothesis: block kb of A scaled by 2^(kb)
+    refs = torch.zeros(16, 16)
+    for kb in range(4):
+        refs += (Aq[:, kb * 32:(kb + 1) * 32] * (2.0 ** kb)) @ \
+                Bq[kb * 32:(kb + 1) * 32, :]
+    err2 = (C2 - refs).abs().max().item()
+    print(f"fp8 block-scale: max abs err {err2:.4f} "
+          f"{'PASS' if err2 < 5e-2 else 'FAIL'}")
+    # ---- fp4 with unit scales
+    idx = torch.randint(0, 16, (16, 128))
+    A4 = torch.tensor([[FP4_VALS[int(i)] for i in row] for row in idx])
+    idxb = torch.randint(0, 16, (128, 16))
+    B4 = torch.tensor([[FP4_VALS[int(i)] for i in row] for row in idxb])
+    C4 = run_case(lib, A4, B4, ones, ones, 4)
+    ref4 = A4 @ B4
+    err4 = (C4 - ref4).abs().max().item()
+    print(f"fp4 unit-scale: max abs err {err4:.4f} "
+          f"{'PASS' if err4 < 1e-2 else 'FAIL'}")
+    if err4 >= 1e-2:
+        print("fp4 sample got", C4[0, :4].tolist(), "want", ref4[0, :4].tolist())
+
+
+if __name__ == "__main__":
+    main()
