@@ -26,7 +26,7 @@ def build():
 
 
 def pack4(vec):
-    out = torch.zeros(32, dtype=torch.uint8)
+    out = torch.zeros(16, dtype=torch.uint8)
     codes = [FP4_VALS.index(round(v, 1)) for v in vec.tolist()]
     for i in range(16):
         out[i] = codes[2 * i] | (codes[2 * i + 1] << 4)
