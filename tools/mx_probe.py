@@ -72,17 +72,30 @@ def main():
     err = (C - ref).abs().max().item()
     print(f"fp8 unit-scale: max abs err {err:.4f} "
           f"{'PASS' if err < 1e-2 else 'FAIL'}")
-    # ---- fp8, per-block scales: scale byte should multiply the lane block
-    sa = torch.tensor([127 + (l // 16) for l in range(64)], dtype=torch.int32)
+    # ---- scale-encoding sweep: put e8m0=128 (x2) in each byte of the
+    # 32-bit scale operand for A then B; a live position doubles C
+    for side in ("A", "B"):
+        for bp in range(4):
+            v = 128 << (8 * bp) | (127 | (127 << 8) | (127 << 16) | (127 << 24)) \
+                & ~(0xFF << (8 * bp))
+            sv = torch.full((64,), v, dtype=torch.int64).int()
+            Cs = run_case(lib, Aq, Bq, sv if side == "A" else ones,
+                          ones if side == "A" else sv, 0)
+            ratio = (Cs / C).nanmean().item()
+            print(f"scale {side} byte{bp}: mean C-ratio {ratio:.3f}")
+    # lane-block attribution: x2 only on lanes 0-15 (A side, byte0)
+    sa = ones.clone()
+    sa[:16] = 128
     C2 = run_case(lib, Aq, Bq, sa, ones, 0)
-    # hypothesis: block kb of A scaled by 2^(kb)
-    refs = torch.zeros(16, 16)
     for kb in range(4):
-        refs += (Aq[:, kb * 32:(kb + 1) * 32] * (2.0 ** kb)) @ \
-                Bq[kb * 32:(kb + 1) * 32, :]
-    err2 = (C2 - refs).abs().max().item()
-    print(f"fp8 block-scale: max abs err {err2:.4f} "
-          f"{'PASS' if err2 < 5e-2 else 'FAIL'}")
+        refs = Aq.clone()
+        refs[:, kb * 32:(kb + 1) * 32] *= 2.0
+        e = ((C2 - refs @ Bq).abs().max().item())
+        print(f"lanes0-15 scale -> k-block {kb}: err {e:.3f}")
+    rows = Aq.clone()
+    # alt hypothesis: lanes 0-15 scale ROWS 0-15? (row = l%16 covers all)
+    e = ((C2 - (Aq * 2) @ Bq).abs().max().item())
+    print(f"lanes0-15 scale -> all A: err {e:.3f}")
     # ---- fp4 with unit scales
     idx = torch.randint(0, 16, (16, 128))
     A4 = torch.tensor([[FP4_VALS[int(i)] for i in row] for row in idx])
