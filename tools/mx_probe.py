@@ -83,6 +83,45 @@ def main():
                           ones if side == "A" else sv, 0)
             ratio = (Cs / C).nanmean().item()
             print(f"scale {side} byte{bp}: mean C-ratio {ratio:.3f}")
+    # ---- full scale-lane map: A = ones, B = block indicator so
+    # C[i, j] = 32 * sA(row i, kb = j % 4); doubling lane L's scale
+    # reveals exactly which (row, kb) it controls. Then symmetric for B.
+    A1 = torch.ones(16, 128)
+    Bi = torch.zeros(128, 16)
+    for k in range(128):
+        for j in range(16):
+            if (k // 32) == j % 4:
+                Bi[k, j] = 1.0
+    base = run_case(lib, A1, Bi, ones, ones, 0)
+    amap = {}
+    for L in range(64):
+        sv = ones.clone()
+        sv[L] = 128
+        CL = run_case(lib, A1, Bi, sv, ones, 0)
+        hits = ((CL - base).abs() > 1.0).nonzero().tolist()
+        cells = sorted({(i, j % 4) for i, j in hits})
+        amap[L] = cells
+    print("A-scale map (lane -> (row, kb)):")
+    for L in range(64):
+        print(f"  {L:2d}: {amap[L]}")
+    Ai = torch.zeros(16, 128)
+    for k in range(128):
+        for i in range(16):
+            if (k // 32) == i % 4:
+                Ai[i, k] = 1.0
+    B1 = torch.ones(128, 16)
+    baseb = run_case(lib, Ai, B1, ones, ones, 0)
+    bmap = {}
+    for L in range(64):
+        sv = ones.clone()
+        sv[L] = 128
+        CL = run_case(lib, Ai, B1, ones, sv, 0)
+        hits = ((CL - baseb).abs() > 1.0).nonzero().tolist()
+        cells = sorted({(i % 4, j) for i, j in hits})
+        bmap[L] = cells
+    print("B-scale map (lane -> (kb, col)):")
+    for L in range(64):
+        print(f"  {L:2d}: {bmap[L]}")
     # lane-block attribution: x2 only on lanes 0-15 (A side, byte0)
     sa = ones.clone()
     sa[:16] = 128
