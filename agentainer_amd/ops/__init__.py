@@ -279,6 +279,77 @@ def linear_fp8(x: torch.Tensor, w_packed: torch.Tensor, sw: torch.Tensor,
     return out
 
 
+# e2m1 (fp4) code values, index = code (bit 3 = sign)
+_FP4_GRID = torch.tensor([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0])
+# operand-role combo for the scaled MFMA (validated on hardware via
+# tests/test_gpu_ops.py::test_mxfp4_gemm; override for probing only)
+MXFP4_COMBO = int(os.environ.get("AGENTAINER_MXFP4_COMBO", "0"))
+
+
+def quantize_weight_mxfp4(w: torch.Tensor):
+    """Offline MXFP4 quantization of a [N, K] weight for the block-scaled
+    MFMA expert GEMM (gfx950 v_mfma_scale_f32_16x16x128_f8f6f4): e2m1
+    codes, 2 per byte, with one e8m0 scale per (row, 64-element k-block)
+    — the scale granularity the instruction's scale lanes expose
+    (tools/mx_probe.py). Returns (packed codes [N*K/2] u8 fragment-linear,
+    scales [N/16 * K/128 * 32] u8 e8m0, kernel-ordered)."""
+    N, K = w.shape
+    assert N % 16 == 0 and K % 128 == 0, "MXFP4: N%16==0, K%128==0"
+    wf = w.float().cpu().view(N, K // 64, 64)
+    amax = wf.abs().amax(dim=-1).clamp(min=1e-8)
+    # e8m0 scale 2^e with amax/2^e <= 6 (e2m1 max)
+    e = torch.ceil(torch.log2(amax / 6.0)).clamp(-127, 127)
+    scale = torch.pow(2.0, e)
+    q = wf / scale.unsqueeze(-1)
+    idx = (q.abs().unsqueeze(-1) - _FP4_GRID).abs().argmin(dim=-1)
+    codes = (idx + torch.where(q < 0, 8, 0)).to(torch.uint8).view(N, K)
+    # pack 2 codes/byte along k (lo nibble = even k)
+    by = (codes[:, 0::2] | (codes[:, 1::2] << 4))  # [N, K/2]
+    # fragment-linear: lane l of (tile, kc) holds cols... B lane map:
+    # B[(l/16)*32 + j][l%16] -> W[tile*16 + l%16][kc*128 + (l/16)*32 + j],
+    # 32 codes = 16 bytes per lane
+    byv = by.view(N // 16, 16, K // 128, 4, 16)     # [T][col][kc][q][16B]
+    packed = byv.permute(0, 2, 3, 1, 4).contiguous().view(-1)
+    # scales: sc[tile][kc][half*16 + col] e8m0-biased
+    e8 = (e + 127).clamp(0, 254).to(torch.uint8).view(N, K // 64)
+    scv = e8.view(N // 16, 16, K // 128, 2).permute(0, 2, 3, 1).contiguous()
+    return packed.to(w.device), scv.view(-1).to(w.device)
+
+
+def dequantize_mxfp4(packed: torch.Tensor, scales: torch.Tensor,
+                     N: int, K: int) -> torch.Tensor:
+    """CPU reference inverse of quantize_weight_mxfp4 (tests/oracles)."""
+    p = packed.cpu().view(N // 16, K // 128, 4, 16, 16)
+    by = p.permute(0, 3, 1, 2, 4).contiguous().view(N, K // 2)
+    lo = (by & 0xF).long()
+    hi = (by >> 4).long()
+    codes = torch.stack([lo, hi], dim=-1).view(N, K)
+    vals = _FP4_GRID[codes & 7] * torch.where(codes >= 8, -1.0, 1.0)
+    sc = scales.cpu().view(N // 16, K // 128, 2, 16).permute(0, 3, 1, 2)
+    sc = sc.contiguous().view(N, K // 64).float()
+    sc = torch.pow(2.0, sc - 127)
+    return (vals.view(N, K // 64, 64) * sc.unsqueeze(-1)).view(N, K)
+
+
+def linear_mxfp4(x: torch.Tensor, w_packed: torch.Tensor,
+                 w_scales: torch.Tensor, N: int) -> torch.Tensor:
+    """x[M,K] bf16 @ MXFP4 W^T: per-row fp8 activation quant + the
+    block-scaled 16x16x128 MFMA — quarter the weight bytes of bf16."""
+    M, K = x.shape
+    mod = _dispatch("skinny_gemm_mxfp4", x)
+    assert mod is not None, "linear_mxfp4 is GPU-only"
+    xc = x.contiguous()
+    x8 = torch.empty(M, K, dtype=torch.uint8, device=x.device)
+    sx = torch.empty(M, dtype=torch.float32, device=x.device)
+    mod.quant_fp8_rows(x8, sx, xc)
+    split = _skinny_split(N // 64, K)
+    ws = _skinny_ws(x.device, N, split)
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    mod.skinny_gemm_mxfp4(out, x8, sx, w_packed, w_scales, N, K, ws, split,
+                          MXFP4_COMBO)
+    return out
+
+
 def gather_kv_pages(dst, k_cache, v_cache, page_ids):
     mod = _dispatch("gather_kv_pages", k_cache)
     if mod:

@@ -33,6 +33,9 @@ void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
                         torch::Tensor w_packed, long N, long K,
                         torch::Tensor ws, long split, bool nt, long kc);
 void quant_fp8_rows(torch::Tensor x8, torch::Tensor sx, torch::Tensor x);
+void skinny_gemm_mxfp4(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
+                       torch::Tensor w_packed, torch::Tensor w_scales, long N,
+                       long K, torch::Tensor ws, long split, long combo);
 void skinny_gemm_fp8(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
                      torch::Tensor w_packed, torch::Tensor sw, long N, long K,
                      torch::Tensor ws, long split);
@@ -57,6 +60,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_kv_pages", &gather_kv_pages, "pages -> host-shaped buffer");
   m.def("skinny_gemm", &skinny_gemm, "out[M,N] = x[M,K] @ W[N,K]^T, M<=64");
   m.def("skinny_gemm_packed", &skinny_gemm_packed, "packed-weight skinny GEMM");
+  m.def("skinny_gemm_mxfp4", &skinny_gemm_mxfp4, "MXFP4 block-scaled expert GEMM");
   m.def("quant_fp8_rows", &quant_fp8_rows, "per-row bf16 -> e4m3 quant");
   m.def("skinny_gemm_fp8", &skinny_gemm_fp8, "fp8 MFMA skinny GEMM");
   m.def("mfma_probe", &mfma_probe, "single 16x16x32 bf16 MFMA on prepacked fragments");

@@ -553,6 +553,204 @@ __global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
 
 }  // namespace
 
+
+// ---------------------------------------------------------------------------
+// MXFP4 expert GEMM (gfx950 block-scaled MFMA): W stored as e2m1 codes
+// (2/byte) with one e8m0 scale per (col, 64-k block); activations e4m3
+// per-row (quant_fp8_rows). One v_mfma_scale_f32_16x16x128_f8f6f4 per
+// 128-k slice — QUARTER the weight bytes of bf16. Operand lane maps were
+// established empirically (tools/mx_probe.py):
+//   A (fp8): lane l holds A[l%16][(l/16)*32 .. +32]  (32 bytes)
+//   B (fp4): lane l holds B[(l/16)*32 + j][l%16], 32 codes = 16 bytes in
+//            the LOW half of the 8xi32 operand
+//   scales:  byte 0 of the scale int; lanes 0..31 cover (row|col = l%16,
+//            64-elem k-half = l/16); lanes 32..63 mirror (fed 127 = x1)
+//   C:       lane l holds rows (l/16)*4 + r, col l%16
+
+namespace {
+
+typedef int i32x8v __attribute__((ext_vector_type(8)));
+typedef int i32x4v __attribute__((ext_vector_type(4)));
+
+template <bool SPLIT, int COMBO>
+__global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
+    void* __restrict__ out, const unsigned char* __restrict__ x8,
+    const float* __restrict__ sx, const unsigned char* __restrict__ wp_packed,
+    const unsigned char* __restrict__ wsc,  // [T][K/128][2][16] e8m0
+    int M, int N, int K, int k_per_split) {
+  const int n_tile = blockIdx.x * 4 + threadIdx.x / WAVE;
+  const int split = blockIdx.y;
+  const int lane = threadIdx.x % WAVE;
+  const int l16 = lane % 16;
+  const int lg = lane / 16;
+  const int k0 = split * k_per_split;
+  const int k1 = min(k0 + k_per_split, K);
+  const int KC128 = K / 128;
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) acc[ms] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // per-(tile, k128) fragment: lane-major 16 B codes
+  const unsigned char* wp =
+      wp_packed + (((long)n_tile * KC128 + k0 / 128) * 64 + lane) * 16;
+  const unsigned char* scp = wsc + ((long)n_tile * KC128 + k0 / 128) * 32;
+
+  constexpr int KC = 256;
+  constexpr int XS = KC + 16;
+  __shared__ unsigned char x_lds[2][64 * XS];
+  const int s_row = threadIdx.x % 64;
+  const int s_col0 = (threadIdx.x / 64) * 8;
+  const unsigned char* s_xp = x8 + (long)min(s_row, M - 1) * K;
+  const bool s_alive = s_row < M;
+
+  uint2 st[8];
+  auto stage_load = [&](int k) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      st[i] = uint2{0, 0};
+      if (s_alive)
+        st[i] = *reinterpret_cast<const uint2*>(s_xp + k + s_col0 + 32 * i);
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      *reinterpret_cast<uint2*>(&x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
+  };
+
+  i32x4v bw_cur[2], bw_nxt[2];
+  int sc_cur[2], sc_nxt[2];
+  auto w_load = [&](i32x4v (&dst)[2], int (&sc)[2], long chunk) {
+#pragma unroll
+    for (int f = 0; f < 2; ++f) {
+      dst[f] = __builtin_nontemporal_load(reinterpret_cast<const i32x4v*>(
+          wp + (chunk * 2 + f) * 1024));
+      sc[f] = (lane < 32) ? (int)scp[(chunk * 2 + f) * 32 + lane] : 127;
+    }
+  };
+
+  auto compute = [&](i32x4v (&bw)[2], int (&sc)[2], int buf) {
+#pragma unroll
+    for (int f = 0; f < 2; ++f) {
+      i32x8v b8 = {bw[f][0], bw[f][1], bw[f][2], bw[f][3], 0, 0, 0, 0};
+#pragma unroll
+      for (int ms = 0; ms < 4; ++ms) {
+        i32x8v a8 = *reinterpret_cast<const i32x8v*>(
+            &x_lds[buf][(ms * 16 + l16) * XS + f * 128 + lg * 32]);
+        if constexpr (COMBO == 0)
+          acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              a8, b8, acc[ms], 0, 4, 0, 127, 0, sc[f]);
+        else if constexpr (COMBO == 1)
+          acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              a8, b8, acc[ms], 4, 0, 0, 127, 0, sc[f]);
+        else if constexpr (COMBO == 2)
+          acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              a8, b8, acc[ms], 0, 4, 0, sc[f], 0, 127);
+        else
+          acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              a8, b8, acc[ms], 4, 0, 0, sc[f], 0, 127);
+      }
+    }
+  };
+
+  stage_load(k0);
+  stage_write(0);
+  w_load(bw_cur, sc_cur, 0);
+  __syncthreads();
+
+  int buf = 0;
+  long chunk = 1;
+  int k = k0;
+  while (true) {
+    bool has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_nxt, sc_nxt, chunk++);
+    }
+    compute(bw_cur, sc_cur, buf);
+    if (has_next) stage_write(buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+    k += KC;
+    if (!has_next) break;
+    has_next = (k + KC < k1);
+    if (has_next) {
+      stage_load(k + KC);
+      w_load(bw_cur, sc_cur, chunk++);
+    }
+    compute(bw_nxt, sc_nxt, buf);
+    if (has_next) stage_write(buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+    k += KC;
+    if (!has_next) break;
+  }
+
+  const int col = n_tile * 16 + l16;
+#pragma unroll
+  for (int ms = 0; ms < 4; ++ms) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = ms * 16 + lg * 4 + r;
+      if (row >= M) continue;
+      const float v = acc[ms][r] * sx[row];
+      if (SPLIT) {
+        reinterpret_cast<float*>(out)[((long)split * M + row) * N + col] = v;
+      } else {
+        reinterpret_cast<short*>(out)[(long)row * N + col] = f2bits(v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void skinny_gemm_mxfp4(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
+                       torch::Tensor w_packed, torch::Tensor w_scales, long N,
+                       long K, torch::Tensor ws, long split, long combo) {
+  TORCH_CHECK(x8.is_contiguous() && w_packed.is_contiguous());
+  TORCH_CHECK(x8.scalar_type() == at::kByte &&
+              w_packed.scalar_type() == at::kByte &&
+              w_scales.scalar_type() == at::kByte);
+  const int M = x8.size(0);
+  TORCH_CHECK(x8.size(1) == K && K % 256 == 0 && N % 64 == 0 && M <= 64);
+  TORCH_CHECK(w_packed.numel() == N * K / 2, "mxfp4 pack numel");
+  TORCH_CHECK(w_scales.numel() == (N / 16) * (K / 128) * 32, "mxfp4 scales");
+  TORCH_CHECK(split >= 1 && (K % (256 * split)) == 0);
+  auto stream = at::hip::getCurrentHIPStream();
+  const int ntiles = (int)N / 64;
+  const int kps = (int)K / (int)split;
+#define MX_LAUNCH(SPLIT_, COMBO_, OUTP)                                        \
+  hipLaunchKernelGGL((skinny_gemm_mxfp4_kernel<SPLIT_, COMBO_>),               \
+                     dim3(ntiles, SPLIT_ ? (int)split : 1), dim3(256), 0,      \
+                     stream, OUTP, x8.data_ptr<uint8_t>(),                     \
+                     sx.data_ptr<float>(), w_packed.data_ptr<uint8_t>(),       \
+                     w_scales.data_ptr<uint8_t>(), M, (int)N, (int)K, kps)
+#define MX_COMBO(SPLIT_, OUTP)                                                 \
+  do {                                                                         \
+    switch (combo) {                                                           \
+      case 1: MX_LAUNCH(SPLIT_, 1, OUTP); break;                               \
+      case 2: MX_LAUNCH(SPLIT_, 2, OUTP); break;                               \
+      case 3: MX_LAUNCH(SPLIT_, 3, OUTP); break;                               \
+      default: MX_LAUNCH(SPLIT_, 0, OUTP); break;                              \
+    }                                                                          \
+  } while (0)
+  if (split == 1) {
+    MX_COMBO(false, out.data_ptr());
+  } else {
+    TORCH_CHECK(ws.numel() >= (long)split * M * N &&
+                ws.scalar_type() == at::kFloat);
+    MX_COMBO(true, ws.data_ptr());
+    const long mn = (long)M * N;
+    hipLaunchKernelGGL(splitk_combine_kernel, dim3((mn + 1023) / 1024),
+                       dim3(256), 0, stream, (short*)out.data_ptr(),
+                       ws.data_ptr<float>(), mn, split);
+  }
+#undef MX_COMBO
+#undef MX_LAUNCH
+}
+
 void quant_fp8_rows(torch::Tensor x8, torch::Tensor sx, torch::Tensor x) {
   TORCH_CHECK(x.is_contiguous() && x.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x8.scalar_type() == at::kByte && sx.scalar_type() == at::kFloat);
