@@ -31,6 +31,12 @@ extern "C" __global__ void mx_gemm_16x16x128(
   } else if (fmt == 4) {
     acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
         a, b, acc, 4, 4, 0, sa[lane], 0, sb[lane]);
+  } else if (fmt == 10) {  // mixed: A fp8, B fp4 via (cbsz=0, blgp=4)
+    acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc, 0, 4, 0, sa[lane], 0, sb[lane]);
+  } else if (fmt == 11) {  // mixed: A fp8, B fp4 via (cbsz=4, blgp=0)
+    acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc, 4, 0, 0, sa[lane], 0, sb[lane]);
   }
   const int col = lane % 16;
   const int r0 = (lane / 16) * 4;

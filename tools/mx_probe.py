@@ -141,6 +141,34 @@ def main():
     idxb = torch.randint(0, 16, (128, 16))
     B4 = torch.tensor([[FP4_VALS[int(i)] for i in row] for row in idxb])
     C4 = run_case(lib, A4, B4, ones, ones, 4)
+    # ---- mixed A fp8 x B fp4, both fmt-arg assignments, B lo/hi half
+    refm = Aq @ B4
+    for fmt in (10, 11):
+        for half in ("lo", "hi"):
+            a_frag = torch.zeros(64, 32, dtype=torch.uint8)
+            b_frag = torch.zeros(64, 32, dtype=torch.uint8)
+            for lane in range(64):
+                row, k0 = lane % 16, (lane // 16) * 32
+                a_frag[lane] = Aq[row, k0:k0 + 32].to(
+                    torch.float8_e4m3fn).view(torch.uint8)
+                pb = pack4(B4[k0:k0 + 32, row].contiguous())
+                if half == "lo":
+                    b_frag[lane, :16] = pb
+                else:
+                    b_frag[lane, 16:] = pb
+            dev = "cuda"
+            c_d = torch.zeros(16, 16, dtype=torch.float32, device=dev)
+            lib.mx_probe_launch(
+                ctypes.c_void_p(c_d.data_ptr()),
+                ctypes.c_void_p(a_frag.to(dev).data_ptr()),
+                ctypes.c_void_p(b_frag.to(dev).data_ptr()),
+                ctypes.c_void_p(ones.to(dev).int().data_ptr()),
+                ctypes.c_void_p(ones.to(dev).int().data_ptr()),
+                ctypes.c_int(fmt))
+            torch.cuda.synchronize()
+            err = (c_d.cpu() - refm).abs().max().item()
+            print(f"mixed fmt={fmt} B-{half}: max err {err:.4f} "
+                  f"{'PASS' if err < 5e-2 else ''}")
     ref4 = A4 @ B4
     err4 = (C4 - ref4).abs().max().item()
     print(f"fp4 unit-scale: max abs err {err4:.4f} "
