@@ -333,15 +333,19 @@ def dequantize_mxfp4(packed: torch.Tensor, scales: torch.Tensor,
 
 def linear_mxfp4(x: torch.Tensor, w_packed: torch.Tensor,
                  w_scales: torch.Tensor, N: int) -> torch.Tensor:
-    """x[M,K] bf16 @ MXFP4 W^T: per-row fp8 activation quant + the
-    block-scaled 16x16x128 MFMA — quarter the weight bytes of bf16."""
+    """x[M,K] bf16 @ MXFP4 W^T: per-row e2m1 activation quant + the
+    block-scaled 16x16x128 fp4 MFMA — quarter the weight bytes of bf16.
+    (Mixed fp8-activation x fp4-weight through the scaled MFMA NaNs on
+    this silicon/compiler — tools/mx_probe.py — so both operands run
+    e2m1; activations carry a per-row f32 scale applied in the
+    epilogue.)"""
     M, K = x.shape
     mod = _dispatch("skinny_gemm_mxfp4", x)
     assert mod is not None, "linear_mxfp4 is GPU-only"
     xc = x.contiguous()
-    x8 = torch.empty(M, K, dtype=torch.uint8, device=x.device)
+    x8 = torch.empty(M, K // 2, dtype=torch.uint8, device=x.device)
     sx = torch.empty(M, dtype=torch.float32, device=x.device)
-    mod.quant_fp8_rows(x8, sx, xc)
+    mod.quant_fp4_rows(x8, sx, xc)
     split = _skinny_split(N // 64, K)
     ws = _skinny_ws(x.device, N, split)
     out = torch.empty(M, N, dtype=x.dtype, device=x.device)

@@ -33,6 +33,7 @@ void skinny_gemm_packed(torch::Tensor out, torch::Tensor x,
                         torch::Tensor w_packed, long N, long K,
                         torch::Tensor ws, long split, bool nt, long kc);
 void quant_fp8_rows(torch::Tensor x8, torch::Tensor sx, torch::Tensor x);
+void quant_fp4_rows(torch::Tensor x4, torch::Tensor sx, torch::Tensor x);
 void skinny_gemm_mxfp4(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
                        torch::Tensor w_packed, torch::Tensor w_scales, long N,
                        long K, torch::Tensor ws, long split, long combo);
@@ -61,6 +62,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, "out[M,N] = x[M,K] @ W[N,K]^T, M<=64");
   m.def("skinny_gemm_packed", &skinny_gemm_packed, "packed-weight skinny GEMM");
   m.def("skinny_gemm_mxfp4", &skinny_gemm_mxfp4, "MXFP4 block-scaled expert GEMM");
+  m.def("quant_fp4_rows", &quant_fp4_rows, "per-row e2m1 activation quant");
   m.def("quant_fp8_rows", &quant_fp8_rows, "per-row bf16 -> e4m3 quant");
   m.def("skinny_gemm_fp8", &skinny_gemm_fp8, "fp8 MFMA skinny GEMM");
   m.def("mfma_probe", &mfma_probe, "single 16x16x32 bf16 MFMA on prepacked fragments");

@@ -554,6 +554,70 @@ __global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
 }  // namespace
 
 
+namespace {
+
+// e2m1 encode: nearest of {0,.5,1,1.5,2,3,4,6} with sign bit 3
+__device__ __forceinline__ unsigned char f2e2m1(float v) {
+  const float a = fabsf(v);
+  unsigned char c;
+  if (a < 0.25f) c = 0;
+  else if (a < 0.75f) c = 1;
+  else if (a < 1.25f) c = 2;
+  else if (a < 1.75f) c = 3;
+  else if (a < 2.5f) c = 4;
+  else if (a < 3.5f) c = 5;
+  else if (a < 5.0f) c = 6;
+  else c = 7;
+  return c | (v < 0.f ? 8 : 0);
+}
+
+// per-row dynamic fp4 quantization: x bf16 [M,K] -> x4 [M,K/2] (2 codes
+// per byte, lo nibble = even k) + sx f32 [M] with amax -> 6.0
+__global__ __launch_bounds__(256) void quant_fp4_rows_kernel(
+    unsigned char* __restrict__ x4, float* __restrict__ sx,
+    const short* __restrict__ x, int K) {
+  const int row = blockIdx.x;
+  const short* xp = x + (long)row * K;
+  float mx = 0.f;
+  for (int i = threadIdx.x * 8; i < K; i += blockDim.x * 8) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(xp + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) mx = fmaxf(mx, fabsf(bits2f(v[j])));
+  }
+  mx = wave_max(mx);
+  __shared__ float red[4];
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x / 64] = mx;
+  __syncthreads();
+  mx = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+  const float scale = fmaxf(mx, 1e-8f) / 6.f;  // e2m1 max = 6
+  const float inv = 1.0f / scale;
+  if (threadIdx.x == 0) sx[row] = scale;
+  for (int i = threadIdx.x * 8; i < K; i += blockDim.x * 8) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(xp + i);
+    unsigned char o[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      o[j] = f2e2m1(bits2f(v[2 * j]) * inv) |
+             (f2e2m1(bits2f(v[2 * j + 1]) * inv) << 4);
+    *reinterpret_cast<unsigned int*>(x4 + ((long)row * K + i) / 2) =
+        *reinterpret_cast<unsigned int*>(o);
+  }
+}
+
+}  // namespace
+
+void quant_fp4_rows(torch::Tensor x4, torch::Tensor sx, torch::Tensor x) {
+  TORCH_CHECK(x.is_contiguous() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x4.scalar_type() == at::kByte && sx.scalar_type() == at::kFloat);
+  const int M = x.size(0), K = x.size(1);
+  TORCH_CHECK(K % 16 == 0 && x4.numel() == (long)M * K / 2);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (M == 0) return;
+  hipLaunchKernelGGL(quant_fp4_rows_kernel, dim3(M), dim3(256), 0, stream,
+                     x4.data_ptr<uint8_t>(), sx.data_ptr<float>(),
+                     (const short*)x.data_ptr(), K);
+}
+
 // ---------------------------------------------------------------------------
 // MXFP4 expert GEMM (gfx950 block-scaled MFMA): W stored as e2m1 codes
 // (2/byte) with one e8m0 scale per (col, 64-k block); activations e4m3
@@ -574,7 +638,7 @@ typedef int i32x4v __attribute__((ext_vector_type(4)));
 
 template <bool SPLIT, int COMBO>
 __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
-    void* __restrict__ out, const unsigned char* __restrict__ x8,
+    void* __restrict__ out, const unsigned char* __restrict__ x4,
     const float* __restrict__ sx, const unsigned char* __restrict__ wp_packed,
     const unsigned char* __restrict__ wsc,  // [T][K/128][2][16] e8m0
     int M, int N, int K, int k_per_split) {
@@ -596,27 +660,29 @@ __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
       wp_packed + (((long)n_tile * KC128 + k0 / 128) * 64 + lane) * 16;
   const unsigned char* scp = wsc + ((long)n_tile * KC128 + k0 / 128) * 32;
 
-  constexpr int KC = 256;
-  constexpr int XS = KC + 16;
+  constexpr int KC = 256;              // k per chunk (codes: KC/2 bytes)
+  constexpr int XS = KC / 2 + 8;       // bytes per LDS row
   __shared__ unsigned char x_lds[2][64 * XS];
   const int s_row = threadIdx.x % 64;
-  const int s_col0 = (threadIdx.x / 64) * 8;
-  const unsigned char* s_xp = x8 + (long)min(s_row, M - 1) * K;
+  const int s_col0 = (threadIdx.x / 64) * 4;   // byte cols, stride 16
+  const unsigned char* s_xp = x4 + (long)min(s_row, M - 1) * (K / 2);
   const bool s_alive = s_row < M;
 
-  uint2 st[8];
+  unsigned int st[8];
   auto stage_load = [&](int k) {
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
-      st[i] = uint2{0, 0};
+      st[i] = 0;
       if (s_alive)
-        st[i] = *reinterpret_cast<const uint2*>(s_xp + k + s_col0 + 32 * i);
+        st[i] = *reinterpret_cast<const unsigned int*>(
+            s_xp + k / 2 + s_col0 + 16 * i);
     }
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < 8; ++i)
-      *reinterpret_cast<uint2*>(&x_lds[buf][s_row * XS + s_col0 + 32 * i]) = st[i];
+      *reinterpret_cast<unsigned int*>(
+          &x_lds[buf][s_row * XS + s_col0 + 16 * i]) = st[i];
   };
 
   i32x4v bw_cur[2], bw_nxt[2];
@@ -636,20 +702,16 @@ __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
       i32x8v b8 = {bw[f][0], bw[f][1], bw[f][2], bw[f][3], 0, 0, 0, 0};
 #pragma unroll
       for (int ms = 0; ms < 4; ++ms) {
-        i32x8v a8 = *reinterpret_cast<const i32x8v*>(
-            &x_lds[buf][(ms * 16 + l16) * XS + f * 128 + lg * 32]);
-        if constexpr (COMBO == 0)
+        // A fp4: lane's 32 codes = 16 bytes, LOW half of the operand
+        const i32x4v a4 = *reinterpret_cast<const i32x4v*>(
+            &x_lds[buf][(ms * 16 + l16) * XS + f * 64 + lg * 16]);
+        i32x8v a8 = {a4[0], a4[1], a4[2], a4[3], 0, 0, 0, 0};
+        if constexpr (COMBO == 1)
           acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              a8, b8, acc[ms], 0, 4, 0, 127, 0, sc[f]);
-        else if constexpr (COMBO == 1)
-          acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              a8, b8, acc[ms], 4, 0, 0, 127, 0, sc[f]);
-        else if constexpr (COMBO == 2)
-          acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              a8, b8, acc[ms], 0, 4, 0, sc[f], 0, 127);
+              a8, b8, acc[ms], 4, 4, 0, sc[f], 0, 127);
         else
           acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              a8, b8, acc[ms], 4, 0, 0, sc[f], 0, 127);
+              a8, b8, acc[ms], 4, 4, 0, 127, 0, sc[f]);
       }
     }
   };
@@ -714,7 +776,7 @@ void skinny_gemm_mxfp4(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
               w_packed.scalar_type() == at::kByte &&
               w_scales.scalar_type() == at::kByte);
   const int M = x8.size(0);
-  TORCH_CHECK(x8.size(1) == K && K % 256 == 0 && N % 64 == 0 && M <= 64);
+  TORCH_CHECK(x8.size(1) == K / 2 && K % 256 == 0 && N % 64 == 0 && M <= 64);
   TORCH_CHECK(w_packed.numel() == N * K / 2, "mxfp4 pack numel");
   TORCH_CHECK(w_scales.numel() == (N / 16) * (K / 128) * 32, "mxfp4 scales");
   TORCH_CHECK(split >= 1 && (K % (256 * split)) == 0);
@@ -729,12 +791,8 @@ void skinny_gemm_mxfp4(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
                      w_scales.data_ptr<uint8_t>(), M, (int)N, (int)K, kps)
 #define MX_COMBO(SPLIT_, OUTP)                                                 \
   do {                                                                         \
-    switch (combo) {                                                           \
-      case 1: MX_LAUNCH(SPLIT_, 1, OUTP); break;                               \
-      case 2: MX_LAUNCH(SPLIT_, 2, OUTP); break;                               \
-      case 3: MX_LAUNCH(SPLIT_, 3, OUTP); break;                               \
-      default: MX_LAUNCH(SPLIT_, 0, OUTP); break;                              \
-    }                                                                          \
+    if (combo == 1) MX_LAUNCH(SPLIT_, 1, OUTP);                                \
+    else MX_LAUNCH(SPLIT_, 0, OUTP);                                          \
   } while (0)
   if (split == 1) {
     MX_COMBO(false, out.data_ptr());
