@@ -696,6 +696,11 @@ __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
     }
   };
 
+  // the scale operands must BOTH be VGPRs: with a literal/SGPR in either
+  // slot the compiler selects an encoding whose scales are inert
+  // (probe-verified; tools/mx_diag.py)
+  int unit_sc;
+  asm volatile("v_mov_b32 %0, 127" : "=v"(unit_sc));
   auto compute = [&](i32x4v (&bw)[2], int (&sc)[2], int buf) {
 #pragma unroll
     for (int f = 0; f < 2; ++f) {
@@ -708,10 +713,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
         i32x8v a8 = {a4[0], a4[1], a4[2], a4[3], 0, 0, 0, 0};
         if constexpr (COMBO == 1)
           acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              a8, b8, acc[ms], 4, 4, 0, sc[f], 0, 127);
+              a8, b8, acc[ms], 4, 4, 0, sc[f], 0, unit_sc);
         else
           acc[ms] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-              a8, b8, acc[ms], 4, 4, 0, 127, 0, sc[f]);
+              a8, b8, acc[ms], 4, 4, 0, unit_sc, 0, sc[f]);
       }
     }
   };
