@@ -141,6 +141,25 @@ def main():
     idxb = torch.randint(0, 16, (128, 16))
     B4 = torch.tensor([[FP4_VALS[int(i)] for i in row] for row in idxb])
     C4 = run_case(lib, A4, B4, ones, ones, 4)
+    # ---- fp4 scale-lane map (scales were only probed under fp8 fmt)
+    idx = torch.randint(1, 8, (16, 128))  # nonzero magnitudes
+    A4m = torch.tensor([[FP4_VALS[int(i)] for i in row] for row in idx])
+    B4i = torch.zeros(128, 16)
+    for k in range(128):
+        for j in range(16):
+            if (k // 32) == j % 4:
+                B4i[k, j] = 1.0
+    base4 = run_case(lib, A4m, B4i, ones, ones, 4)
+    amap4 = {}
+    for L in range(64):
+        sv = ones.clone()
+        sv[L] = 128
+        CL = run_case(lib, A4m, B4i, sv, ones, 4)
+        hits = ((CL - base4).abs() > 0.9).nonzero().tolist()
+        amap4[L] = sorted({(i, j % 4) for i, j in hits})
+    print("fp4 A-scale map:")
+    for L in range(64):
+        print(f"  {L:2d}: {amap4[L]}")
     # ---- mixed A fp8 x B fp4, both fmt-arg assignments, B lo/hi half
     refm = Aq @ B4
     for fmt in (10, 11):
