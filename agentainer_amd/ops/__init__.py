@@ -295,7 +295,7 @@ def quantize_weight_mxfp4(w: torch.Tensor):
     scales [N/16 * K/128 * 32] u8 e8m0, kernel-ordered)."""
     N, K = w.shape
     assert N % 16 == 0 and K % 128 == 0, "MXFP4: N%16==0, K%128==0"
-    wf = w.float().cpu().view(N, K // 64, 64)
+    wf = w.float().cpu().view(N, K // 32, 32)   # true MX block-32
     amax = wf.abs().amax(dim=-1).clamp(min=1e-8)
     # e8m0 scale 2^e with amax/2^e <= 6 (e2m1 max)
     e = torch.ceil(torch.log2(amax / 6.0)).clamp(-127, 127)
@@ -305,14 +305,15 @@ def quantize_weight_mxfp4(w: torch.Tensor):
     codes = (idx + torch.where(q < 0, 8, 0)).to(torch.uint8).view(N, K)
     # pack 2 codes/byte along k (lo nibble = even k)
     by = (codes[:, 0::2] | (codes[:, 1::2] << 4))  # [N, K/2]
-    # fragment-linear: lane l of (tile, kc) holds cols... B lane map:
+    # fragment-linear: B lane map (probed):
     # B[(l/16)*32 + j][l%16] -> W[tile*16 + l%16][kc*128 + (l/16)*32 + j],
     # 32 codes = 16 bytes per lane
     byv = by.view(N // 16, 16, K // 128, 4, 16)     # [T][col][kc][q][16B]
     packed = byv.permute(0, 2, 3, 1, 4).contiguous().view(-1)
-    # scales: sc[tile][kc][half*16 + col] e8m0-biased
-    e8 = (e + 127).clamp(0, 254).to(torch.uint8).view(N, K // 64)
-    scv = e8.view(N // 16, 16, K // 128, 2).permute(0, 2, 3, 1).contiguous()
+    # scales: one e8m0 per (col, 32-block); scale lane l covers
+    # (kb = l/16, col = l%16) -> scv[T][kc][kb][col], 64 B per fragment
+    e8 = (e + 127).clamp(0, 254).to(torch.uint8).view(N, K // 32)
+    scv = e8.view(N // 16, 16, K // 128, 4).permute(0, 2, 3, 1).contiguous()
     return packed.to(w.device), scv.view(-1).to(w.device)
 
 
@@ -325,10 +326,10 @@ def dequantize_mxfp4(packed: torch.Tensor, scales: torch.Tensor,
     hi = (by >> 4).long()
     codes = torch.stack([lo, hi], dim=-1).view(N, K)
     vals = _FP4_GRID[codes & 7] * torch.where(codes >= 8, -1.0, 1.0)
-    sc = scales.cpu().view(N // 16, K // 128, 2, 16).permute(0, 3, 1, 2)
-    sc = sc.contiguous().view(N, K // 64).float()
+    sc = scales.cpu().view(N // 16, K // 128, 4, 16).permute(0, 3, 1, 2)
+    sc = sc.contiguous().view(N, K // 32).float()
     sc = torch.pow(2.0, sc - 127)
-    return (vals.view(N, K // 64, 64) * sc.unsqueeze(-1)).view(N, K)
+    return (vals.view(N, K // 32, 32) * sc.unsqueeze(-1)).view(N, K)
 
 
 def linear_mxfp4(x: torch.Tensor, w_packed: torch.Tensor,

@@ -640,7 +640,7 @@ template <bool SPLIT, int COMBO>
 __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
     void* __restrict__ out, const unsigned char* __restrict__ x4,
     const float* __restrict__ sx, const unsigned char* __restrict__ wp_packed,
-    const unsigned char* __restrict__ wsc,  // [T][K/128][2][16] e8m0
+    const unsigned char* __restrict__ wsc,  // [T][K/128][4][16] e8m0
     int M, int N, int K, int k_per_split) {
   const int n_tile = blockIdx.x * 4 + threadIdx.x / WAVE;
   const int split = blockIdx.y;
@@ -658,7 +658,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
   // per-(tile, k128) fragment: lane-major 16 B codes
   const unsigned char* wp =
       wp_packed + (((long)n_tile * KC128 + k0 / 128) * 64 + lane) * 16;
-  const unsigned char* scp = wsc + ((long)n_tile * KC128 + k0 / 128) * 32;
+  const unsigned char* scp = wsc + ((long)n_tile * KC128 + k0 / 128) * 64;
 
   constexpr int KC = 256;              // k per chunk (codes: KC/2 bytes)
   constexpr int XS = KC / 2 + 8;       // bytes per LDS row
@@ -692,7 +692,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_mxfp4_kernel(
     for (int f = 0; f < 2; ++f) {
       dst[f] = __builtin_nontemporal_load(reinterpret_cast<const i32x4v*>(
           wp + (chunk * 2 + f) * 1024));
-      sc[f] = (lane < 32) ? (int)scp[(chunk * 2 + f) * 32 + lane] : 127;
+      // fp4 scale-lane map is bijective: lane l = (kb l/16, col l%16)
+      sc[f] = (int)scp[(chunk * 2 + f) * 64 + lane];
     }
   };
 
@@ -783,7 +784,7 @@ void skinny_gemm_mxfp4(torch::Tensor out, torch::Tensor x8, torch::Tensor sx,
   const int M = x8.size(0);
   TORCH_CHECK(x8.size(1) == K / 2 && K % 256 == 0 && N % 64 == 0 && M <= 64);
   TORCH_CHECK(w_packed.numel() == N * K / 2, "mxfp4 pack numel");
-  TORCH_CHECK(w_scales.numel() == (N / 16) * (K / 128) * 32, "mxfp4 scales");
+  TORCH_CHECK(w_scales.numel() == (N / 16) * (K / 128) * 64, "mxfp4 scales");
   TORCH_CHECK(split >= 1 && (K % (256 * split)) == 0);
   auto stream = at::hip::getCurrentHIPStream();
   const int ntiles = (int)N / 64;
