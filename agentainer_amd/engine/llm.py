@@ -128,7 +128,8 @@ class ModelInstance:
                 if "expert_fp8" in inspect.signature(
                         self.model.pack_decode_weights).parameters:
                     self.model.pack_decode_weights(
-                        expert_fp8=bool(engine_cfg.get("expert_fp8", False)))
+                        expert_fp8=bool(engine_cfg.get("expert_fp8", False)),
+                        expert_fp4=bool(engine_cfg.get("expert_fp4", False)))
                 else:
                     self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))

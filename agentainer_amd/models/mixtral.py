@@ -74,6 +74,9 @@ class MixtralMoE(torch.nn.Module):
         # fp8 expert path (config 5): (packed bytes, scales) per expert
         self.gate_up_fp8: List[Optional[tuple]] = [None] * self.n_local
         self.down_fp8: List[Optional[tuple]] = [None] * self.n_local
+        # MXFP4 expert path (block-scaled fp4 MFMA): quarter the bytes
+        self.gate_up_fp4: List[Optional[tuple]] = [None] * self.n_local
+        self.down_fp4: List[Optional[tuple]] = [None] * self.n_local
 
     # prefill batches at least this big use token-shuffle all-to-all EP
     A2A_MIN_TOKENS = 128
@@ -92,9 +95,13 @@ class MixtralMoE(torch.nn.Module):
         weight.scatter_(1, topi, topv)
         out = torch.zeros_like(h, dtype=torch.float32)
         use_fp8 = (h.is_cuda and T <= 64 and self.gate_up_fp8[0] is not None)
+        use_fp4 = (h.is_cuda and T <= 64 and self.gate_up_fp4[0] is not None)
         for i in range(self.n_local):
             e = self.e0 + i
-            if use_fp8:
+            if use_fp4:
+                wp, sw = self.gate_up_fp4[i]
+                gu = ops.linear_mxfp4(h, wp, sw, 2 * self.inter)
+            elif use_fp8:
                 wp, sw = self.gate_up_fp8[i]
                 gu = ops.linear_fp8(h, wp, sw, 2 * self.inter)
             else:
@@ -102,7 +109,10 @@ class MixtralMoE(torch.nn.Module):
             gate, up = gu[:, :self.inter], gu[:, self.inter:]
             act = torch.empty(T, self.inter, dtype=gu.dtype, device=gu.device)
             ops.silu_mul(act, gate, up)
-            if use_fp8:
+            if use_fp4:
+                wp, sw = self.down_fp4[i]
+                eo = ops.linear_mxfp4(act, wp, sw, h.size(1))
+            elif use_fp8:
                 wp, sw = self.down_fp8[i]
                 eo = ops.linear_fp8(act, wp, sw, h.size(1))
             else:
@@ -277,12 +287,18 @@ class MixtralForCausalLM(torch.nn.Module):
                 p.fill_(1.0)
 
     @torch.no_grad()
-    def pack_decode_weights(self, expert_fp8: bool = False):
+    def pack_decode_weights(self, expert_fp8: bool = False,
+                            expert_fp4: bool = False):
         for layer in self.layers:
             layer.attn.qkv_packed = ops.pack_weight(layer.attn.qkv_proj.data)
             layer.attn.o_packed = ops.pack_weight(layer.attn.o_proj.data)
             for i in range(layer.moe.n_local):
-                if expert_fp8:
+                if expert_fp4:
+                    layer.moe.gate_up_fp4[i] = ops.quantize_weight_mxfp4(
+                        layer.moe.gate_up[i].data)
+                    layer.moe.down_fp4[i] = ops.quantize_weight_mxfp4(
+                        layer.moe.down[i].data)
+                elif expert_fp8:
                     layer.moe.gate_up_fp8[i] = ops.quantize_weight_fp8(
                         layer.moe.gate_up[i].data)
                     layer.moe.down_fp8[i] = ops.quantize_weight_fp8(

@@ -57,6 +57,8 @@ def parse_args():
                    help="fp8 MFMA expert decode GEMMs (Mixtral, config 5)")
     p.add_argument("--kv-fp8", action="store_true",
                    help="OCP e4m3 KV cache (half the KV bytes, 2x agents)")
+    p.add_argument("--expert-fp4", action="store_true",
+                   help="MXFP4 block-scaled expert GEMMs (quarter bytes)")
     return p.parse_args()
 
 
@@ -141,6 +143,7 @@ def main():
     cfg.data["engine"]["max_batch_tokens"] = args.max_batch_tokens
     cfg.data["engine"]["max_decode_batch"] = max(args.agents, 1)
     cfg.data["engine"]["expert_fp8"] = bool(args.expert_fp8)
+    cfg.data["engine"]["expert_fp4"] = bool(args.expert_fp4)
     if args.kv_fp8:
         cfg.data["engine"]["kv_dtype"] = "fp8"
     store = Store(os.path.join(tmp, "state"), sync="interval")
@@ -219,7 +222,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no number
-            "dtype": ("bf16+fp8exp" if args.expert_fp8 else
+            "dtype": ("bf16+fp4exp" if args.expert_fp4 else
+                      "bf16+fp8exp" if args.expert_fp8 else
                       "bf16+fp8kv" if args.kv_fp8 else "bf16"),
             "data": "synthetic",
             "config": {
