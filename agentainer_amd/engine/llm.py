@@ -118,18 +118,22 @@ class ModelInstance:
         if weights_path:
             self.model.load_safetensors(weights_path)
         if device.startswith("cuda"):
-            # packed decode copies double weight memory; skip when the
-            # model alone already claims a large share of HBM (70B TP=1)
+            # packed decode copies add memory on top of the bf16 weights:
+            # x1 for a bf16 repack, x0.5 for fp8 experts, x0.25 for MXFP4 —
+            # size the guard by what is actually being packed (the old
+            # flat x2 guard silently skipped quantization for Mixtral-8x7B)
             param_bytes = sum(p.numel() * p.element_size()
                               for p in self.model.parameters())
+            fp8 = bool(engine_cfg.get("expert_fp8", False))
+            fp4 = bool(engine_cfg.get("expert_fp4", False))
+            factor = 0.25 if fp4 else 0.5 if fp8 else 1.0
             free, total = torch.cuda.mem_get_info()
-            if param_bytes * 2 < 0.6 * total:
+            if param_bytes * (1.0 + factor) < 0.75 * total:
                 import inspect
                 if "expert_fp8" in inspect.signature(
                         self.model.pack_decode_weights).parameters:
-                    self.model.pack_decode_weights(
-                        expert_fp8=bool(engine_cfg.get("expert_fp8", False)),
-                        expert_fp4=bool(engine_cfg.get("expert_fp4", False)))
+                    self.model.pack_decode_weights(expert_fp8=fp8,
+                                                   expert_fp4=fp4)
                 else:
                     self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))

@@ -295,13 +295,17 @@ def quantize_weight_mxfp4(w: torch.Tensor):
     scales [N/16 * K/128 * 32] u8 e8m0, kernel-ordered)."""
     N, K = w.shape
     assert N % 16 == 0 and K % 128 == 0, "MXFP4: N%16==0, K%128==0"
-    wf = w.float().cpu().view(N, K // 32, 32)   # true MX block-32
+    wf = w.float().view(N, K // 32, 32)          # true MX block-32
     amax = wf.abs().amax(dim=-1).clamp(min=1e-8)
     # e8m0 scale 2^e with amax/2^e <= 6 (e2m1 max)
     e = torch.ceil(torch.log2(amax / 6.0)).clamp(-127, 127)
     scale = torch.pow(2.0, e)
     q = wf / scale.unsqueeze(-1)
-    idx = (q.abs().unsqueeze(-1) - _FP4_GRID).abs().argmin(dim=-1)
+    # nearest e2m1 via bucketize on the midpoints (device-native: this
+    # runs at model load for up to 512 Mixtral expert matrices)
+    mids = torch.tensor([0.25, 0.75, 1.25, 1.75, 2.5, 3.5, 5.0],
+                        device=wf.device)
+    idx = torch.bucketize(q.abs().contiguous(), mids)
     codes = (idx + torch.where(q < 0, 8, 0)).to(torch.uint8).view(N, K)
     # pack 2 codes/byte along k (lo nibble = even k)
     by = (codes[:, 0::2] | (codes[:, 1::2] << 4))  # [N, K/2]
