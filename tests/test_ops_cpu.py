@@ -176,3 +176,25 @@ def test_gather_scatter_pages_roundtrip():
     ops.scatter_kv_pages(k2, v2, buf, ids)
     assert torch.equal(k2[1], k_cache[1]) and torch.equal(v2[3], v_cache[3])
     assert k2[0].abs().sum() == 0
+
+
+def test_mxfp4_pack_roundtrip():
+    """quantize_weight_mxfp4 <-> dequantize_mxfp4: every reconstructed
+    value is within the e2m1 grid resolution of its 32-block's scale."""
+    torch.manual_seed(3)
+    N, K = 64, 512
+    w = (torch.randn(N, K) * 0.07).to(torch.bfloat16)
+    packed, scales = ops.quantize_weight_mxfp4(w)
+    assert packed.numel() == N * K // 2
+    assert scales.numel() == (N // 16) * (K // 128) * 64
+    wd = ops.dequantize_mxfp4(packed, scales, N, K)
+    blocks = w.float().view(N, K // 32, 32)
+    err = (wd.view(N, K // 32, 32) - blocks).abs().amax(-1)
+    bound = blocks.abs().amax(-1).clamp(min=1e-8) * 0.26 + 1e-6
+    assert bool((err <= bound).all()), float((err / bound).max())
+    # exactly representable values roundtrip exactly
+    grid = torch.tensor([0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0, -6.0, -0.5])
+    w2 = grid[torch.randint(0, 9, (16, 128))].to(torch.bfloat16)
+    w2.view(16, 4, 32)[:, :, 0] = 6.0  # pin block scale to 2^0
+    p2, s2 = ops.quantize_weight_mxfp4(w2)
+    assert torch.equal(ops.dequantize_mxfp4(p2, s2, 16, 128), w2.float())
