@@ -535,7 +535,7 @@ def test_mxfp4_gemm_matches_dequant_reference():
         assert rel < 2e-2, (M, N, K, rel)
 
 
-def test_mixtral_fp4_experts_generate(gpu_tmp_engine=None, tmp_path=None):
+def test_mixtral_fp4_experts_generate():
     """Mixtral decode with MXFP4 experts completes generations."""
     import tempfile
 
