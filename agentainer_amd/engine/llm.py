@@ -126,14 +126,18 @@ class ModelInstance:
                               for p in self.model.parameters())
             fp8 = bool(engine_cfg.get("expert_fp8", False))
             fp4 = bool(engine_cfg.get("expert_fp4", False))
-            factor = 0.25 if fp4 else 0.5 if fp8 else 1.0
+            dq = str(engine_cfg.get("dense_quant", "") or "")
+            factor = (0.25 if (fp4 or dq == "mxfp4")
+                      else 0.5 if (fp8 or dq == "fp8") else 1.0)
             free, total = torch.cuda.mem_get_info()
             if param_bytes * (1.0 + factor) < 0.75 * total:
                 import inspect
-                if "expert_fp8" in inspect.signature(
-                        self.model.pack_decode_weights).parameters:
+                sig = inspect.signature(self.model.pack_decode_weights)
+                if "expert_fp8" in sig.parameters:
                     self.model.pack_decode_weights(expert_fp8=fp8,
                                                    expert_fp4=fp4)
+                elif "dense_quant" in sig.parameters:
+                    self.model.pack_decode_weights(dense_quant=dq)
                 else:
                     self.model.pack_decode_weights()
         page_size = int(engine_cfg.get("kv_page_size", 16))

@@ -89,10 +89,16 @@ class LlamaAttention(torch.nn.Module):
             if cfg.qkv_bias else None)
         self.qkv_packed = None
         self.o_packed = None
+        # opt-in quantized decode weights ("turbo": engine.dense_quant)
+        self.qkv_q = None
+        self.o_q = None
 
     def forward(self, h, k_cache, v_cache, md: AttnMetadata, cos_sin, tp_group=None):
         T = h.size(0)
-        qkv = ops.linear(h, self.qkv_proj, self.qkv_packed)
+        if self.qkv_q is not None and T <= 64 and h.is_cuda:
+            qkv = ops.linear_quant(h, self.qkv_q)
+        else:
+            qkv = ops.linear(h, self.qkv_proj, self.qkv_packed)
         if self.qkv_bias is not None:
             qkv = qkv + self.qkv_bias
         q_sz = self.n_heads * self.head_dim
@@ -113,7 +119,10 @@ class LlamaAttention(torch.nn.Module):
         else:
             ops.paged_decode_attention(out, q, k_cache, v_cache, md.page_table,
                                        md.seq_lens, self.scale)
-        o = ops.linear(out.view(T, q_sz), self.o_proj, self.o_packed)
+        if self.o_q is not None and T <= 64 and h.is_cuda:
+            o = ops.linear_quant(out.view(T, q_sz), self.o_q)
+        else:
+            o = ops.linear(out.view(T, q_sz), self.o_proj, self.o_packed)
         if tp_group is not None:
             torch.distributed.all_reduce(o, group=tp_group)
         return o
@@ -130,14 +139,23 @@ class LlamaMLP(torch.nn.Module):
         self.inter = inter
         self.gate_up_packed = None
         self.down_packed = None
+        self.gate_up_q = None
+        self.down_q = None
 
     def forward(self, h, tp_group=None):
-        gu = ops.linear(h, self.gate_up, self.gate_up_packed)
+        turbo = self.gate_up_q is not None and h.size(0) <= 64 and h.is_cuda
+        if turbo:
+            gu = ops.linear_quant(h, self.gate_up_q)
+        else:
+            gu = ops.linear(h, self.gate_up, self.gate_up_packed)
         gate, up = gu[:, :self.inter], gu[:, self.inter:]
         act = torch.empty(gu.size(0), self.inter, dtype=gu.dtype,
                           device=gu.device)
         ops.silu_mul(act, gate, up)
-        out = ops.linear(act, self.down, self.down_packed)
+        if turbo:
+            out = ops.linear_quant(act, self.down_q)
+        else:
+            out = ops.linear(act, self.down, self.down_packed)
         if tp_group is not None:
             torch.distributed.all_reduce(out, group=tp_group)
         return out
@@ -224,16 +242,39 @@ class LlamaForCausalLM(torch.nn.Module):
                           getattr(self, 'lm_head_packed', None))  # bf16
 
     @torch.no_grad()
-    def pack_decode_weights(self):
+    def pack_decode_weights(self, dense_quant: str = ""):
         """Pre-shuffle every projection weight into the fragment-linear
         layout the skinny decode GEMM streams (ops.pack_weight). Keeps the
         natural-layout Parameters for the prefill hipBLASLt path — the
-        packed copies double weight memory, well inside 288 GB HBM3E."""
+        packed copies double weight memory, well inside 288 GB HBM3E.
+
+        dense_quant: "" (bf16, the default/headline dtype), "fp8"
+        (per-channel e4m3), or "mxfp4" (e2m1 + e8m0 block-32 scales) —
+        opt-in reduced-precision DECODE projections (prefill stays bf16)."""
         for layer in self.layers:
-            layer.attn.qkv_packed = ops.pack_weight(layer.attn.qkv_proj.data)
-            layer.attn.o_packed = ops.pack_weight(layer.attn.o_proj.data)
-            layer.mlp.gate_up_packed = ops.pack_weight(layer.mlp.gate_up.data)
-            layer.mlp.down_packed = ops.pack_weight(layer.mlp.down.data)
+            if dense_quant == "mxfp4":
+                for mod_, wname, qname in (
+                        (layer.attn, "qkv_proj", "qkv_q"),
+                        (layer.attn, "o_proj", "o_q"),
+                        (layer.mlp, "gate_up", "gate_up_q"),
+                        (layer.mlp, "down", "down_q")):
+                    w = getattr(mod_, wname).data
+                    p, sc = ops.quantize_weight_mxfp4(w)
+                    setattr(mod_, qname, ("mxfp4", p, sc, w.size(0)))
+            elif dense_quant == "fp8":
+                for mod_, wname, qname in (
+                        (layer.attn, "qkv_proj", "qkv_q"),
+                        (layer.attn, "o_proj", "o_q"),
+                        (layer.mlp, "gate_up", "gate_up_q"),
+                        (layer.mlp, "down", "down_q")):
+                    w = getattr(mod_, wname).data
+                    p, sw = ops.quantize_weight_fp8(w)
+                    setattr(mod_, qname, ("fp8", p, sw, w.size(0)))
+            else:
+                layer.attn.qkv_packed = ops.pack_weight(layer.attn.qkv_proj.data)
+                layer.attn.o_packed = ops.pack_weight(layer.attn.o_proj.data)
+                layer.mlp.gate_up_packed = ops.pack_weight(layer.mlp.gate_up.data)
+                layer.mlp.down_packed = ops.pack_weight(layer.mlp.down.data)
         self.lm_head_packed = ops.pack_weight(
             self.lm_head.data if isinstance(self.lm_head, torch.nn.Parameter)
             else self.lm_head)

@@ -359,6 +359,15 @@ def linear_mxfp4(x: torch.Tensor, w_packed: torch.Tensor,
     return out
 
 
+def linear_quant(x: torch.Tensor, qt) -> torch.Tensor:
+    """Dispatch a quantized decode GEMM: qt = (kind, packed, scales, N)
+    with kind "mxfp4" or "fp8" (engine.dense_quant / expert paths)."""
+    kind, packed, scales, N = qt
+    if kind == "mxfp4":
+        return linear_mxfp4(x, packed, scales, N)
+    return linear_fp8(x, packed, scales, N)
+
+
 def gather_kv_pages(dst, k_cache, v_cache, page_ids):
     mod = _dispatch("gather_kv_pages", k_cache)
     if mod:

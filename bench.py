@@ -59,6 +59,9 @@ def parse_args():
                    help="OCP e4m3 KV cache (half the KV bytes, 2x agents)")
     p.add_argument("--expert-fp4", action="store_true",
                    help="MXFP4 block-scaled expert GEMMs (quarter bytes)")
+    p.add_argument("--dense-quant", default="", choices=["", "fp8", "mxfp4"],
+                   help="opt-in quantized dense decode projections (turbo; "
+                        "NOT the headline dtype)")
     return p.parse_args()
 
 
@@ -144,6 +147,7 @@ def main():
     cfg.data["engine"]["max_decode_batch"] = max(args.agents, 1)
     cfg.data["engine"]["expert_fp8"] = bool(args.expert_fp8)
     cfg.data["engine"]["expert_fp4"] = bool(args.expert_fp4)
+    cfg.data["engine"]["dense_quant"] = args.dense_quant
     if args.kv_fp8:
         cfg.data["engine"]["kv_dtype"] = "fp8"
     store = Store(os.path.join(tmp, "state"), sync="interval")
@@ -222,7 +226,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no number
-            "dtype": ("bf16+fp4exp" if args.expert_fp4 else
+            "dtype": (f"bf16+{args.dense_quant}w" if args.dense_quant else
+                      "bf16+fp4exp" if args.expert_fp4 else
                       "bf16+fp8exp" if args.expert_fp8 else
                       "bf16+fp8kv" if args.kv_fp8 else "bf16"),
             "data": "synthetic",

@@ -314,3 +314,34 @@ def test_threaded_engine_chat_gpu(tmp_path):
     assert len(hist) == 1
     engine.shutdown()
     store.close()
+
+
+def test_dense_quant_turbo_generation(tmp_path):
+    """Opt-in quantized dense decode (engine.dense_quant): generations
+    complete and stay deterministic within the mode, for both fp8 and
+    MXFP4 projections."""
+    for dq in ("fp8", "mxfp4"):
+        cfg = load_config(path="/nonexistent.yaml", env={})
+        root = str(tmp_path / dq)
+        cfg.data["store"]["path"] = root
+        cfg.data["engine"]["sync_mode"] = True
+        cfg.data["engine"]["kv_pool_gb"] = 1.0
+        cfg.data["engine"]["dense_quant"] = dq
+        store = Store(root + "/state", sync="never")
+        torch.manual_seed(4)
+        engine = LLMEngine(store, cfg, device="cuda", state_root=root)
+        manager = Manager(store, engine, cfg)
+        try:
+            a = manager.deploy(name="t1", model="tiny-llama")
+            manager.start(a.id)
+            b = manager.deploy(name="t2", model="tiny-llama")
+            manager.start(b.id)
+            inst = engine._instances["tiny-llama"]
+            assert inst.model.layers[0].attn.qkv_q is not None
+            assert inst.model.layers[0].attn.qkv_q[0] == dq
+            prompt = list(range(3, 40))
+            o1 = _gen(engine, manager, a, prompt)
+            o2 = _gen(engine, manager, b, prompt)
+            assert o1 == o2 and len(o1) == 8, (dq, o1, o2)
+        finally:
+            engine.shutdown()
