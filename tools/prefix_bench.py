@@ -38,7 +38,9 @@ for sharing in (True, False):
         manager.start(a.id)
         agents.append(a)
     inst = engine._instances["llama3-8b"]
-    pk = inst.binding(agents[0].id).prefix_tokens or []
+    full = inst.tokenizer.encode(f"[system] {sysprompt}\n")
+    n_full = (len(full) // inst.kvm.page_size) * inst.kvm.page_size
+    pk = full[:n_full]  # identical prompt in BOTH modes
     reqs = []
     torch.cuda.synchronize()
     t0 = time.time()
