@@ -18,12 +18,20 @@ from agentainer_amd.store import Store
 N_AGENTS = int(sys.argv[1]) if len(sys.argv) > 1 else 64
 SYS_TOK = int(sys.argv[2]) if len(sys.argv) > 2 else 1024
 
+engine = None
 for sharing in (True, False):
+    import gc
     import tempfile
+    if engine is not None:
+        engine.shutdown()
+        del engine  # release the previous 200+ GB pool BEFORE reallocating
+        gc.collect()
+        torch.cuda.empty_cache()
     tmp = tempfile.mkdtemp(prefix=f"pfx-{sharing}-")
     cfg = load_config(path="/nonexistent.yaml", env={})
     cfg.data["store"]["path"] = tmp
     cfg.data["engine"]["sync_mode"] = True
+    cfg.data["engine"]["kv_pool_gb"] = 48.0
     cfg.data["engine"]["prefix_sharing"] = sharing
     store = Store(tmp + "/state", sync="never")
     torch.manual_seed(0)
@@ -64,4 +72,3 @@ for sharing in (True, False):
     print(f"sharing={sharing}: {N_AGENTS} agents x {len(pk)}-tok prefix | "
           f"wall {wall:.2f}s  prefill_tokens {inst.prefill_tokens}  "
           f"used_pages {inst.kvm.used_pages}")
-    engine.shutdown()
