@@ -197,3 +197,26 @@ def test_chat_sse_not_running_queues(client):
                     json={"message": "queued"})
     assert r.status_code == 202
     assert r.json()["data"]["status"] == "pending"
+
+
+def test_chat_sse_client_disconnect_leaves_pending(client):
+    """A client abandoning an SSE stream mid-generation must not lose the
+    request: the generator's cleanup leaves the WAL entry pending (no
+    ack), and the replay worker completes it later (at-least-once)."""
+    rt = client.runtime
+    a = _deploy(client, name="sse-drop")
+    client.post(f"/agents/{a['id']}/start", headers=AUTH)
+    status, gen = rt.agent_request_stream(a["id"], {"message": "dropped"})
+    assert status == 200
+    first = next(gen)  # start consuming, then abandon
+    assert "text" in first or "token" in first
+    gen.close()  # GeneratorExit inside the stream
+    # in-flight marker released; entry not acked
+    assert not rt._inflight
+    pend = rt.requests.pending(a["id"])
+    assert len(pend) == 1
+    # replay completes it
+    assert rt.replay.tick() == 1
+    assert rt.requests.pending(a["id"]) == []
+    done = rt.requests.by_queue(a["id"], "completed")
+    assert len(done) == 1 and "dropped" in done[0].response["response"]
