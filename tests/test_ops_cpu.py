@@ -198,3 +198,12 @@ def test_mxfp4_pack_roundtrip():
     w2.view(16, 4, 32)[:, :, 0] = 6.0  # pin block scale to 2^0
     p2, s2 = ops.quantize_weight_mxfp4(w2)
     assert torch.equal(ops.dequantize_mxfp4(p2, s2, 16, 128), w2.float())
+
+
+def test_linear_quant_dispatch_kinds():
+    """linear_quant routes by kind tag; CPU tensors raise loudly (these
+    paths are GPU-only by policy — no silent eager fallback)."""
+    x = torch.randn(4, 256, dtype=torch.bfloat16)
+    with pytest.raises((AssertionError, RuntimeError)):
+        ops.linear_quant(x, ("mxfp4", torch.zeros(1, dtype=torch.uint8),
+                             torch.zeros(1, dtype=torch.uint8), 64))
