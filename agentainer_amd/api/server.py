@@ -361,6 +361,43 @@ def create_app(rt: Runtime) -> FastAPI:
         return envelope(True, "", rt.logger.get_logs(
             level=level, component=component, agent_id=agent_id, limit=limit))
 
+    @app.get("/logs/stream")
+    async def logs_stream(agent_id: str = "",
+                          _tok: str = Depends(require_auth)):
+        """Live log tail over SSE (the reference's TailLogs published to a
+        channel nobody could subscribe to — here the stream is usable:
+        `agentainer logs -f`). Optional ?agent_id= filter."""
+        import queue as _q
+
+        def sse():  # sync generator — starlette iterates in a threadpool
+            lines: "_q.Queue[str]" = _q.Queue(maxsize=1000)
+
+            def on_line(line: str) -> None:
+                try:
+                    lines.put_nowait(line)
+                except _q.Full:
+                    pass  # slow client: drop rather than block the logger
+
+            unsub = rt.logger.tail(on_line)
+            try:
+                while True:
+                    try:
+                        line = lines.get(timeout=15.0)
+                    except _q.Empty:
+                        yield ": heartbeat\n\n"
+                        continue
+                    if agent_id:
+                        try:
+                            if json.loads(line).get("agent_id") != agent_id:
+                                continue
+                        except json.JSONDecodeError:
+                            continue
+                    yield f"data: {line}\n\n"
+            finally:
+                unsub()
+
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
     return app
 
 

@@ -181,11 +181,28 @@ def list_cmd(client: Client):
 
 
 @cli.command()
-@click.argument("agent_id")
+@click.argument("agent_id", required=False, default="")
 @click.option("--limit", default=100, type=int)
+@click.option("--follow", "-f", is_flag=True,
+              help="stream live log lines (SSE); ctrl-c to stop")
 @pass_client
-def logs(client: Client, agent_id: str, limit: int):
-    """Show an agent's engine log."""
+def logs(client: Client, agent_id: str, limit: int, follow: bool):
+    """Show an agent's engine log (or, with -f, tail the live stream)."""
+    if follow:
+        params = {"agent_id": agent_id} if agent_id else {}
+        with httpx.stream("GET", f"{client.url}/logs/stream", params=params,
+                          headers={"Authorization": f"Bearer {client.token}"},
+                          timeout=None) as r:
+            if r.status_code != 200:
+                click.echo(f"error ({r.status_code})", err=True)
+                sys.exit(1)
+            for line in r.iter_lines():
+                if line.startswith("data: "):
+                    click.echo(line[len("data: "):])
+        return
+    if not agent_id:
+        click.echo("error: AGENT_ID required unless --follow", err=True)
+        sys.exit(2)
     resp = client.call("GET", f"/agents/{agent_id}/logs", params={"limit": limit})
     for e in resp.get("data") or []:
         click.echo(json.dumps(e))

@@ -220,3 +220,4 @@ def test_chat_sse_client_disconnect_leaves_pending(client):
     assert rt.requests.pending(a["id"]) == []
     done = rt.requests.by_queue(a["id"], "completed")
     assert len(done) == 1 and "dropped" in done[0].response["response"]
+

@@ -285,3 +285,44 @@ def test_kill_mid_stream(tmp_path):
         assert rec["response"]["tokens"] == 300  # full regeneration
     finally:
         srv2.terminate()
+
+
+@pytest.mark.timeout(120)
+def test_logs_follow_stream(tmp_path):
+    """`GET /logs/stream` (the `agentainer logs -f` tail): lines published
+    after the subscription arrive as SSE frames over a REAL server
+    (TestClient's ASGI transport buffers, so this needs uvicorn)."""
+    import threading
+
+    root = str(tmp_path / "root")
+    srv = Server(root, _free_port())
+    try:
+        srv.start()
+        st, resp = srv.call("POST", "/agents", {"name": "tailed", "model": "echo"})
+        aid = resp["data"]["id"]
+
+        def traffic():
+            time.sleep(0.4)
+            srv.call("POST", f"/agents/{aid}/start")
+            srv.call("POST", f"/agent/{aid}/chat", body={"message": "make logs"},
+                     auth=False)
+
+        th = threading.Thread(target=traffic)
+        th.start()
+        got = []
+        deadline = time.time() + 30
+        with httpx.stream("GET", f"{srv.base}/logs/stream",
+                          headers=AUTH, timeout=30) as r:
+            assert r.status_code == 200
+            for line in r.iter_lines():
+                if time.time() > deadline:
+                    break
+                if line.startswith("data: "):
+                    got.append(json.loads(line[len("data: "):]))
+                if any(g.get("action") == "start" or
+                       "start" in str(g.get("message", "")) for g in got):
+                    break
+        th.join(timeout=10)
+        assert got, "no log lines streamed"
+    finally:
+        srv.terminate()
