@@ -1205,6 +1205,9 @@ class LLMEngine:
                 "prefill_tokens": inst.prefill_tokens,
                 "kv_pages_used": inst.kvm.used_pages,
                 "kv_pages_free": inst.kvm.free_pages,
+                "kv_pages_shared": len(inst.kvm._refs),
+                "shared_prefixes": len(inst._prefixes),
+                "kv_dtype": str(inst.kvm.dtype).replace("torch.", ""),
                 "running": len(inst.running),
                 "stuck": inst.stuck(),
                 "last_step_age_s": round(time.time() - inst.last_step_t, 3),

@@ -178,3 +178,20 @@ def test_prefix_not_adopted_for_short_prompt(tmp_path):
         assert not inst.kvm._refs
     finally:
         rt.shutdown()
+
+
+def test_stats_expose_sharing(tmp_path):
+    rt = _runtime(tmp_path, "stats", prefix_sharing=True)
+    try:
+        for n in ("s1", "s2"):
+            a = rt.agents.deploy(name=n, model="tiny-llama",
+                                 system_prompt=SYSPROMPT,
+                                 sampling={"max_tokens": 3})
+            rt.agents.start(a.id)
+            rt.agent_request(a.id, "POST", "/chat", body={"message": "x"})
+        st = rt.engine.stats()["models"]["tiny-llama"]
+        assert st["shared_prefixes"] == 1
+        assert st["kv_pages_shared"] >= 2
+        assert st["kv_dtype"] == "bfloat16"
+    finally:
+        rt.shutdown()
