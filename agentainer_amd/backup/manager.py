@@ -61,12 +61,29 @@ class BackupManager:
             "version": BACKUP_VERSION,
             "agents": [],
         }
+        kv_dir = os.path.join(self.backup_dir, backup["id"] + ".kv")
+        engine = getattr(self.manager, "engine", None)
         for a in agents:
-            backup["agents"].append({
+            entry = {
                 "config": a.to_dict(),
                 "conversations": self.store.lrange(f"agent:{a.id}:conversations"),
                 "metrics": self.store.hgetall(f"agent:{a.id}:metrics"),
-            })
+            }
+            # conversation KV snapshot (live agents offload copy-on-read):
+            # restore becomes KV-exact instead of replay-regenerated
+            if engine is not None and hasattr(engine, "export_kv"):
+                try:
+                    ckpt = engine.export_kv(a.id)
+                except Exception:  # noqa: BLE001 — backup stays best-effort
+                    ckpt = None
+                if ckpt is not None:
+                    import torch
+                    os.makedirs(kv_dir, exist_ok=True)
+                    torch.save({"length": ckpt.length, "n_pages": ckpt.n_pages,
+                                "data": ckpt.data.cpu()},
+                               os.path.join(kv_dir, a.id + ".pt"))
+                    entry["kv"] = a.id + ".pt"
+            backup["agents"].append(entry)
         with open(self._path(backup["id"]), "w", encoding="utf-8") as f:
             json.dump(backup, f)
             f.flush()
@@ -98,6 +115,18 @@ class BackupManager:
                 self.store.rpush(f"agent:{agent.id}:conversations", conv)
             for k, v in (entry.get("metrics") or {}).items():
                 self.store.hset(f"agent:{agent.id}:metrics", k, v)
+            kv_name = entry.get("kv")
+            engine = getattr(self.manager, "engine", None)
+            if kv_name and engine is not None and hasattr(engine, "import_kv"):
+                p = os.path.join(self.backup_dir, backup_id + ".kv", kv_name)
+                if os.path.exists(p):
+                    import torch
+
+                    from ..engine.kvcache import KVCheckpoint
+                    d = torch.load(p, map_location="cpu", weights_only=True)
+                    engine.import_kv(agent.id, KVCheckpoint(
+                        length=d["length"], n_pages=d["n_pages"],
+                        data=d["data"]))
             restored.append(agent)
         return restored
 
@@ -132,10 +161,16 @@ class BackupManager:
         os.unlink(path)
 
     def export(self, backup_id: str, out_path: str) -> str:
-        """Bundle a backup into one tar.gz (manager.go:396-456)."""
+        """Bundle a backup (+ any KV checkpoints) into one tar.gz
+        (manager.go:396-456)."""
         self.load(backup_id)  # existence check
         with tarfile.open(out_path, "w:gz") as tar:
             tar.add(self._path(backup_id), arcname=backup_id + ".json")
+            kv_dir = os.path.join(self.backup_dir, backup_id + ".kv")
+            if os.path.isdir(kv_dir):
+                for fn in sorted(os.listdir(kv_dir)):
+                    tar.add(os.path.join(kv_dir, fn),
+                            arcname=f"{backup_id}.kv/{fn}")
         return out_path
 
     def import_(self, tar_path: str) -> List[str]:
@@ -154,4 +189,18 @@ class BackupManager:
                 with open(self._path(data["id"]), "w", encoding="utf-8") as out:
                     json.dump(data, out)
                 ids.append(data["id"])
+            # KV checkpoint payloads: "<backup_id>.kv/<agent_id>.pt"
+            for member in tar.getmembers():
+                parts = member.name.split("/")
+                if (len(parts) != 2 or not parts[0].endswith(".kv")
+                        or not parts[1].endswith(".pt")
+                        or ".." in member.name):
+                    continue
+                f = tar.extractfile(member)
+                if f is None:
+                    continue
+                d = os.path.join(self.backup_dir, parts[0])
+                os.makedirs(d, exist_ok=True)
+                with open(os.path.join(d, parts[1]), "wb") as out:
+                    out.write(f.read())
         return ids

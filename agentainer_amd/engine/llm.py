@@ -940,6 +940,35 @@ class LLMEngine:
             return True
         return False
 
+    def export_kv(self, agent_id: str) -> Optional[KVCheckpoint]:
+        """Snapshot an agent's conversation KV WITHOUT detaching it (backup
+        path): live agents offload copy-on-read under the step mutex;
+        stopped agents return their cached/disk checkpoint."""
+        model = self._agent_model.get(agent_id)
+        if model in (None, "echo"):
+            return None
+        inst = self._instances.get(model)
+        if inst is None:
+            return None
+        b = inst.binding(agent_id)
+        if b is None:
+            return self._ckpts.get(agent_id) or self._load_disk_ckpt(agent_id)
+        if not inst.sync_mode:
+            inst._step_mutex.acquire()
+        try:
+            if inst.async_decode:
+                inst.drain_async()
+            return inst.kvm.offload(b.seq_id, free=False)
+        finally:
+            if not inst.sync_mode:
+                inst._step_mutex.release()
+
+    def import_kv(self, agent_id: str, ckpt: KVCheckpoint) -> None:
+        """Stage a checkpoint so the agent's NEXT start restores it (backup
+        restore path — restored agents begin stopped)."""
+        self._ckpts[agent_id] = ckpt
+        self._save_disk_ckpt(agent_id, ckpt)
+
     def purge_agent(self, agent_id: str) -> None:
         """Drop every engine-side trace of a REMOVED agent: in-memory and
         on-disk KV checkpoints (detach keeps them on purpose — a crashed

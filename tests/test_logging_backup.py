@@ -77,3 +77,55 @@ def test_backup_export_import(runtime, tmp_path):
     ids = runtime.backups.import_(out)
     assert ids == [b["id"]]
     assert runtime.backups.load(b["id"])["name"] == "exported"
+
+
+def test_backup_restores_kv_exact(tmp_path):
+    """Backups snapshot the conversation KV (live copy-on-read offload):
+    a restored agent continues from the SAME context it had at backup
+    time, byte-exact — not a replay regeneration."""
+    import torch
+
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import LLMEngine
+    from agentainer_amd.service import Runtime
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    root = str(tmp_path / "root")
+    cfg.data["store"]["path"] = root
+    cfg.data["engine"]["kv_pool_gb"] = 0.01
+    s = Store(root + "/state", sync="interval")
+    torch.manual_seed(0)
+    rt = Runtime(cfg, engine=LLMEngine(s, cfg, device="cpu", state_root=root),
+                 store=s, state_root=root)
+    try:
+        a = rt.agents.deploy(name="bk", model="tiny-llama",
+                             sampling={"max_tokens": 6})
+        rt.agents.start(a.id)
+        ctl = rt.agents.deploy(name="ctl", model="tiny-llama",
+                               sampling={"max_tokens": 6})
+        rt.agents.start(ctl.id)
+        rt.agent_request(a.id, "POST", "/chat", body={"message": "alpha"})
+        rt.agent_request(ctl.id, "POST", "/chat", body={"message": "alpha"})
+        bk = rt.backups.create("kv-test", agent_ids=[a.id])
+        assert bk["agents"][0].get("kv"), "live KV not snapshotted"
+        # export/import round trip carries the checkpoint payload
+        tar = str(tmp_path / "bundle.tar.gz")
+        rt.backups.export(bk["id"], tar)
+        rt.backups.delete(bk["id"])
+        import shutil
+        shutil.rmtree(f"{rt.backups.backup_dir}/{bk['id']}.kv")
+        ids = rt.backups.import_(tar)
+        assert ids == [bk["id"]]
+        restored = rt.backups.restore(bk["id"])
+        assert len(restored) == 1
+        r = restored[0]
+        rt.agents.start(r.id)
+        inst = rt.engine._instances["tiny-llama"]
+        assert inst.kvm.seq_len(r.id) > 0  # KV restored, not empty
+        # continuation equals the uninterrupted control's continuation
+        p_r = rt.agent_request(r.id, "POST", "/chat", body={"message": "beta"})[1]
+        p_c = rt.agent_request(ctl.id, "POST", "/chat", body={"message": "beta"})[1]
+        assert p_r["response"] == p_c["response"]
+    finally:
+        rt.shutdown()
