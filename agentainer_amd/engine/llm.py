@@ -953,6 +953,11 @@ class LLMEngine:
         b = inst.binding(agent_id)
         if b is None:
             return self._ckpts.get(agent_id) or self._load_disk_ckpt(agent_id)
+        if inst.tp_size > 1:
+            # a live TP snapshot would capture only rank 0's shard (the
+            # workers' shards live in their own processes); stop/resume
+            # checkpoints per-rank correctly — skip live KV here
+            return None
         if not inst.sync_mode:
             inst._step_mutex.acquire()
         try:
