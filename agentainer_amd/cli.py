@@ -317,9 +317,14 @@ def backup():
 @backup.command("create")
 @click.option("--name", required=True)
 @click.option("--description", default="")
+@click.option("--agent", "agents", multiple=True,
+              help="limit to specific agent ids (repeatable)")
 @pass_client
-def backup_create(client: Client, name, description):
-    resp = client.call("POST", "/backups", {"name": name, "description": description})
+def backup_create(client: Client, name, description, agents):
+    body = {"name": name, "description": description}
+    if agents:
+        body["agent_ids"] = list(agents)
+    resp = client.call("POST", "/backups", body)
     click.echo(json.dumps(resp.get("data")))
 
 
