@@ -88,6 +88,10 @@ class RequestManager:
     def store_request(self, agent_id: str, method: str, path: str,
                       headers: Optional[Dict[str, str]] = None, body: Any = None,
                       req_id: Optional[str] = None) -> Request:
+        if req_id is not None and self.store.get(self._rec(agent_id, req_id)):
+            # a client-supplied id (X-Agentainer-Request-ID) must never
+            # clobber an existing WAL record — fall back to a fresh uuid
+            req_id = None
         req = Request(
             id=req_id or str(uuid.uuid4()), agent_id=agent_id, method=method,
             path=path, headers=dict(headers or {}), body=body,

@@ -139,3 +139,20 @@ def test_replay_skips_inflight(runtime):
     runtime._inflight.discard(req.id)  # dispatch died (crash analog)
     assert runtime.replay.tick() == 1
     assert runtime.requests.pending(a.id) == []
+
+
+def test_client_supplied_id_cannot_clobber(runtime):
+    """A duplicate X-Agentainer-Request-ID must not overwrite an existing
+    WAL record — the second request gets a fresh id."""
+    a = _deploy_started(runtime, name="idem")
+    r1 = runtime.requests.store_request(a.id, "POST", "/chat",
+                                        body={"message": "one"},
+                                        req_id="fixed-id")
+    assert r1.id == "fixed-id"
+    runtime.requests.store_response(a.id, r1.id, {"response": "done"})
+    r2 = runtime.requests.store_request(a.id, "POST", "/chat",
+                                        body={"message": "two"},
+                                        req_id="fixed-id")
+    assert r2.id != "fixed-id"
+    kept = runtime.requests.get(a.id, "fixed-id")
+    assert kept.status == COMPLETED and kept.response == {"response": "done"}
