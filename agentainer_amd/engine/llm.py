@@ -127,6 +127,9 @@ class ModelInstance:
             fp8 = bool(engine_cfg.get("expert_fp8", False))
             fp4 = bool(engine_cfg.get("expert_fp4", False))
             dq = str(engine_cfg.get("dense_quant", "") or "")
+            if dq not in ("", "fp8", "mxfp4"):
+                raise ValueError(f"unsupported dense_quant {dq!r} "
+                                 "(expected fp8 or mxfp4)")
             factor = (0.25 if (fp4 or dq == "mxfp4")
                       else 0.5 if (fp8 or dq == "fp8") else 1.0)
             free, total = torch.cuda.mem_get_info()
