@@ -299,6 +299,43 @@ def create_app(rt: Runtime) -> FastAPI:
     async def engine_metrics(_tok: str = Depends(require_auth)):
         return envelope(True, "", rt.engine.stats())
 
+    @app.get("/metrics/prometheus")
+    async def prometheus_metrics(_tok: str = Depends(require_auth)):
+        """Engine + fleet counters in Prometheus text exposition format
+        (scrape-ready; beyond the reference, which had no metrics export)."""
+        st = await run_in_threadpool(rt.engine.stats)
+        lines = []
+
+        def emit(name, value, labels=None, mtype="gauge", help_=""):
+            if not any(ln.startswith(f"# TYPE {name} ") for ln in lines):
+                if help_:
+                    lines.append(f"# HELP {name} {help_}")
+                lines.append(f"# TYPE {name} {mtype}")
+            lab = ("{" + ",".join(f'{k}="{v}"' for k, v in labels.items()) + "}"
+                   if labels else "")
+            lines.append(f"{name}{lab} {value}")
+
+        for model, m in (st.get("models") or {}).items():
+            L = {"model": model}
+            emit("agentainer_engine_steps_total", m.get("steps", 0), L,
+                 "counter", "engine steps executed")
+            emit("agentainer_decode_tokens_total", m.get("decode_tokens", 0),
+                 L, "counter", "decode tokens sampled")
+            emit("agentainer_prefill_tokens_total", m.get("prefill_tokens", 0),
+                 L, "counter", "prompt tokens prefilled")
+            emit("agentainer_kv_pages_used", m.get("kv_pages_used", 0), L)
+            emit("agentainer_kv_pages_free", m.get("kv_pages_free", 0), L)
+            emit("agentainer_kv_pages_shared", m.get("kv_pages_shared", 0), L)
+            emit("agentainer_running_requests", m.get("running", 0), L)
+            emit("agentainer_engine_stuck", int(bool(m.get("stuck"))), L)
+        if "hbm_total_bytes" in st:
+            emit("agentainer_hbm_free_bytes", st.get("hbm_free_bytes", 0))
+            emit("agentainer_hbm_total_bytes", st.get("hbm_total_bytes", 0))
+        emit("agentainer_agents_attached", len(st.get("agents") or {}))
+        from fastapi.responses import PlainTextResponse
+        return PlainTextResponse("\n".join(lines) + "\n",
+                                 media_type="text/plain; version=0.0.4")
+
     # ---------- backups / audit / logs ----------
 
     @app.post("/backups")

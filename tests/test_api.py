@@ -221,3 +221,16 @@ def test_chat_sse_client_disconnect_leaves_pending(client):
     done = rt.requests.by_queue(a["id"], "completed")
     assert len(done) == 1 and "dropped" in done[0].response["response"]
 
+
+
+def test_prometheus_metrics_endpoint(client):
+    a = _deploy(client, name="prom")
+    client.post(f"/agents/{a['id']}/start", headers=AUTH)
+    client.post(f"/agent/{a['id']}/chat", json={"message": "hi"})
+    r = client.get(f"/metrics/prometheus?token={DEFAULT_TOKEN}")
+    assert r.status_code == 200
+    body = r.text
+    assert "agentainer_agents_attached" in body
+    assert body.count("# TYPE agentainer_agents_attached") == 1
+    # unauthenticated scrape rejected
+    assert client.get("/metrics/prometheus").status_code == 401
