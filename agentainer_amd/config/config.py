@@ -36,9 +36,16 @@ DEFAULTS: Dict[str, Dict[str, Any]] = {
         "dtype": "bf16",
         "kv_page_size": 16,        # tokens per KV page
         "kv_pool_gb": 0.0,         # 0 => auto-size from free HBM
+        "kv_dtype": "bf16",        # fp8 = e4m3 pages (2x agents per GPU)
         "max_batch_tokens": 8192,  # prefill token budget per step
         "max_decode_batch": 256,
+        "max_seqs": 2048,          # sequence-slot budget (device mirrors)
         "graph_capture": True,     # hipGraph-capture the decode step
+        "async_decode": True,      # one-step-lag speculative resolution
+        "prefix_sharing": True,    # COW system-prompt KV across agents
+        "dense_quant": "",         # fp8|mxfp4 decode projections (turbo)
+        "expert_fp8": False,       # e4m3 Mixtral expert GEMMs
+        "expert_fp4": False,       # MXFP4 Mixtral expert GEMMs
         "tp_degree": 1,
     },
     "security": {

@@ -90,3 +90,21 @@ def test_deployment_validation(tmp_path):
         "    - name: a\n      model: echo\n      dependencies: [ghost]\n")
     with pytest.raises(DeploymentError, match="ghost"):
         load_deployment(str(p))
+
+
+def test_env_overrides_quant_and_kv_options(tmp_path):
+    """Every engine option added this round is env-overridable (the env
+    layer only applies to keys present in DEFAULTS — regression guard)."""
+    cfg = load_config(path="/nonexistent.yaml", env={
+        "AGENTAINER_ENGINE_KV_DTYPE": "fp8",
+        "AGENTAINER_ENGINE_DENSE_QUANT": "mxfp4",
+        "AGENTAINER_ENGINE_EXPERT_FP4": "true",
+        "AGENTAINER_ENGINE_PREFIX_SHARING": "false",
+        "AGENTAINER_ENGINE_MAX_SEQS": "4096",
+    })
+    e = cfg.get("engine")
+    assert e["kv_dtype"] == "fp8"
+    assert e["dense_quant"] == "mxfp4"
+    assert e["expert_fp4"] is True
+    assert e["prefix_sharing"] is False
+    assert e["max_seqs"] == 4096
