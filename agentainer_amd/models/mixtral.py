@@ -25,7 +25,8 @@ MoE execution model, chosen for the MI355X serving regime:
     activations dynamically per token (ops.quantize_weight_fp8 /
     linear_fp8, v_mfma_f32_16x16x32_fp8_fp8) — half the weight bytes of
     bf16 on the streaming-bound decode path. Enable with
-    engine.expert_fp8: true (or agent dtype fp8).
+    engine.expert_fp8: true (weights are shared per model instance, so
+    the quantization choice is engine-level, not per-agent).
   * MXFP4 expert GEMMs (engine.expert_fp4): e2m1 weights with e8m0
     block-32 scales on the gfx950 block-scaled MFMA
     (v_mfma_scale_f32_16x16x128_f8f6f4, operand maps probed in
