@@ -112,3 +112,26 @@ def test_registry_survives_restart(tmp_path):
     assert g.status == RUNNING  # desired state preserved; reconciler fixes runtime state
     assert g.auto_restart is True
     s2.close()
+
+
+def test_double_start_and_double_stop_are_clean(runtime):
+    """Idempotency-adjacent: re-starting a running agent and re-stopping a
+    stopped one either no-op or raise cleanly — never corrupt state."""
+    a = runtime.agents.deploy(name="dbl", model="echo")
+    runtime.agents.start(a.id)
+    try:
+        runtime.agents.start(a.id)  # second start
+    except Exception:
+        pass
+    assert runtime.agents.get(a.id).status == "running"
+    st, p = runtime.agent_request(a.id, "POST", "/chat", body={"message": "ok"})
+    assert st == 200
+    runtime.agents.stop(a.id)
+    try:
+        runtime.agents.stop(a.id)
+    except Exception:
+        pass
+    assert runtime.agents.get(a.id).status == "stopped"
+    runtime.agents.resume(a.id)
+    st, _ = runtime.agent_request(a.id, "POST", "/chat", body={"message": "ok"})
+    assert st == 200
