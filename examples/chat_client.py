@@ -58,3 +58,17 @@ for _ in range(40):
     time.sleep(0.5)
 
 print("metrics:", call("GET", f"/agents/{aid}/metrics")["data"])
+
+# streaming (SSE): tokens as they decode
+print("streamed: ", end="", flush=True)
+with httpx.stream("POST", f"{BASE}/agent/{aid}/chat",
+                  json={"message": "stream this", "stream": True},
+                  timeout=120) as r:
+    import json as _json
+    for line in r.iter_lines():
+        if line.startswith("data: "):
+            ev = _json.loads(line[len("data: "):])
+            if ev.get("done"):
+                print(f"  [{ev.get('tokens', '?')} tokens]")
+            elif "text" in ev:
+                print(ev["text"], end="", flush=True)
