@@ -241,7 +241,7 @@ class KVCacheManager:
         with self._lock:
             src = self._seqs[src_seq_id]
             dst = self._seqs[seq_id]
-            if dst.length != 0 or dst.pages:
+            if dst.length != 0:
                 raise ValueError(f"adopt_prefix: {seq_id} is not empty")
             if n_tokens % self.page_size != 0 or n_tokens <= 0:
                 raise ValueError("adopt_prefix: n_tokens must be whole pages")
@@ -251,12 +251,19 @@ class KVCacheManager:
             shared = src.pages[:n_pg]
             for p in shared:
                 self._refs[p] = self._refs.get(p, 1) + 1
-            dst.pages = list(shared)
+            # pages already attached at length 0 are admission reservations
+            # (reserve()) sized for the FULL prompt; the shared prefix now
+            # supplies the first n_pg pages' worth, so release exactly that
+            # surplus and keep the rest as the post-prefix tail
+            tail = dst.pages
+            for _ in range(min(n_pg, len(tail))):
+                self._free.append(tail.pop())
+            dst.pages = list(shared) + tail
             dst.length = n_tokens
             if self.is_gpu:
                 slot = self._slot_of[seq_id]
-                self.dev_page_table[slot, :n_pg] = torch.tensor(
-                    shared, dtype=torch.int32, device=self.device)
+                self.dev_page_table[slot, :len(dst.pages)] = torch.tensor(
+                    dst.pages, dtype=torch.int32, device=self.device)
                 self.dev_seq_lens[slot] = n_tokens
 
     def _release_pages(self, pages: List[int]) -> None:
@@ -290,6 +297,29 @@ class KVCacheManager:
             s.length = 0
             if self.is_gpu:
                 self.dev_seq_lens[self._slot_of[seq_id]] = 0
+
+    def reserve(self, seq_id: str, n_tokens: int) -> None:
+        """Pre-allocate pages covering n_tokens future appends (chunked
+        prefill admission: room for the whole prompt+generation is claimed
+        up front, so other agents admitted between chunks cannot starve a
+        half-prefilled sequence into OutOfPages mid-plan). Raises
+        OutOfPages atomically — no pages are taken on failure."""
+        with self._lock:
+            s = self._seqs[seq_id]
+            need_pages = -(-(s.length + n_tokens) // self.page_size)
+            if need_pages > self.max_pages_per_seq:
+                raise OutOfPages("sequence would exceed max_pages_per_seq")
+            short = need_pages - len(s.pages)
+            if short <= 0:
+                return
+            if short > len(self._free):
+                raise OutOfPages(f"KV pool exhausted ({self.n_pages} pages)")
+            for _ in range(short):
+                page = self._free.pop()
+                s.pages.append(page)
+                if self.is_gpu:
+                    self.dev_page_table[self._slot_of[seq_id],
+                                        len(s.pages) - 1] = page
 
     def can_append(self, seq_id: str, n_tokens: int) -> bool:
         with self._lock:

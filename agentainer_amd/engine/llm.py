@@ -80,6 +80,9 @@ class AgentBinding:
     paused: bool = False
     requests: int = 0
     tokens: int = 0
+    # /clear: the next admitted request prefills onto a reset KV sequence
+    # (plan-mirrored to TP workers via needs_reset)
+    pending_reset: bool = False
     queue: "queue.Queue[GenRequest]" = field(default_factory=queue.Queue)
     active: Optional[GenRequest] = None
     # whole-page token prefix shared with other agents (system prompt),
@@ -223,6 +226,11 @@ class ModelInstance:
         # already invalidated (their outputs are discarded).
         self._prefill_stream = (torch.cuda.Stream()
                                 if self.is_gpu else None)
+        # set whenever pages were freed (reset/rollback) while a decode
+        # step may still be in flight: the next side-stream prefill must
+        # order itself after that decode so a speculative K/V write can't
+        # land in a page the prefill just re-allocated
+        self._pages_freed = False
         self._pad_slot = -1
         if self.is_gpu:
             # reserve one sequence slot as the graph's pad row target
@@ -310,7 +318,10 @@ class ModelInstance:
             return ran
         admitted = self._admit()
         if admitted:
-            self._prefill(admitted)
+            try:
+                self._prefill(admitted)
+            except OutOfPages as e:
+                self._fail_prefill(admitted, str(e))
         ran_decode = False
         with self._lock:
             batch = [r for r in self.running if not r.done.is_set()]
@@ -340,6 +351,12 @@ class ModelInstance:
             seq_ids = [self._bindings[r.agent_id].seq_id for r in batch]
             rows = kvm.decode_batch_prepare(seq_ids)
             entry = self._get_graph(bucket)
+            # the PREVIOUS step's non-blocking H2D copies read these pinned
+            # staging buffers; wait for that DMA before rewriting them (a
+            # torn read would feed wrong slots/tokens into the graph)
+            ev = entry.get("h2d_ev")
+            if ev is not None:
+                ev.synchronize()
             entry["rows_pin"][:B] = torch.tensor(rows, dtype=torch.long)
             entry["rows_pin"][B:] = self._pad_slot
             entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
@@ -350,6 +367,9 @@ class ModelInstance:
                 [r.generated[-1] if idx[i] < 0 else 0
                  for i, r in enumerate(batch)], dtype=torch.long)
             entry["ids"][:B].copy_(entry["ids_pin"][:B], non_blocking=True)
+            if ev is None:
+                ev = entry["h2d_ev"] = torch.cuda.Event()
+            ev.record()
             if prev is not None and any(x >= 0 for x in idx):
                 gidx = torch.tensor([max(x, 0) for x in idx], dtype=torch.long,
                                     device=dev)
@@ -378,11 +398,50 @@ class ModelInstance:
         admitted = self._admit()
         if admitted:
             if self._prefill_stream is not None:
-                with torch.cuda.stream(self._prefill_stream):
-                    self._prefill(admitted)
+                if self._pages_freed:
+                    # pages freed by a reset/rollback may be re-allocated by
+                    # this prefill while the in-flight decode still writes
+                    # its (rolled-back) speculative K/V there — serialize
+                    # the streams for this one step
+                    self._prefill_stream.wait_stream(torch.cuda.current_stream())
+                    self._pages_freed = False
+                try:
+                    with torch.cuda.stream(self._prefill_stream):
+                        self._prefill(admitted)
+                except OutOfPages as e:
+                    self._fail_prefill(admitted, str(e))
             else:
-                self._prefill(admitted)
+                try:
+                    self._prefill(admitted)
+                except OutOfPages as e:
+                    self._fail_prefill(admitted, str(e))
         return bool(batch) or bool(admitted) or prev is not None
+
+    def _fail_prefill(self, reqs: List[GenRequest], msg: str) -> None:
+        """Admission reserves full KV room, so a prefill-time OutOfPages
+        should be impossible — but if one ever escapes, fail the batch's
+        requests EXPLICITLY (clients see the error instead of hanging to
+        timeout) and reset their sequences to a consistent state."""
+        with self._lock:
+            for r in reqs:
+                if r in self._chunking:
+                    self._chunking.remove(r)
+                if r in self._running_set:
+                    self.running.remove(r)
+                    self._running_set.discard(r)
+                if not r.done.is_set():
+                    r.error = f"KV pool exhausted: {msg}"
+                    if r.stream_q is not None:
+                        r.stream_q.put(None)
+                    r.done.set()
+                b = self._bindings.get(r.agent_id)
+                if b is not None:
+                    if b.active is r:
+                        b.active = None
+                    if self.kvm.has_seq(b.seq_id):
+                        self.kvm.reset_seq(b.seq_id)
+                        self._pages_freed = True
+                    self._pump_agent(b)
 
     def _resolve_spec(self, prev, launched) -> None:
         if prev is None:
@@ -408,6 +467,8 @@ class ModelInstance:
                         b = self._bindings.get(r.agent_id)
                         if b is not None:
                             rollback_seqs.append(b.seq_id)
+        if rollback_seqs:
+            self._pages_freed = True  # see _step_async phase 3
         self.kvm.rollback_many(rollback_seqs)
 
     def drain_async(self) -> None:
@@ -455,21 +516,31 @@ class ModelInstance:
             # prompts no longer stall the whole batch)
             req.slice_len = min(need, budget)
             # KV room: prompt + generation. Context truncation fires on
-            # pool pressure AND on the model's rope/position horizon —
-            # a conversation crossing max_position would otherwise step
-            # the rope table out of bounds
+            # pool pressure, on the model's rope/position horizon (a
+            # conversation crossing max_position would step the rope
+            # table out of bounds), and on an explicit /clear
             req.needs_reset = (
-                self.kvm.seq_len(b.seq_id) + need + req.max_new
+                b.pending_reset
+                or self.kvm.seq_len(b.seq_id) + need + req.max_new
                 > self.cfg.max_position
                 or not self.kvm.can_append(b.seq_id, need + req.max_new))
             if req.needs_reset:
-                # truncate conversation: reset KV (applied in the prefill
-                # plan so TP workers mirror it), re-prefill just this turn
-                if not self.kvm.can_append_after_reset(b.seq_id,
-                                                       need + req.max_new):
-                    req.error = "KV pool exhausted"
-                    req.done.set()
-                    continue
+                # truncate conversation NOW (rank 0); the prefill plan
+                # carries the flag so TP workers mirror the reset
+                self.kvm.reset_seq(b.seq_id)
+                self._pages_freed = True  # in-flight-step hazard, see _step_async
+            try:
+                # claim pages for the whole prompt + generation (+1 for a
+                # speculative async append) up front: admission is the ONLY
+                # place the KV pool can reject work, so a half-prefilled
+                # chunked prompt can never die OutOfPages mid-plan and the
+                # decode path never allocates under pressure
+                self.kvm.reserve(b.seq_id, need + req.max_new + 1)
+            except OutOfPages:
+                req.error = "KV pool exhausted"
+                req.done.set()
+                continue
+            b.pending_reset = False
             out.append(req)
             budget -= req.slice_len
         return out
@@ -574,7 +645,9 @@ class ModelInstance:
         for seq_id, tokens, needs_reset, adopt in plan:
             if not self.kvm.has_seq(seq_id):
                 self.kvm.create_seq(seq_id)  # lazily-created shared-prefix seq
-            if needs_reset:
+            if needs_reset and self.tp_rank != 0:
+                # rank 0 already reset at admission (before reserving KV
+                # room); workers mirror the truncation via the plan flag
                 self.kvm.reset_seq(seq_id)
             if adopt is not None and self.kvm.seq_len(seq_id) == 0:
                 self.kvm.adopt_prefix(seq_id, adopt[0], adopt[1])
@@ -763,14 +836,28 @@ class ModelInstance:
         elif not self.kvm.has_seq(seq_id):
             self.kvm.create_seq(seq_id)
 
+    def _acquire_step_mutex_unless_stuck(self) -> bool:
+        """Blocking acquire that gives up when the engine thread is wedged
+        mid-step (watchdog fired): a stuck HIP stream holds _step_mutex
+        forever, and the health monitor's restart path must not deadlock
+        behind it (it IS the recovery path for exactly that state)."""
+        while True:
+            if self._step_mutex.acquire(timeout=2.0):
+                return True
+            if self.stuck():
+                return False
+
     def bind(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint],
              worker_has_ckpt: bool = False):
+        locked = False
         if not self.sync_mode:
-            self._step_mutex.acquire()
+            locked = self._acquire_step_mutex_unless_stuck()
+            if not locked:
+                raise EngineDead(f"engine {self.name} is stuck (watchdog)")
         try:
             self._bind_locked(agent, seq_id, ckpt, worker_has_ckpt)
         finally:
-            if not self.sync_mode:
+            if locked:
                 self._step_mutex.release()
 
     def _bind_locked(self, agent, seq_id: str, ckpt: Optional[KVCheckpoint],
@@ -792,16 +879,23 @@ class ModelInstance:
             self.refcount += 1
 
     def unbind(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:
+        locked = True
         if not self.sync_mode:
-            self._step_mutex.acquire()
+            locked = self._acquire_step_mutex_unless_stuck()
         try:
-            return self._unbind_locked(agent_id, offload)
+            # stuck engine: force-detach — skip the drain/offload (any GPU
+            # work would hang on the wedged stream) so the health monitor's
+            # stop->start restart completes; the agent resumes with a fresh
+            # KV sequence and its pending WAL replays (at-least-once)
+            return self._unbind_locked(agent_id, offload and locked,
+                                       force=not locked)
         finally:
-            if not self.sync_mode:
+            if locked and not self.sync_mode:
                 self._step_mutex.release()
 
-    def _unbind_locked(self, agent_id: str, offload: bool) -> Optional[KVCheckpoint]:
-        if self.async_decode:
+    def _unbind_locked(self, agent_id: str, offload: bool,
+                       force: bool = False) -> Optional[KVCheckpoint]:
+        if self.async_decode and not force:
             self.drain_async()
         with self._lock:
             b = self._bindings.pop(agent_id, None)
@@ -961,14 +1055,17 @@ class LLMEngine:
             # workers' shards live in their own processes); stop/resume
             # checkpoints per-rank correctly — skip live KV here
             return None
+        locked = True
         if not inst.sync_mode:
-            inst._step_mutex.acquire()
+            locked = inst._acquire_step_mutex_unless_stuck()
+            if not locked:
+                return None  # wedged engine: no live snapshot possible
         try:
             if inst.async_decode:
                 inst.drain_async()
             return inst.kvm.offload(b.seq_id, free=False)
         finally:
-            if not inst.sync_mode:
+            if locked and not inst.sync_mode:
                 inst._step_mutex.release()
 
     def import_kv(self, agent_id: str, ckpt: KVCheckpoint) -> None:
@@ -1009,6 +1106,15 @@ class LLMEngine:
         eb = getattr(self, "_echo_attached", {}).get(agent_id)
         if eb is not None:
             eb["paused"] = False
+
+    def reset_conversation(self, agent_id: str) -> None:
+        """/clear hook: schedule a KV reset so the next chat starts from an
+        empty sequence (clearing only the history list would leave the next
+        turn's system prompt prefilled on top of the stale multi-turn KV —
+        the agent would keep attending to the cleared conversation)."""
+        b = self._binding(agent_id)
+        if b is not None:
+            b.pending_reset = True
 
     def _binding(self, agent_id: str) -> Optional[AgentBinding]:
         model = self._agent_model.get(agent_id)
@@ -1224,7 +1330,9 @@ class LLMEngine:
             agents[aid] = {"requests": st["requests"], "tokens": st["tokens"],
                            "paused": st["paused"]}
         models = {}
-        for name, inst in self._instances.items():
+        with self._lock:  # _get_instance can insert concurrently
+            instances = list(self._instances.items())
+        for name, inst in instances:
             with inst._lock:
                 for aid, b in inst._bindings.items():
                     agents[aid] = {

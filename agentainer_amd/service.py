@@ -273,6 +273,12 @@ class Runtime:
             return {"history": self.store.lrange(f"agent:{agent.id}:conversations")}
         if path == "/clear":
             self.store.delete(f"agent:{agent.id}:conversations")
+            # also reset the engine-side KV sequence: a cleared history
+            # with resident KV would leave the next turn attending to the
+            # old conversation (and double the system prompt)
+            reset = getattr(self.engine, "reset_conversation", None)
+            if reset is not None:
+                reset(agent.id)
             return {"status": "cleared"}
         if path == "/metrics":
             return {"metrics": self.store.hgetall(f"agent:{agent.id}:metrics")}
