@@ -1385,6 +1385,13 @@ class LLMEngine:
             op = cmd[0]
             if op == "shutdown":
                 break
+            if op == "barrier":
+                # timing fence (bench.py --tp): drain this rank's stream,
+                # then rendezvous so rank 0's clock bounds every rank
+                if self.device.startswith("cuda"):
+                    torch.cuda.synchronize()
+                par.barrier()
+                continue
             if op == "instance":
                 with self._lock:
                     if cmd[1] not in self._instances:

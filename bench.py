@@ -21,6 +21,13 @@ Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Tensor-parallel mode (BASELINE config 4, Llama-3-70B TP=8 over xGMI):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+      --master-addr 127.0.0.1 bench.py --gpus 8 --tp 8 --steps K --warmup W
+Rank 0 schedules ONE engine sharded across all ranks (broadcast step
+plans, RCCL all-reduce per sub-layer); value = rank 0's completed
+requests / barrier-bounded elapsed.
 """
 
 from __future__ import annotations
@@ -48,8 +55,12 @@ def parse_args():
     p.add_argument("--steps", type=int, default=300)
     p.add_argument("--warmup", type=int, default=60)
     p.add_argument("--agents", type=int, default=64)
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree (= world size); default "
+                        "model becomes llama3-70b (config 4)")
     p.add_argument("--model", default=None,
-                   help="default: llama3-8b on GPU, tiny-llama on CPU")
+                   help="default: llama3-8b on GPU, tiny-llama on CPU "
+                        "(llama3-70b / tiny-llama-tp when --tp > 1)")
     p.add_argument("--prompt-len", type=int, default=256)
     p.add_argument("--gen-len", type=int, default=64)
     p.add_argument("--max-batch-tokens", type=int, default=16384)
@@ -110,8 +121,13 @@ class ClosedLoopClient:
                 self.store.ltrim(hist, -50, -1)
             if record:
                 self.completed.append((req.enq_t, req.fin_t))
-            # KV hygiene: reset the conversation so seq length stays bounded
-            self.inst.kvm.reset_seq(self.agent.id)
+            # KV hygiene: schedule a conversation reset so seq length stays
+            # bounded. pending_reset is applied AT ADMISSION inside the
+            # engine (plan-mirrored), so TP workers reset in lockstep and
+            # no pages are freed under an in-flight speculative step.
+            b = self.inst.binding(self.agent.id)
+            if b is not None:
+                b.pending_reset = True
             self.submit()
 
 
@@ -130,7 +146,14 @@ def main():
         dist = dist_mod
         dist.init_process_group(backend="nccl" if has_gpu else "gloo")
 
-    model_name = args.model or ("llama3-8b" if has_gpu else "tiny-llama")
+    tp = max(args.tp, 1)
+    if tp > 1 and world != tp:
+        raise SystemExit(f"--tp {tp} requires a torchrun world of {tp} "
+                         f"(got WORLD_SIZE={world})")
+    if tp > 1:
+        model_name = args.model or ("llama3-70b" if has_gpu else "tiny-llama-tp")
+    else:
+        model_name = args.model or ("llama3-8b" if has_gpu else "tiny-llama")
     if not has_gpu and args.model is None:
         # CPU fallback sizes so the no-GPU smoke run finishes in seconds
         args.agents = min(args.agents, 4)
@@ -138,6 +161,10 @@ def main():
         args.gen_len = min(args.gen_len, 8)
         args.steps = min(args.steps, 40)
         args.warmup = min(args.warmup, 8)
+    # any driver-chosen --steps must complete requests INSIDE the timed
+    # region (a gen_len longer than the window yields a vacuous 0 req/s):
+    # cap generation length at half the timed steps
+    args.gen_len = max(1, min(args.gen_len, max(args.steps // 2, 1)))
 
     tmp = tempfile.mkdtemp(prefix=f"bench-rank{rank}-")
     cfg = load_config(path="/nonexistent.yaml", env={})
@@ -150,10 +177,28 @@ def main():
     cfg.data["engine"]["dense_quant"] = args.dense_quant
     if args.kv_fp8:
         cfg.data["engine"]["kv_dtype"] = "fp8"
+    if tp > 1:
+        cfg.data["engine"]["tp_degree"] = tp
     store = Store(os.path.join(tmp, "state"), sync="interval")
     engine = LLMEngine(store, cfg, device=device, state_root=tmp)
+    if tp > 1 and rank != 0:
+        # SPMD worker: execute rank 0's broadcast step plans (incl. the
+        # "barrier" timing fences) until shutdown
+        engine.run_worker()
+        dist.destroy_process_group()
+        return
     manager = Manager(store, engine, cfg)
     wal = RequestManager(store)
+
+    def fence():
+        """Synchronize + barrier on every rank (TP: via the plan channel)."""
+        if tp > 1:
+            import agentainer_amd.parallel as par
+            par.broadcast_obj(("barrier",))
+        if has_gpu:
+            torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
 
     t_load0 = time.time()
     agents = []
@@ -185,16 +230,10 @@ def main():
         c.completed.clear()
 
     # timed region
-    if dist:
-        dist.barrier()
-    if has_gpu:
-        torch.cuda.synchronize()
+    fence()
     t0 = time.time()
     run_steps(args.steps, record=True)
-    if has_gpu:
-        torch.cuda.synchronize()
-    if dist:
-        dist.barrier()
+    fence()
     t1 = time.time()
 
     elapsed = t1 - t0
@@ -203,14 +242,30 @@ def main():
     decode_tokens = inst.decode_tokens
     prefill_tokens = inst.prefill_tokens
 
-    if dist:
+    if dist and tp == 1:
+        # weak DP: whole-job aggregate = sum of completions / MAX elapsed,
+        # with the e2e populations of EVERY rank merged so p50/p99 are
+        # whole-job too (not rank-0-only)
+        dd = device if has_gpu else "cpu"
         stats = torch.tensor([elapsed, float(n_done)],
-                             dtype=torch.float64,
-                             device=device if has_gpu else "cpu")
+                             dtype=torch.float64, device=dd)
         gathered = [torch.zeros_like(stats) for _ in range(world)]
         dist.all_gather(gathered, stats)
         elapsed = max(float(g[0]) for g in gathered)  # MAX over ranks
         n_done = sum(float(g[1]) for g in gathered)
+        n = torch.tensor([len(e2es)], dtype=torch.int64, device=dd)
+        ns = [torch.zeros_like(n) for _ in range(world)]
+        dist.all_gather(ns, n)
+        width = max(int(x.item()) for x in ns)
+        if width > 0:
+            buf = torch.full((width,), float("nan"), dtype=torch.float64,
+                             device=dd)
+            if e2es:
+                buf[: len(e2es)] = torch.tensor(e2es, dtype=torch.float64,
+                                                device=dd)
+            bufs = [torch.zeros_like(buf) for _ in range(world)]
+            dist.all_gather(bufs, buf)
+            e2es = sorted(v for b in bufs for v in b.tolist() if v == v)
 
     if rank == 0:
         p50 = statistics.median(e2es) if e2es else None
@@ -224,7 +279,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000, 3),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if tp > 1 else "weak",
             "vs_baseline": None,  # reference publishes no number
             "dtype": (f"bf16+{args.dense_quant}w" if args.dense_quant else
                       "bf16+fp4exp" if args.expert_fp4 else
@@ -233,12 +288,13 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": model_name,
-                "agents_per_gpu": args.agents,
-                "global_batch": args.agents * max(world, 1),
+                "agents_per_gpu": args.agents if tp == 1 else None,
+                "agents": args.agents * (max(world, 1) if tp == 1 else 1),
+                "global_batch": args.agents * (max(world, 1) if tp == 1 else 1),
                 "prompt_len": args.prompt_len,
                 "gen_len": args.gen_len,
                 "seq_len": args.prompt_len + args.gen_len,
-                "parallelism": f"dp{max(world,1)}",
+                "parallelism": (f"tp{world}" if tp > 1 else f"dp{max(world,1)}"),
                 "p50_e2e_s": round(p50, 4) if p50 is not None else None,
                 "p99_e2e_s": round(p99, 4) if p99 is not None else None,
                 "decode_tokens_rank0": decode_tokens,
@@ -246,6 +302,8 @@ def main():
             },
         }
         print(json.dumps(out))
+    if tp > 1:
+        engine.shutdown()  # broadcasts shutdown so workers exit run_worker
     if dist:
         dist.destroy_process_group()
 
