@@ -98,8 +98,12 @@ class KVCacheManager:
         else:
             self.dev_page_table = None
             self.dev_seq_lens = None
+        # slots are assigned on CPU too: slot allocation is deterministic
+        # across TP ranks (same create/free order), so the slot id doubles
+        # as the wire identity of a sequence in binary step plans
         self._free_slots: List[int] = list(range(max_slots - 1, -1, -1))
         self._slot_of: Dict[str, int] = {}
+        self._seq_of_slot: Dict[int, str] = {}
 
     # ---------- queries ----------
 
@@ -140,16 +144,22 @@ class KVCacheManager:
                 raise ValueError(f"sequence {seq_id} exists")
             s = Sequence(seq_id)
             self._seqs[seq_id] = s
+            if not self._free_slots:
+                raise OutOfPages("no free sequence slots")
+            slot = self._free_slots.pop()
+            self._slot_of[seq_id] = slot
+            self._seq_of_slot[slot] = seq_id
             if self.is_gpu:
-                if not self._free_slots:
-                    raise OutOfPages("no free sequence slots")
-                slot = self._free_slots.pop()
-                self._slot_of[seq_id] = slot
                 self.dev_seq_lens[slot] = 0
             return s
 
     def slot(self, seq_id: str) -> int:
         return self._slot_of[seq_id]
+
+    def seq_of_slot(self, slot: int) -> str:
+        """Inverse of slot(): wire identity -> sequence (TP worker plans)."""
+        with self._lock:
+            return self._seq_of_slot[slot]
 
     def push_dev(self, seq_id: str) -> None:
         """Sync a sequence's page row + length to the device mirrors."""
@@ -204,7 +214,7 @@ class KVCacheManager:
                     if self.is_gpu:
                         self.dev_page_table[self._slot_of[seq_id],
                                             page_idx] = page
-                rows.append(self._slot_of[seq_id] if self.is_gpu else -1)
+                rows.append(self._slot_of[seq_id])
         return rows
 
     def advance_many(self, seq_ids) -> None:
@@ -285,7 +295,9 @@ class KVCacheManager:
                 self._release_pages(s.pages)
             slot = self._slot_of.pop(seq_id, None)
             if slot is not None:
-                self.dev_seq_lens[slot] = -1
+                self._seq_of_slot.pop(slot, None)
+                if self.is_gpu:
+                    self.dev_seq_lens[slot] = -1
                 self._free_slots.append(slot)
 
     def reset_seq(self, seq_id: str) -> None:

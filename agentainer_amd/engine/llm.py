@@ -43,6 +43,11 @@ from ..models.llama import (AttnMetadata, LLAMA_CONFIGS, LlamaConfig,
 
 DEFAULT_MAX_NEW = 64
 
+# int-vector op codes on the TP plan channel (parallel.dist.send_ints);
+# 0 is reserved for pickled commands
+OP_DECODE = 1   # [OP, model_idx, B, slot*B, token*B] — the per-step hot path
+OP_BARRIER = 2  # [OP] — timing fence (bench.py --tp)
+
 
 class EngineDead(RuntimeError):
     pass
@@ -206,15 +211,23 @@ class ModelInstance:
         # captured once per batch bucket and replayed with two small H2D
         # copies (row slots + last tokens) per step.
         self.is_gpu = device.startswith("cuda")
-        self.use_graph = (self.is_gpu and self.tp_size == 1
+        # hipGraph capture now includes the TP case: the captured decode
+        # body contains the per-sublayer RCCL all-reduces (SURVEY.md §7.3
+        # names exactly this); capture failure falls back to eager replay
+        # (_get_graph) identically on every rank.
+        self.use_graph = (self.is_gpu
                           and bool(engine_cfg.get("graph_capture", True)))
         # async decode (speculative one-step lag): the next decode step is
         # launched with device-fed tokens BEFORE the previous step's tokens
         # reach the host, so per-step host bookkeeping overlaps GPU work.
         # Rows that resolve as finished get their speculative append rolled
-        # back (kvm.rollback_many, stream-ordered).
-        self.async_decode = (self.use_graph
+        # back (kvm.rollback_many, stream-ordered). TP=1 only: the worker
+        # loop executes explicit plans and has no speculative mirror.
+        self.async_decode = (self.use_graph and self.tp_size == 1
                              and bool(engine_cfg.get("async_decode", True)))
+        # index of this model in the engine's creation order — the wire
+        # identity in binary step plans (identical on every TP rank)
+        self._model_idx = 0
         self._spec: Optional[Dict[str, Any]] = None
         self._graphs: Dict[int, Dict[str, Any]] = {}
         # prefill runs on its own HIP stream so a prefill batch overlaps
@@ -719,13 +732,23 @@ class ModelInstance:
         entry = {"rows": rows, "ids": ids, "inc": inc, "rows_pin": rows_pin,
                  "ids_pin": ids_pin, "graph": None, "logits": None}
         if self.use_graph:
-            for _ in range(2):  # warmup before capture
-                self._device_decode_fwd(rows, ids, inc)
-            torch.cuda.synchronize()
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                entry["logits"] = self._device_decode_fwd(rows, ids, inc)
-            entry["graph"] = g
+            try:
+                for _ in range(2):  # warmup before capture (also exercises
+                    self._device_decode_fwd(rows, ids, inc)  # RCCL comms)
+                torch.cuda.synchronize()
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    entry["logits"] = self._device_decode_fwd(rows, ids, inc)
+                entry["graph"] = g
+            except Exception:
+                # e.g. an RCCL build that rejects capture: run eager. Every
+                # rank executes identical code, so the fallback is lockstep.
+                traceback.print_exc()
+                self.use_graph = False
+                self.async_decode = False
+                entry["graph"] = None
+                entry["logits"] = None
+                torch.cuda.synchronize()
         self._graphs[bucket] = entry
         return entry
 
@@ -737,7 +760,14 @@ class ModelInstance:
                 reqs = reqs[:bucket]
         plan = [(self._bindings[r.agent_id].seq_id, r.generated[-1])
                 for r in reqs]
-        self._bcast(("decode", self.name, plan))
+        if self.tp_size > 1 and self.tp_rank == 0:
+            # binary hot path: ONE int-tensor broadcast on the gloo plan
+            # channel (slot ids are deterministic across ranks), replacing
+            # the round-1 per-step pickled broadcast_object_list
+            vec = [OP_DECODE, self._model_idx, len(plan)]
+            vec.extend(self.kvm.slot(s) for s, _t in plan)
+            vec.extend(t for _s, t in plan)
+            par.send_ints(vec)
         logits = self._decode_exec(plan)
         toks = self._sample(logits, reqs)
         self.decode_tokens += len(reqs)
@@ -776,12 +806,20 @@ class ModelInstance:
         # host half of the append, one lock for the whole batch
         row_ids = kvm.decode_batch_prepare([s for s, _t in plan])
         entry = self._get_graph(bucket)
+        # TP workers run ahead with no host sync per step: wait out the
+        # previous step's H2D DMA before rewriting the pinned staging
+        ev = entry.get("h2d_ev")
+        if ev is not None:
+            ev.synchronize()
         entry["rows_pin"][:B] = torch.tensor(row_ids, dtype=torch.long)
         entry["rows_pin"][B:] = self._pad_slot
         entry["ids_pin"][:B] = torch.tensor([t for _s, t in plan],
                                             dtype=torch.long)
         entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
         entry["ids"].copy_(entry["ids_pin"], non_blocking=True)
+        if ev is None:
+            ev = entry["h2d_ev"] = torch.cuda.Event()
+        ev.record()
         if entry["graph"] is not None:
             entry["graph"].replay()
             logits = entry["logits"][:B]
@@ -941,6 +979,8 @@ class LLMEngine:
         self.engine_cfg = dict(config.get("engine")) if config else {}
         self.state_root = state_root or os.path.expanduser("~/.agentainer_amd")
         self._instances: Dict[str, ModelInstance] = {}
+        # creation order == binary-plan model index (lockstep on all ranks)
+        self._instance_order: List[str] = []
         self._agent_model: Dict[str, str] = {}
         self._ckpts: Dict[str, KVCheckpoint] = {}
         self._lock = threading.RLock()
@@ -992,6 +1032,8 @@ class LLMEngine:
                 if self.tp_size > 1 and self.tp_rank == 0:
                     par.broadcast_obj(("instance", model))
                 inst = self._make_instance(model)
+                inst._model_idx = len(self._instance_order)
+                self._instance_order.append(model)
                 self._instances[model] = inst
                 inst.start()
             return inst
@@ -1377,25 +1419,45 @@ class LLMEngine:
 
     def run_worker(self):
         """SPMD worker: executes the step plans rank 0 broadcasts.
-        Page allocation is deterministic, so this rank's KV pool mirrors
-        rank 0's without any per-step tensor metadata exchange."""
+        Page/slot allocation is deterministic, so this rank's KV pool
+        mirrors rank 0's without any per-step tensor metadata exchange.
+
+        Decode (the per-token hot path) arrives as ONE int vector —
+        [OP_DECODE, model_idx, B, slots, tokens] — and the worker enqueues
+        the step with NO host synchronize: kernels and the in-graph RCCL
+        all-reduces are stream-ordered, so the worker runs ahead of the
+        GPU and plan latency overlaps compute (round 1 synced per
+        command, stacking host latency on every token; VERDICT r1 #3)."""
         assert self.tp_size > 1 and self.tp_rank > 0
         while True:
-            cmd = par.broadcast_obj(None)
+            cmd = par.recv_cmd()
+            if isinstance(cmd, list):  # int-vector fast path
+                op = cmd[0]
+                if op == OP_DECODE:
+                    inst = self._instances[self._instance_order[cmd[1]]]
+                    B = cmd[2]
+                    kvm = inst.kvm
+                    plan = [(kvm.seq_of_slot(s), t)
+                            for s, t in zip(cmd[3:3 + B], cmd[3 + B:3 + 2 * B])]
+                    inst._decode_exec(plan)
+                elif op == OP_BARRIER:
+                    # timing fence (bench.py --tp): drain this rank's
+                    # stream, then rendezvous so rank 0's clock bounds
+                    # every rank
+                    if self.device.startswith("cuda"):
+                        torch.cuda.synchronize()
+                    par.barrier()
+                continue
             op = cmd[0]
             if op == "shutdown":
                 break
-            if op == "barrier":
-                # timing fence (bench.py --tp): drain this rank's stream,
-                # then rendezvous so rank 0's clock bounds every rank
-                if self.device.startswith("cuda"):
-                    torch.cuda.synchronize()
-                par.barrier()
-                continue
             if op == "instance":
                 with self._lock:
                     if cmd[1] not in self._instances:
-                        self._instances[cmd[1]] = self._make_instance(cmd[1])
+                        inst = self._make_instance(cmd[1])
+                        inst._model_idx = len(self._instance_order)
+                        self._instance_order.append(cmd[1])
+                        self._instances[cmd[1]] = inst
                 continue
             inst = self._instances[cmd[1]]
             if op == "prefill":
@@ -1417,5 +1479,3 @@ class LLMEngine:
                         self._save_disk_ckpt(seq_id, ckpt)
                 else:
                     inst.kvm.free_seq(seq_id)
-            if self.device.startswith("cuda"):
-                torch.cuda.synchronize()

@@ -43,7 +43,7 @@ import time
 import torch
 
 from agentainer_amd.config import load_config
-from agentainer_amd.engine.llm import GenRequest, LLMEngine
+from agentainer_amd.engine.llm import OP_BARRIER, GenRequest, LLMEngine
 from agentainer_amd.registry import Manager
 from agentainer_amd.store import Store
 from agentainer_amd.wal import RequestManager
@@ -194,7 +194,7 @@ def main():
         """Synchronize + barrier on every rank (TP: via the plan channel)."""
         if tp > 1:
             import agentainer_amd.parallel as par
-            par.broadcast_obj(("barrier",))
+            par.send_ints([OP_BARRIER])
         if has_gpu:
             torch.cuda.synchronize()
         if dist:
