@@ -295,29 +295,47 @@ class LlamaForCausalLM(torch.nn.Module):
 
     @torch.no_grad()
     def _load_hf_state_dict(self, sd: Dict[str, torch.Tensor]):
+        """Copy full HF tensors into this rank's (possibly TP-sharded)
+        parameters: column-parallel q/k/v/gate/up slice output rows by
+        rank, row-parallel o/down slice input columns; embeddings, norms
+        and lm_head are replicated. TP=1 reduces to the identity slices,
+        so weights-path deploys shard inside the loader for any degree
+        (the round-1 loader only accepted full shapes)."""
         cfg = self.cfg
+        r, s = self.tp_rank, self.tp_size
+        qh = cfg.n_heads // s
+        kvh = max(cfg.n_kv_heads // s, 1)
+        hd = cfg.head_dim
+        inter = cfg.intermediate_size // s
+
         def t(name):
             return sd[name].to(torch.bfloat16)
+
+        def rows(x, n):  # column-parallel: rank's slice of output rows
+            return x[r * n:(r + 1) * n]
+
         self.embed.copy_(t("model.embed_tokens.weight"))
         if not cfg.tie_embeddings and "lm_head.weight" in sd:
             self.lm_head.copy_(t("lm_head.weight"))
         self.final_ln.copy_(t("model.norm.weight"))
         for i, layer in enumerate(self.layers):
             pfx = f"model.layers.{i}."
-            q = t(pfx + "self_attn.q_proj.weight")
-            k = t(pfx + "self_attn.k_proj.weight")
-            v = t(pfx + "self_attn.v_proj.weight")
+            q = rows(t(pfx + "self_attn.q_proj.weight"), qh * hd)
+            k = rows(t(pfx + "self_attn.k_proj.weight"), kvh * hd)
+            v = rows(t(pfx + "self_attn.v_proj.weight"), kvh * hd)
             layer.attn.qkv_proj.copy_(torch.cat([q, k, v], dim=0))
             if layer.attn.qkv_bias is not None:
                 layer.attn.qkv_bias.copy_(torch.cat(
-                    [t(pfx + "self_attn.q_proj.bias"),
-                     t(pfx + "self_attn.k_proj.bias"),
-                     t(pfx + "self_attn.v_proj.bias")], dim=0))
-            layer.attn.o_proj.copy_(t(pfx + "self_attn.o_proj.weight"))
-            g = t(pfx + "mlp.gate_proj.weight")
-            u = t(pfx + "mlp.up_proj.weight")
+                    [rows(t(pfx + "self_attn.q_proj.bias"), qh * hd),
+                     rows(t(pfx + "self_attn.k_proj.bias"), kvh * hd),
+                     rows(t(pfx + "self_attn.v_proj.bias"), kvh * hd)], dim=0))
+            layer.attn.o_proj.copy_(
+                t(pfx + "self_attn.o_proj.weight")[:, r * qh * hd:(r + 1) * qh * hd])
+            g = rows(t(pfx + "mlp.gate_proj.weight"), inter)
+            u = rows(t(pfx + "mlp.up_proj.weight"), inter)
             layer.mlp.gate_up.copy_(torch.cat([g, u], dim=0))
-            layer.mlp.down.copy_(t(pfx + "mlp.down_proj.weight"))
+            layer.mlp.down.copy_(
+                t(pfx + "mlp.down_proj.weight")[:, r * inter:(r + 1) * inter])
             layer.input_ln.copy_(t(pfx + "input_layernorm.weight"))
             layer.post_ln.copy_(t(pfx + "post_attention_layernorm.weight"))
 
