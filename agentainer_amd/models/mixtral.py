@@ -84,14 +84,23 @@ class MixtralMoE(torch.nn.Module):
         self.gate_up_fp4: List[Optional[tuple]] = [None] * self.n_local
         self.down_fp4: List[Optional[tuple]] = [None] * self.n_local
 
-    # prefill batches at least this big use token-shuffle all-to-all EP
+    # batches at least this big use token-shuffle all-to-all EP by default
     A2A_MIN_TOKENS = 128
+    # ep_mode: "auto" = token-count threshold (a2a_min_tokens), "a2a" =
+    # always token-shuffle (config 5's decode all-to-all, forceable for
+    # crossover measurement), "dense" = always dense-routed + all-reduce.
+    # Set per engine via engine.moe_ep_mode / engine.moe_a2a_min_tokens
+    # (make_mixtral_instance threads them onto every layer's MoE).
+    ep_mode = "auto"
+    a2a_min_tokens = A2A_MIN_TOKENS
 
     def forward(self, h: torch.Tensor, ep_group=None) -> torch.Tensor:
         T = h.size(0)
-        if (ep_group is not None and T >= self.A2A_MIN_TOKENS
-                and torch.distributed.get_world_size(ep_group) > 1):
-            return self.forward_a2a(h, ep_group)
+        if ep_group is not None and torch.distributed.get_world_size(ep_group) > 1:
+            use_a2a = (self.ep_mode == "a2a"
+                       or (self.ep_mode == "auto" and T >= self.a2a_min_tokens))
+            if use_a2a:
+                return self.forward_a2a(h, ep_group)
         logits = F.linear(h, self.router).float()           # [T, E]
         probs = torch.softmax(logits, dim=-1)
         topv, topi = probs.topk(self.top_k, dim=-1)         # [T, k]
@@ -339,6 +348,11 @@ MIXTRAL_CONFIGS: Dict[str, MixtralConfig] = {
         name="tiny-mixtral", vocab_size=512, hidden_size=512, n_layers=2,
         n_heads=4, n_kv_heads=1, intermediate_size=512, max_position=4096,
         n_experts=4, top_k=2, tie_embeddings=False),
+    # 2-shardable tiny MoE for the gloo TP/EP tests (n_kv=2, 4 experts)
+    "tiny-mixtral-tp": MixtralConfig(
+        name="tiny-mixtral-tp", vocab_size=512, hidden_size=1024, n_layers=2,
+        n_heads=8, n_kv_heads=2, intermediate_size=512, max_position=4096,
+        n_experts=4, top_k=2, tie_embeddings=False),
     "mixtral-8x7b": MixtralConfig(
         name="mixtral-8x7b", vocab_size=32064, hidden_size=4096, n_layers=32,
         n_heads=32, n_kv_heads=8, intermediate_size=14336, max_position=8192,
@@ -350,5 +364,13 @@ def make_mixtral_instance(name: str, device: str, engine_cfg):
     """ModelInstance over a Mixtral model (engine factory hook)."""
     from ..engine.llm import ModelInstance
 
-    return ModelInstance(name, MIXTRAL_CONFIGS[name], device, engine_cfg,
+    inst = ModelInstance(name, MIXTRAL_CONFIGS[name], device, engine_cfg,
                          model_cls=MixtralForCausalLM)
+    mode = str(engine_cfg.get("moe_ep_mode", "auto"))
+    if mode not in ("auto", "a2a", "dense"):
+        raise ValueError(f"unsupported moe_ep_mode {mode!r}")
+    thr = int(engine_cfg.get("moe_a2a_min_tokens", MixtralMoE.A2A_MIN_TOKENS))
+    for layer in inst.model.layers:
+        layer.moe.ep_mode = mode
+        layer.moe.a2a_min_tokens = thr
+    return inst
