@@ -63,6 +63,9 @@ DEFAULTS: Dict[str, Dict[str, Any]] = {
         "interval_s": 30.0,  # monitor.go:117-129 defaults
         "timeout_s": 5.0,
         "retries": 3,
+        # amd-smi device-level fault probe (ECC / device presence) on GPU
+        "gpu_fault_probe": True,
+        "gpu_fault_interval_s": 30.0,
     },
     "metrics": {
         "sample_interval_s": 10.0,  # collector.go:205

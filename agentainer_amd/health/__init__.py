@@ -1,3 +1,3 @@
-from .monitor import HealthMonitor
+from .monitor import GpuFaultDetector, HealthMonitor
 
-__all__ = ["HealthMonitor"]
+__all__ = ["GpuFaultDetector", "HealthMonitor"]

@@ -19,10 +19,12 @@ microseconds; the reference's goroutine-per-agent was shaped by HTTP).
 
 from __future__ import annotations
 
+import json
+import subprocess
 import threading
 import time
 import traceback
-from typing import Any, Dict, Optional
+from typing import Any, Dict, List, Optional
 
 from ..registry import RUNNING, Manager
 from ..store import Store
@@ -33,16 +35,107 @@ DEFAULT_RETRIES = 3
 STATUS_TTL_S = 24 * 3600.0
 
 
+def _sum_keys(obj: Any, needle: str) -> int:
+    """Recursively sum every integer field whose key contains `needle`
+    (amd-smi's JSON nesting varies across versions; count totals don't)."""
+    total = 0
+    if isinstance(obj, dict):
+        for k, v in obj.items():
+            if needle in k.lower() and isinstance(v, (int, float)):
+                total += int(v)
+            else:
+                total += _sum_keys(v, needle)
+    elif isinstance(obj, list):
+        for v in obj:
+            total += _sum_keys(v, needle)
+    return total
+
+
+class GpuFaultDetector:
+    """Device-level GPU fault detection ("HBM ECC /
+    'GPU fell off the bus' detection via amd-smi"). The reference detects
+    container death three ways (events/reconcile/probes,
+    state_sync.go:253-309, monitor.go:207-251); the MI355X analog of the
+    hardware dying under an agent is the accelerator itself faulting —
+    which an IN-PROCESS probe cannot see (a hung HIP stream blocks the
+    caller), so this probes via an external `amd-smi` subprocess under a
+    hard timeout.
+
+    Faults reported:
+      * gpu_missing      — amd-smi sees no GPU / exits nonzero / times out
+      * ecc_uncorrectable — the uncorrectable-ECC total INCREASED since
+                            the baseline sample
+    Correctable ECC increases are recorded in the status but are not a
+    fault (HBM scrubbing handles them)."""
+
+    def __init__(self, interval_s: float = 30.0, timeout_s: float = 10.0,
+                 cmd: Optional[List[str]] = None):
+        self.interval_s = interval_s
+        self.timeout_s = timeout_s
+        self.cmd = cmd or ["amd-smi", "metric", "--ecc", "--json"]
+        self.fault: Optional[str] = None
+        self.status: Dict[str, Any] = {}
+        self._baseline_ue: Optional[int] = None
+        self._baseline_ce: int = 0
+        self._next_at = 0.0
+
+    def _sample(self) -> Dict[str, Any]:
+        try:
+            out = subprocess.run(self.cmd, capture_output=True, text=True,
+                                 timeout=self.timeout_s)
+        except (subprocess.TimeoutExpired, OSError) as e:
+            return {"ok": False, "error": type(e).__name__}
+        if out.returncode != 0:
+            return {"ok": False, "error": f"rc={out.returncode}",
+                    "stderr": out.stderr[-500:]}
+        try:
+            blob = json.loads(out.stdout)
+        except ValueError:
+            return {"ok": False, "error": "unparseable amd-smi output"}
+        gpus = blob if isinstance(blob, list) else [blob]
+        if not gpus:
+            return {"ok": False, "error": "no GPUs reported"}
+        return {"ok": True,
+                "n_gpus": len(gpus),
+                "uncorrectable": _sum_keys(blob, "uncorrect"),
+                "correctable": _sum_keys(blob, "correctable")
+                - _sum_keys(blob, "uncorrect")}
+
+    def check(self, now: Optional[float] = None) -> Optional[str]:
+        """Rate-limited probe; returns the current fault verdict."""
+        now = now if now is not None else time.time()
+        if now < self._next_at:
+            return self.fault
+        self._next_at = now + self.interval_s
+        s = self._sample()
+        s["last_check"] = now
+        self.status = s
+        if not s["ok"]:
+            self.fault = "gpu_missing"
+            return self.fault
+        ue = s["uncorrectable"]
+        if self._baseline_ue is None:
+            self._baseline_ue = ue
+            self._baseline_ce = s["correctable"]
+        if ue > self._baseline_ue:
+            self.fault = "ecc_uncorrectable"
+        else:
+            self.fault = None
+        return self.fault
+
+
 class HealthMonitor:
     def __init__(self, store: Store, manager: Manager,
                  interval_s: float = DEFAULT_INTERVAL_S,
                  timeout_s: float = DEFAULT_TIMEOUT_S,
-                 retries: int = DEFAULT_RETRIES):
+                 retries: int = DEFAULT_RETRIES,
+                 gpu_fault: Optional[GpuFaultDetector] = None):
         self.store = store
         self.manager = manager
         self.interval_s = interval_s
         self.timeout_s = timeout_s
         self.retries = retries
+        self.gpu_fault = gpu_fault
         self._watch: Dict[str, Dict[str, Any]] = {}
         self._lock = threading.RLock()
         self._stop = threading.Event()
@@ -110,6 +203,11 @@ class HealthMonitor:
 
     def check_due(self, now: Optional[float] = None) -> None:
         now = now if now is not None else time.time()
+        if self.gpu_fault is not None:
+            self.gpu_fault.check(now)
+            self.store.set("health:gpu",
+                           {"fault": self.gpu_fault.fault,
+                            **self.gpu_fault.status}, ttl=STATUS_TTL_S)
         with self._lock:
             due = [aid for aid, w in self._watch.items() if w["next_at"] <= now]
         for aid in due:
@@ -125,6 +223,9 @@ class HealthMonitor:
             healthy = bool(self.manager.engine.health_probe(agent_id))
         except Exception:
             healthy = False
+        gpu_fault = self.gpu_fault.fault if self.gpu_fault is not None else None
+        if gpu_fault is not None:
+            healthy = False  # every agent on a faulted device is down
         with self._lock:
             w["next_at"] = now + w["interval"]
             if healthy:
@@ -139,6 +240,8 @@ class HealthMonitor:
             "consecutive_failures": failures,
             "last_check": now,
         }
+        if gpu_fault is not None:
+            status["gpu_fault"] = gpu_fault
         self.store.set(f"health:{agent_id}", status, ttl=STATUS_TTL_S)
         if not healthy and failures >= threshold:
             agent = self.manager.try_get(agent_id)

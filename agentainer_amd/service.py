@@ -85,11 +85,22 @@ class Runtime:
                                    interval_s=float(feats.get("replay_interval_s", 5.0)),
                                    inflight=self._inflight)
         self.reconciler = Reconciler(self.agents, interval_s=10.0)
+        # device-level fault probing (amd-smi: ECC counts, device
+        # presence) on GPU engines — SURVEY.md §5 failure detection
+        gpu_fault = None
+        if (getattr(engine, "device", "cpu").startswith("cuda")
+                and bool(self.config.get("health", "gpu_fault_probe", True))):
+            from .health import GpuFaultDetector
+
+            gpu_fault = GpuFaultDetector(
+                interval_s=float(self.config.get("health",
+                                                 "gpu_fault_interval_s", 30.0)))
         self.health = HealthMonitor(
             self.store, self.agents,
             interval_s=float(self.config.get("health", "interval_s", 30.0)),
             timeout_s=float(self.config.get("health", "timeout_s", 5.0)),
             retries=int(self.config.get("health", "retries", 3)),
+            gpu_fault=gpu_fault,
         )
         self.metrics = MetricsCollector(
             self.store, self.agents,
