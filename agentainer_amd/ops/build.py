@@ -21,8 +21,14 @@ from torch.utils import cpp_extension
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
-OUT = os.path.join(HERE, "_hip_ops.so")
-OBJ_DIR = os.path.join(HERE, "_build")
+# AGENTAINER_ASAN=1 builds a host-AddressSanitized variant (SURVEY.md §5
+# race-detection row): the C++ binding layer (all torch interop, tensor
+# lifetime, host bookkeeping) compiles with -fsanitize=address; device
+# .hip kernels stay unsanitized (device ASAN needs xnack+, not a serving
+# config). Run it with LD_PRELOAD=$(clang -print-file-name=libclang_rt.asan-x86_64.so).
+ASAN = os.environ.get("AGENTAINER_ASAN", "") == "1"
+OUT = os.path.join(HERE, "_hip_ops_asan.so" if ASAN else "_hip_ops.so")
+OBJ_DIR = os.path.join(HERE, "_build_asan" if ASAN else "_build")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 SOURCES = [
@@ -77,8 +83,10 @@ def build(verbose: bool = True, force: bool = False) -> str:
                 and os.path.getmtime(obj) > _newest([src_path, os.path.join(CSRC, "common.h")])
                 and not force):
             continue
-        cmd = (["hipcc", "-c", src_path, "-o", obj] + cxx_flags + inc_flags
-               + (["-x", "hip"] if src.endswith(".cpp") else []))
+        san = (["-fsanitize=address", "-shared-libsan"]
+               if ASAN and src.endswith(".cpp") else [])
+        cmd = (["hipcc", "-c", src_path, "-o", obj] + cxx_flags + san
+               + inc_flags + (["-x", "hip"] if src.endswith(".cpp") else []))
         if verbose:
             print("[ops.build]", " ".join(cmd), flush=True)
         procs.append((src, subprocess.Popen(cmd, stdout=subprocess.PIPE,
@@ -92,6 +100,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
     if failed:
         raise RuntimeError("HIP extension build failed")
     link = (["hipcc", "-shared", "-fPIC", "-o", OUT] + objs
+            + (["-fsanitize=address", "-shared-libsan"] if ASAN else [])
             + [f"-L{torch_lib}", "-ltorch", "-ltorch_python", "-lc10",
                "-ltorch_hip", "-lc10_hip", f"-Wl,-rpath,{torch_lib}"])
     if verbose:
