@@ -68,8 +68,10 @@ class GpuFaultDetector:
     Correctable ECC increases are recorded in the status but are not a
     fault (HBM scrubbing handles them)."""
 
-    def __init__(self, interval_s: float = 30.0, timeout_s: float = 10.0,
+    def __init__(self, interval_s: float = 30.0, timeout_s: float = 30.0,
                  cmd: Optional[List[str]] = None):
+        # timeout covers amd-smi's cold start (first invocation on a fresh
+        # node measures >10s while it enumerates devices)
         self.interval_s = interval_s
         self.timeout_s = timeout_s
         self.cmd = cmd or ["amd-smi", "metric", "--ecc", "--json"]
@@ -92,7 +94,12 @@ class GpuFaultDetector:
             blob = json.loads(out.stdout)
         except ValueError:
             return {"ok": False, "error": "unparseable amd-smi output"}
-        gpus = blob if isinstance(blob, list) else [blob]
+        if isinstance(blob, dict) and isinstance(blob.get("gpu_data"), list):
+            gpus = blob["gpu_data"]  # amd-smi 26.x: {"gpu_data": [...]}
+        elif isinstance(blob, list):
+            gpus = blob
+        else:
+            gpus = [blob]
         if not gpus:
             return {"ok": False, "error": "no GPUs reported"}
         return {"ok": True,
