@@ -1,3 +1,4 @@
-from .collector import LatencyWindow, MetricsCollector
+from .collector import (DeviceMetricsSampler, LatencyWindow,
+                        MetricsCollector)
 
-__all__ = ["LatencyWindow", "MetricsCollector"]
+__all__ = ["DeviceMetricsSampler", "LatencyWindow", "MetricsCollector"]

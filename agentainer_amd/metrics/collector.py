@@ -19,6 +19,7 @@ the reference where storage stubs left the collector dormant
 from __future__ import annotations
 
 import json
+import subprocess
 import threading
 import time
 import traceback
@@ -30,6 +31,70 @@ from ..store import Store
 CURRENT_TTL_S = 3600.0
 HISTORY_RETENTION_S = 24 * 3600.0
 SAMPLE_INTERVAL_S = 10.0
+
+
+def _numeric_leaves(obj: Any, needle: str, path: str = "") -> Dict[str, float]:
+    """Flatten numeric JSON leaves whose key path contains `needle`."""
+    out: Dict[str, float] = {}
+    if isinstance(obj, dict):
+        for k, v in obj.items():
+            p = f"{path}.{k}" if path else str(k)
+            if isinstance(v, (int, float)) and needle in p.lower():
+                out[p] = float(v)
+            else:
+                out.update(_numeric_leaves(v, needle, p))
+    elif isinstance(obj, list):
+        for i, v in enumerate(obj):
+            out.update(_numeric_leaves(v, needle, f"{path}[{i}]"))
+    return out
+
+
+class DeviceMetricsSampler:
+    """Device-level counters via amd-smi (SURVEY.md §5 metrics row: the
+    engine-counter story needs HBM + xGMI link throughput, which only
+    the platform tool exposes). Best-effort: xGMI counters exist only on
+    multi-GPU hives; repeated failures disable the sampler so a missing
+    tool never stalls the 10s metrics loop."""
+
+    def __init__(self, timeout_s: float = 30.0,
+                 cmd: Optional[List[str]] = None):
+        self.cmd = cmd or ["amd-smi", "metric", "--json"]
+        self.timeout_s = timeout_s
+        self._fails = 0
+        self._prev: Dict[str, float] = {}
+        self._prev_ts: Optional[float] = None
+
+    @property
+    def enabled(self) -> bool:
+        return self._fails < 3
+
+    def sample(self, now: Optional[float] = None) -> Dict[str, Any]:
+        if not self.enabled:
+            return {}
+        now = now if now is not None else time.time()
+        try:
+            out = subprocess.run(self.cmd, capture_output=True, text=True,
+                                 timeout=self.timeout_s)
+            blob = json.loads(out.stdout)
+        except Exception:
+            self._fails += 1
+            return {}
+        self._fails = 0
+        xgmi = _numeric_leaves(blob, "xgmi")
+        result: Dict[str, Any] = {}
+        if xgmi:
+            result["xgmi_counters"] = xgmi
+            if self._prev_ts is not None:
+                dt = max(1e-9, now - self._prev_ts)
+                rates = {k: (v - self._prev.get(k, v)) / dt
+                         for k, v in xgmi.items()
+                         if v >= self._prev.get(k, v)}
+                # only counter-like fields produce meaningful rates, but
+                # publishing all deltas keeps this schema-agnostic
+                result["xgmi_per_s"] = rates
+            self._prev = xgmi
+            self._prev_ts = now
+        return result
 
 
 class LatencyWindow:
@@ -61,11 +126,13 @@ class LatencyWindow:
 class MetricsCollector:
     def __init__(self, store: Store, manager: Manager,
                  sample_interval_s: float = SAMPLE_INTERVAL_S,
-                 history_retention_s: float = HISTORY_RETENTION_S):
+                 history_retention_s: float = HISTORY_RETENTION_S,
+                 device_sampler: Optional[DeviceMetricsSampler] = None):
         self.store = store
         self.manager = manager
         self.sample_interval_s = sample_interval_s
         self.history_retention_s = history_retention_s
+        self.device_sampler = device_sampler
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self._latency: Dict[str, LatencyWindow] = {}
@@ -117,6 +184,18 @@ class MetricsCollector:
             if agent.status != RUNNING:
                 continue
             self.sample_agent(agent.id, per_agent.get(agent.id, {}), now=now)
+        # device-level sample: HBM from the engine, xGMI from amd-smi
+        dev: Dict[str, Any] = {"ts": now}
+        for k in ("hbm_total_bytes", "hbm_free_bytes", "hbm_torch_allocated"):
+            if k in engine_stats:
+                dev[k] = engine_stats[k]
+        if self.device_sampler is not None:
+            dev.update(self.device_sampler.sample(now))
+        if len(dev) > 1:
+            self.store.set("metrics:current:device", dev, ttl=CURRENT_TTL_S)
+            self.store.zadd("metrics:history:device", now, json.dumps(dev))
+            self.store.zremrangebyscore("metrics:history:device", 0,
+                                        now - self.history_retention_s)
 
     def sample_agent(self, agent_id: str, eng: Dict[str, Any],
                      now: Optional[float] = None) -> Dict[str, Any]:

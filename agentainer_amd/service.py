@@ -102,9 +102,15 @@ class Runtime:
             retries=int(self.config.get("health", "retries", 3)),
             gpu_fault=gpu_fault,
         )
+        dev_sampler = None
+        if getattr(engine, "device", "cpu").startswith("cuda"):
+            from .metrics import DeviceMetricsSampler
+
+            dev_sampler = DeviceMetricsSampler()
         self.metrics = MetricsCollector(
             self.store, self.agents,
             sample_interval_s=float(self.config.get("metrics", "sample_interval_s", 10.0)),
+            device_sampler=dev_sampler,
         )
         self.backups = BackupManager(self.store, self.agents, os.path.join(root, "backups"))
         self._started = False

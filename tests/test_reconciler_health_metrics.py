@@ -147,3 +147,73 @@ def test_watchdog_marks_stuck_engine_unhealthy(tmp_path):
     inst._step_started = None
     assert eng.health_probe(a.id) is True
     store.close()
+
+
+# ---------------- device metrics (xGMI / HBM) -----------------------------
+
+def test_device_metrics_sampler_stub(tmp_path):
+    """xGMI link counters flow into metrics:current:device with per-
+    interval rates (SURVEY.md §5 metrics row: xGMI link throughput)."""
+    import json
+    import stat
+
+    from agentainer_amd.metrics import DeviceMetricsSampler
+
+    state = tmp_path / "n"
+    state.write_text("0")
+    script = tmp_path / "smi"
+    script.write_text(f"""#!/usr/bin/env python3
+import json
+i = int(open({str(state)!r}).read() or 0)
+open({str(state)!r}, "w").write(str(i + 1))
+print(json.dumps({{"gpu_data": [{{"gpu": 0, "xgmi": {{
+    "link_0_read_kb": 1000 * (i + 1), "link_0_write_kb": 500 * (i + 1)}}}}]}}))
+""")
+    script.chmod(script.stat().st_mode | stat.S_IEXEC)
+    s = DeviceMetricsSampler(cmd=[str(script)])
+    first = s.sample(now=100.0)
+    assert "xgmi_counters" in first and "xgmi_per_s" not in first
+    second = s.sample(now=110.0)
+    rates = second["xgmi_per_s"]
+    read_key = [k for k in rates if "read" in k][0]
+    assert abs(rates[read_key] - 100.0) < 1e-6  # 1000 KB over 10 s
+
+
+def test_device_metrics_sampler_disables_after_failures(tmp_path):
+    from agentainer_amd.metrics import DeviceMetricsSampler
+
+    s = DeviceMetricsSampler(cmd=[str(tmp_path / "missing")])
+    for _ in range(3):
+        assert s.sample() == {}
+    assert not s.enabled
+    assert s.sample() == {}  # permanently off, no subprocess attempts
+
+
+def test_collector_publishes_device_sample(tmp_path):
+    """Engine HBM fields land in metrics:current:device even without a
+    working amd-smi (CPU path: engine stats only, no device sampler)."""
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.echo import EchoEngine
+    from agentainer_amd.metrics import MetricsCollector
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    store = Store(str(tmp_path / "state"), sync="never")
+    eng = EchoEngine(store)
+    # fake a GPU-shaped stats() payload
+    orig = eng.stats
+
+    def stats():
+        out = orig()
+        out["hbm_total_bytes"] = 288 << 30
+        out["hbm_free_bytes"] = 100 << 30
+        return out
+
+    eng.stats = stats
+    man = Manager(store, eng, cfg)
+    col = MetricsCollector(store, man)
+    col.sample_all(now=100.0)
+    dev = store.get("metrics:current:device")
+    assert dev["hbm_total_bytes"] == 288 << 30
+    store.close()
