@@ -51,7 +51,8 @@ class KVCheckpoint:
 class KVCacheManager:
     def __init__(self, n_layers: int, n_kv: int, head_dim: int, page_size: int,
                  n_pages: int, device="cpu", dtype=torch.bfloat16,
-                 max_slots: int = 1024, max_pages_per_seq: int = 512):
+                 max_slots: int = 1024, max_pages_per_seq: int = 512,
+                 mirrors: Optional[bool] = None):
         assert head_dim % 8 == 0
         self.n_layers = n_layers
         self.n_kv = n_kv
@@ -84,13 +85,17 @@ class KVCacheManager:
         self._lock = threading.RLock()
         # shorts per page per layer: K plane + V plane
         self.page_shorts = 2 * n_kv * head_dim * page_size
-        # ---- device mirrors for the in-graph decode step (GPU only) ----
+        # ---- device mirrors for the in-graph decode step ----
         # dev_page_table[slot] / dev_seq_lens[slot] are the ground truth the
-        # captured decode graph reads; host bookkeeping mirrors them.
+        # captured decode graph reads; host bookkeeping mirrors them. On GPU
+        # they live in HBM; `mirrors=True` on CPU materializes them as CPU
+        # tensors so the device-decode path (and the async-TP protocol on
+        # top of it) is testable without hardware.
         self.is_gpu = str(device) not in ("cpu",)
+        self.mirrors = self.is_gpu if mirrors is None else bool(mirrors)
         self.max_slots = max_slots
         self.max_pages_per_seq = max_pages_per_seq
-        if self.is_gpu:
+        if self.mirrors:
             self.dev_page_table = torch.zeros(max_slots, max_pages_per_seq,
                                               dtype=torch.int32, device=device)
             self.dev_seq_lens = torch.full((max_slots,), -1, dtype=torch.int32,
@@ -149,7 +154,7 @@ class KVCacheManager:
             slot = self._free_slots.pop()
             self._slot_of[seq_id] = slot
             self._seq_of_slot[slot] = seq_id
-            if self.is_gpu:
+            if self.mirrors:
                 self.dev_seq_lens[slot] = 0
             return s
 
@@ -163,7 +168,7 @@ class KVCacheManager:
 
     def push_dev(self, seq_id: str) -> None:
         """Sync a sequence's page row + length to the device mirrors."""
-        if not self.is_gpu:
+        if not self.mirrors:
             return
         with self._lock:
             s = self._seqs[seq_id]
@@ -186,7 +191,7 @@ class KVCacheManager:
                     raise OutOfPages(f"KV pool exhausted ({self.n_pages} pages)")
                 page = self._free.pop()
                 s.pages.append(page)
-                if self.is_gpu:
+                if self.mirrors:
                     self.dev_page_table[self._slot_of[seq_id], page_idx] = page
 
     def advance_host(self, seq_id: str) -> None:
@@ -211,7 +216,7 @@ class KVCacheManager:
                             f"KV pool exhausted ({self.n_pages} pages)")
                     page = self._free.pop()
                     s.pages.append(page)
-                    if self.is_gpu:
+                    if self.mirrors:
                         self.dev_page_table[self._slot_of[seq_id],
                                             page_idx] = page
                 rows.append(self._slot_of[seq_id])
@@ -233,9 +238,9 @@ class KVCacheManager:
             slots = []
             for seq_id in seq_ids:
                 self._seqs[seq_id].length -= 1
-                if self.is_gpu:
+                if self.mirrors:
                     slots.append(self._slot_of[seq_id])
-        if self.is_gpu and slots:
+        if self.mirrors and slots:
             idx = torch.tensor(slots, dtype=torch.long, device=self.device)
             self.dev_seq_lens.index_add_(
                 0, idx, torch.full((len(slots),), -1, dtype=torch.int32,
@@ -270,7 +275,7 @@ class KVCacheManager:
                 self._free.append(tail.pop())
             dst.pages = list(shared) + tail
             dst.length = n_tokens
-            if self.is_gpu:
+            if self.mirrors:
                 slot = self._slot_of[seq_id]
                 self.dev_page_table[slot, :len(dst.pages)] = torch.tensor(
                     dst.pages, dtype=torch.int32, device=self.device)
@@ -296,7 +301,7 @@ class KVCacheManager:
             slot = self._slot_of.pop(seq_id, None)
             if slot is not None:
                 self._seq_of_slot.pop(slot, None)
-                if self.is_gpu:
+                if self.mirrors:
                     self.dev_seq_lens[slot] = -1
                 self._free_slots.append(slot)
 
@@ -307,7 +312,7 @@ class KVCacheManager:
             self._release_pages(s.pages)
             s.pages = []
             s.length = 0
-            if self.is_gpu:
+            if self.mirrors:
                 self.dev_seq_lens[self._slot_of[seq_id]] = 0
 
     def reserve(self, seq_id: str, n_tokens: int) -> None:
@@ -329,7 +334,7 @@ class KVCacheManager:
             for _ in range(short):
                 page = self._free.pop()
                 s.pages.append(page)
-                if self.is_gpu:
+                if self.mirrors:
                     self.dev_page_table[self._slot_of[seq_id],
                                         len(s.pages) - 1] = page
 

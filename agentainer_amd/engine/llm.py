@@ -47,6 +47,25 @@ DEFAULT_MAX_NEW = 64
 # 0 is reserved for pickled commands
 OP_DECODE = 1   # [OP, model_idx, B, slot*B, token*B] — the per-step hot path
 OP_BARRIER = 2  # [OP] — timing fence (bench.py --tp)
+# async (speculative) decode: rollbacks from the previous resolve are
+# prepended, then per-row (slot, token|-1, prev_idx, eff_seed, temp_bits,
+# top_p_bits) — workers feed carried-over rows from their OWN previous
+# device sample (logits are bitwise-identical across ranks after the
+# all-reduce, and sampling is seed-deterministic, so every rank samples
+# the same token without any extra communication)
+OP_DECODE_ASYNC = 3  # [OP, model_idx, B, nrb, rb_slot*nrb, row*6*B]
+OP_ROLLBACK = 4      # [OP, model_idx, n, slot*n] — flush before other cmds
+
+
+def _f2i(x: float) -> int:
+    """float32 bit pattern as int (plan-channel encoding)."""
+    import struct
+    return struct.unpack("<I", struct.pack("<f", float(x)))[0]
+
+
+def _i2f(x: int) -> float:
+    import struct
+    return struct.unpack("<f", struct.pack("<I", int(x) & 0xFFFFFFFF))[0]
 
 
 class EngineDead(RuntimeError):
@@ -169,7 +188,10 @@ class ModelInstance:
                                   max(cfg.n_kv_heads // self.tp_size, 1),
                                   cfg.head_dim, page_size, n_pages,
                                   device=device, dtype=kv_dtype,
-                                  max_slots=max_seqs)
+                                  max_slots=max_seqs,
+                                  mirrors=(device.startswith("cuda") or bool(
+                                      engine_cfg.get("async_decode_emulate",
+                                                     False))))
         self.max_batch_tokens = int(engine_cfg.get("max_batch_tokens", 8192))
         self.max_decode_batch = int(engine_cfg.get("max_decode_batch", 256))
         self.refcount = 0
@@ -217,17 +239,31 @@ class ModelInstance:
         # (_get_graph) identically on every rank.
         self.use_graph = (self.is_gpu
                           and bool(engine_cfg.get("graph_capture", True)))
+        # test-only: run the device-decode machinery (mirrors, slot-fed
+        # decode, async protocol) on CPU tensors so gloo world-2 tests can
+        # validate the exact code path the GPU runs
+        self._emulate_dev = (not self.is_gpu
+                             and bool(engine_cfg.get("async_decode_emulate",
+                                                     False)))
+        self.dev_decode = self.is_gpu or self._emulate_dev
         # async decode (speculative one-step lag): the next decode step is
         # launched with device-fed tokens BEFORE the previous step's tokens
         # reach the host, so per-step host bookkeeping overlaps GPU work.
         # Rows that resolve as finished get their speculative append rolled
-        # back (kvm.rollback_many, stream-ordered). TP=1 only: the worker
-        # loop executes explicit plans and has no speculative mirror.
-        self.async_decode = (self.use_graph and self.tp_size == 1
+        # back (kvm.rollback_many, stream-ordered). Under TP, rank 0
+        # broadcasts OP_DECODE_ASYNC plans; every rank samples on-device
+        # from its own (identical) logits, so the carried-over token feed
+        # needs no cross-rank traffic.
+        self.async_decode = ((self.use_graph or self._emulate_dev)
                              and bool(engine_cfg.get("async_decode", True)))
         # index of this model in the engine's creation order — the wire
         # identity in binary step plans (identical on every TP rank)
         self._model_idx = 0
+        # rollback slots applied locally but not yet mirrored to workers
+        # (flushed at the head of the next broadcast, any kind)
+        self._pending_rb: List[int] = []
+        # worker-side speculative mirror: previous step's device sample
+        self._spec_worker: Optional[Dict[str, Any]] = None
         self._spec: Optional[Dict[str, Any]] = None
         self._graphs: Dict[int, Dict[str, Any]] = {}
         # prefill runs on its own HIP stream so a prefill batch overlaps
@@ -245,7 +281,7 @@ class ModelInstance:
         # land in a page the prefill just re-allocated
         self._pages_freed = False
         self._pad_slot = -1
-        if self.is_gpu:
+        if self.dev_decode:
             # reserve one sequence slot as the graph's pad row target
             self._pad_slot = self.kvm._free_slots.pop()
             self.kvm.dev_seq_lens[self._pad_slot] = -1
@@ -329,6 +365,10 @@ class ModelInstance:
             self._step_started = None
             self.last_step_t = time.time()
             return ran
+        if self._spec is not None:
+            # async was disabled mid-flight (graph-capture fallback):
+            # resolve the orphaned speculative step before going eager
+            self.drain_async()
         admitted = self._admit()
         if admitted:
             try:
@@ -363,26 +403,48 @@ class ModelInstance:
             B = len(batch)
             seq_ids = [self._bindings[r.agent_id].seq_id for r in batch]
             rows = kvm.decode_batch_prepare(seq_ids)
+            prev = self._spec
+            pos = {id(r): i for i, r in enumerate(prev["reqs"])} if prev else {}
+            idx = [pos.get(id(r), -1) for r in batch]
+            # effective sampling seed: a carried-over row has ONE unresolved
+            # token in flight ahead of this sample (generated is appended at
+            # resolve, one step later), so its counter runs one ahead of
+            # len(generated) — this keeps the sampled stream identical to
+            # the synchronous path's seed+len(generated) sequence
+            seeds_eff = [(r.seed + len(r.generated)
+                          + (1 if idx[i] >= 0 else 0)) & 0x7FFFFFFFFFFFFFFF
+                         for i, r in enumerate(batch)]
+            if self.tp_size > 1 and self.tp_rank == 0:
+                # async plan: pending rollbacks + per-row feed/sample spec;
+                # workers replicate the launch and sample on-device from
+                # their own identical logits (no return traffic)
+                rb, self._pending_rb = self._pending_rb, []
+                vec = [OP_DECODE_ASYNC, self._model_idx, B, len(rb)] + rb
+                for i, r in enumerate(batch):
+                    vec += [rows[i],
+                            r.generated[-1] if idx[i] < 0 else -1,
+                            idx[i],
+                            seeds_eff[i],
+                            _f2i(r.temperature), _f2i(r.top_p)]
+                par.send_ints(vec)
             entry = self._get_graph(bucket)
             # the PREVIOUS step's non-blocking H2D copies read these pinned
             # staging buffers; wait for that DMA before rewriting them (a
             # torn read would feed wrong slots/tokens into the graph)
-            ev = entry.get("h2d_ev")
+            ev = entry.get("h2d_ev") if self.is_gpu else None
             if ev is not None:
                 ev.synchronize()
             entry["rows_pin"][:B] = torch.tensor(rows, dtype=torch.long)
             entry["rows_pin"][B:] = self._pad_slot
             entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
-            prev = self._spec
-            pos = {id(r): i for i, r in enumerate(prev["reqs"])} if prev else {}
-            idx = [pos.get(id(r), -1) for r in batch]
             entry["ids_pin"][:B] = torch.tensor(
                 [r.generated[-1] if idx[i] < 0 else 0
                  for i, r in enumerate(batch)], dtype=torch.long)
             entry["ids"][:B].copy_(entry["ids_pin"][:B], non_blocking=True)
-            if ev is None:
-                ev = entry["h2d_ev"] = torch.cuda.Event()
-            ev.record()
+            if self.is_gpu:
+                if ev is None:
+                    ev = entry["h2d_ev"] = torch.cuda.Event()
+                ev.record()
             if prev is not None and any(x >= 0 for x in idx):
                 gidx = torch.tensor([max(x, 0) for x in idx], dtype=torch.long,
                                     device=dev)
@@ -391,9 +453,15 @@ class ModelInstance:
                 gathered = prev["sampled"].index_select(0, gidx)
                 entry["ids"][:B].copy_(
                     torch.where(mask, gathered, entry["ids"][:B]))
-            entry["graph"].replay()
-            logits = entry["logits"][:B]
-            sampled = self._sample_device(logits, batch)
+            if entry["graph"] is not None:
+                entry["graph"].replay()
+                logits = entry["logits"][:B]
+            else:  # eager fallback (capture failure / CPU emulation)
+                logits = self._device_decode_fwd(entry["rows"], entry["ids"],
+                                                 entry["inc"])[:B]
+            sampled = self._sample_device_params(
+                logits, [r.temperature for r in batch],
+                [r.top_p for r in batch], seeds_eff)
             kvm.advance_many(seq_ids)
             self.decode_tokens += B
             self.occupancy_acc += B / max(1, self.max_decode_batch)
@@ -482,6 +550,11 @@ class ModelInstance:
                             rollback_seqs.append(b.seq_id)
         if rollback_seqs:
             self._pages_freed = True  # see _step_async phase 3
+            if self.tp_size > 1 and self.tp_rank == 0:
+                # applied locally now; workers get them at the head of the
+                # NEXT broadcast (async plan or any other command)
+                self._pending_rb.extend(self.kvm.slot(s)
+                                        for s in rollback_seqs)
         self.kvm.rollback_many(rollback_seqs)
 
     def drain_async(self) -> None:
@@ -563,10 +636,23 @@ class ModelInstance:
 
     def _sample_device(self, logits: torch.Tensor,
                        reqs: List[GenRequest]) -> torch.Tensor:
+        return self._sample_device_params(
+            logits,
+            [r.temperature for r in reqs],
+            [r.top_p for r in reqs],
+            [(r.seed + len(r.generated)) & 0x7FFFFFFFFFFFFFFF for r in reqs])
+
+    def _sample_device_params(self, logits: torch.Tensor,
+                              temps_l: List[float], tops_l: List[float],
+                              seeds_l: List[int]) -> torch.Tensor:
+        """Param-vector core shared by rank 0 (from GenRequests) and TP
+        workers (from the async plan): given bitwise-identical logits and
+        the same (temp, top_p, effective-seed) rows, every rank samples
+        the same tokens."""
         B = logits.size(0)
         out = torch.empty(B, dtype=torch.long, device=logits.device)
-        greedy_rows = [i for i, r in enumerate(reqs) if r.temperature <= 0.0]
-        samp_rows = [i for i, r in enumerate(reqs) if r.temperature > 0.0]
+        greedy_rows = [i for i in range(B) if temps_l[i] <= 0.0]
+        samp_rows = [i for i in range(B) if temps_l[i] > 0.0]
         lb = logits.to(torch.bfloat16).contiguous()
         if greedy_rows:
             if len(greedy_rows) == B:
@@ -578,19 +664,24 @@ class ModelInstance:
                 out[greedy_rows] = sub
         if samp_rows:
             sub = torch.empty(len(samp_rows), dtype=torch.long, device=logits.device)
-            temps = torch.tensor([reqs[i].temperature for i in samp_rows],
+            temps = torch.tensor([temps_l[i] for i in samp_rows],
                                  dtype=torch.float32, device=logits.device)
-            tps = torch.tensor([reqs[i].top_p for i in samp_rows],
+            tps = torch.tensor([tops_l[i] for i in samp_rows],
                                dtype=torch.float32, device=logits.device)
-            seeds = torch.tensor(
-                [(reqs[i].seed + len(reqs[i].generated)) & 0x7FFFFFFFFFFFFFFF
-                 for i in samp_rows], dtype=torch.int64, device=logits.device)
+            seeds = torch.tensor([seeds_l[i] for i in samp_rows],
+                                 dtype=torch.int64, device=logits.device)
             ops.topp_sample(sub, lb[samp_rows].contiguous(), temps, tps, seeds)
             out[samp_rows] = sub
         return out
 
     def _bcast(self, cmd):
         if self.tp_size > 1 and self.tp_rank == 0:
+            if self._pending_rb:
+                # flush speculative rollbacks FIRST: the command below may
+                # read KV state (prefill positions, unbind) that is only
+                # consistent across ranks once workers applied them
+                rb, self._pending_rb = self._pending_rb, []
+                par.send_ints([OP_ROLLBACK, self._model_idx, len(rb)] + rb)
             par.broadcast_obj(cmd)
 
     def _prefill(self, reqs: List[GenRequest]):
@@ -672,7 +763,7 @@ class ModelInstance:
             ids.extend(tokens)
             positions.extend(range(prev, prev + n))
             slots.extend(self.kvm.append_slots(seq_id, n))
-        if self.is_gpu:
+        if self.dev_decode:
             for sid in seq_ids:
                 self.kvm.push_dev(sid)
         md = AttnMetadata(
@@ -727,8 +818,8 @@ class ModelInstance:
         ids = torch.zeros(bucket, dtype=torch.long, device=dev)
         inc = torch.ones(bucket, dtype=torch.int32, device=dev)
         rows_pin = torch.full((bucket,), self._pad_slot, dtype=torch.long,
-                              pin_memory=True)
-        ids_pin = torch.zeros(bucket, dtype=torch.long, pin_memory=True)
+                              pin_memory=self.is_gpu)
+        ids_pin = torch.zeros(bucket, dtype=torch.long, pin_memory=self.is_gpu)
         entry = {"rows": rows, "ids": ids, "inc": inc, "rows_pin": rows_pin,
                  "ids_pin": ids_pin, "graph": None, "logits": None}
         if self.use_graph:
@@ -754,13 +845,16 @@ class ModelInstance:
 
     def _decode(self, reqs: List[GenRequest]):
         B0 = len(reqs)
-        if self.is_gpu:
+        if self.dev_decode:
             bucket = min(self._bucket(B0), max(self.max_decode_batch, 1))
             if bucket < B0:
                 reqs = reqs[:bucket]
         plan = [(self._bindings[r.agent_id].seq_id, r.generated[-1])
                 for r in reqs]
         if self.tp_size > 1 and self.tp_rank == 0:
+            if self._pending_rb:  # async->eager fallback edge
+                rb, self._pending_rb = self._pending_rb, []
+                par.send_ints([OP_ROLLBACK, self._model_idx, len(rb)] + rb)
             # binary hot path: ONE int-tensor broadcast on the gloo plan
             # channel (slot ids are deterministic across ranks), replacing
             # the round-1 per-step pickled broadcast_object_list
@@ -779,7 +873,7 @@ class ModelInstance:
 
     def _decode_exec(self, plan) -> torch.Tensor:
         """One batched decode step; runs identically on every TP rank."""
-        if self.is_gpu:
+        if self.dev_decode:
             return self._decode_exec_gpu(plan)
         dev = self.device
         ids, positions, slots, seq_ids = [], [], [], []
@@ -808,7 +902,7 @@ class ModelInstance:
         entry = self._get_graph(bucket)
         # TP workers run ahead with no host sync per step: wait out the
         # previous step's H2D DMA before rewriting the pinned staging
-        ev = entry.get("h2d_ev")
+        ev = entry.get("h2d_ev") if self.is_gpu else None
         if ev is not None:
             ev.synchronize()
         entry["rows_pin"][:B] = torch.tensor(row_ids, dtype=torch.long)
@@ -817,9 +911,10 @@ class ModelInstance:
                                             dtype=torch.long)
         entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
         entry["ids"].copy_(entry["ids_pin"], non_blocking=True)
-        if ev is None:
-            ev = entry["h2d_ev"] = torch.cuda.Event()
-        ev.record()
+        if self.is_gpu:
+            if ev is None:
+                ev = entry["h2d_ev"] = torch.cuda.Event()
+            ev.record()
         if entry["graph"] is not None:
             entry["graph"].replay()
             logits = entry["logits"][:B]
@@ -828,6 +923,74 @@ class ModelInstance:
                                              entry["inc"])[:B]
         kvm.advance_many([s for s, _t in plan])
         return logits
+
+    def _decode_async_worker(self, cmd: List[int]) -> None:
+        """Worker half of the speculative decode step (OP_DECODE_ASYNC).
+
+        Mirrors rank 0's phase-1 launch exactly: apply the prepended
+        rollbacks, prepare pages, feed carried-over rows from this rank's
+        OWN previous device sample (identical logits => identical sample),
+        replay the graph, sample on-device with the plan's per-row
+        (temp, top_p, effective-seed), advance. No host sync, no return
+        traffic — the worker runs ahead of its GPU."""
+        kvm = self.kvm
+        B = cmd[2]
+        nrb = cmd[3]
+        off = 4
+        rb = cmd[off:off + nrb]
+        off += nrb
+        if rb:
+            kvm.rollback_many([kvm.seq_of_slot(s) for s in rb])
+        rows_l: List[int] = []
+        toks_l: List[int] = []
+        idx_l: List[int] = []
+        seeds_l: List[int] = []
+        temps_l: List[float] = []
+        tops_l: List[float] = []
+        for i in range(B):
+            slot, tok, pidx, seed, tbits, pbits = cmd[off + 6 * i:off + 6 * i + 6]
+            rows_l.append(slot)
+            toks_l.append(tok if tok >= 0 else 0)
+            idx_l.append(pidx)
+            seeds_l.append(seed)
+            temps_l.append(_i2f(tbits))
+            tops_l.append(_i2f(pbits))
+        seq_ids = [kvm.seq_of_slot(s) for s in rows_l]
+        kvm.decode_batch_prepare(seq_ids)  # same pages, deterministic
+        bucket = min(self._bucket(B), max(self.max_decode_batch, 1))
+        entry = self._get_graph(bucket)
+        ev = entry.get("h2d_ev") if self.is_gpu else None
+        if ev is not None:
+            ev.synchronize()
+        entry["rows_pin"][:B] = torch.tensor(rows_l, dtype=torch.long)
+        entry["rows_pin"][B:] = self._pad_slot
+        entry["rows"].copy_(entry["rows_pin"], non_blocking=True)
+        entry["ids_pin"][:B] = torch.tensor(toks_l, dtype=torch.long)
+        entry["ids"][:B].copy_(entry["ids_pin"][:B], non_blocking=True)
+        if self.is_gpu:
+            if ev is None:
+                ev = entry["h2d_ev"] = torch.cuda.Event()
+            ev.record()
+        prev = self._spec_worker
+        if prev is not None and any(x >= 0 for x in idx_l):
+            dev = self.device
+            gidx = torch.tensor([max(x, 0) for x in idx_l], dtype=torch.long,
+                                device=dev)
+            mask = torch.tensor([x >= 0 for x in idx_l], dtype=torch.bool,
+                                device=dev)
+            gathered = prev["sampled"].index_select(0, gidx)
+            entry["ids"][:B].copy_(torch.where(mask, gathered,
+                                               entry["ids"][:B]))
+        if entry["graph"] is not None:
+            entry["graph"].replay()
+            logits = entry["logits"][:B]
+        else:
+            logits = self._device_decode_fwd(entry["rows"], entry["ids"],
+                                             entry["inc"])[:B]
+        sampled = self._sample_device_params(logits, temps_l, tops_l, seeds_l)
+        kvm.advance_many(seq_ids)
+        self.decode_tokens += B
+        self._spec_worker = {"sampled": sampled}
 
     def _finish_or_run(self, r: GenRequest, tok: int):
         """Called with lock held, after appending tok."""
@@ -1440,6 +1603,14 @@ class LLMEngine:
                     plan = [(kvm.seq_of_slot(s), t)
                             for s, t in zip(cmd[3:3 + B], cmd[3 + B:3 + 2 * B])]
                     inst._decode_exec(plan)
+                elif op == OP_DECODE_ASYNC:
+                    inst = self._instances[self._instance_order[cmd[1]]]
+                    inst._decode_async_worker(cmd)
+                elif op == OP_ROLLBACK:
+                    inst = self._instances[self._instance_order[cmd[1]]]
+                    n = cmd[2]
+                    inst.kvm.rollback_many(
+                        [inst.kvm.seq_of_slot(s) for s in cmd[3:3 + n]])
                 elif op == OP_BARRIER:
                     # timing fence (bench.py --tp): drain this rank's
                     # stream, then rendezvous so rank 0's clock bounds

@@ -99,6 +99,9 @@ def paged_decode_attention(out: torch.Tensor, q: torch.Tensor,
     ratio = n_q // n_kv
     for b in range(B):
         length = int(seq_lens[b])
+        if length <= 0:
+            out[b] = 0  # graph pad row (dev_seq_lens == -1); HIP kernel skips
+            continue
         for g in range(n_kv):
             K, V = _gather_kv(k_cache, v_cache, page_table[b], length, g)
             for qh in range(g * ratio, (g + 1) * ratio):
