@@ -85,7 +85,9 @@ def tp_group():
 # + pickled-bytes pair on the same channel.
 
 _plan_group = None
-_PLAN_CAP = 4 + 2 * 4096  # op + args + 4096 (slot, token) pairs
+# op + args + up to 4096 async rows at 6 ints each (gloo broadcast is
+# latency-bound — 128 B and 192 KB cost the same, tools/plan_latency.py)
+_PLAN_CAP = 64 + 6 * 4096
 _plan_buf: Optional[torch.Tensor] = None
 _OP_PICKLE = 0  # engine op codes start at 1 (llm.OP_*)
 
