@@ -25,6 +25,9 @@ DEFAULTS: Dict[str, Dict[str, Any]] = {
     "server": {
         "host": "127.0.0.1",
         "port": 8081,
+        # worker-thread cap for blocking request handlers; each in-flight
+        # generation parks one thread, so size for max concurrent agents
+        "threadpool_size": 1024,
     },
     "store": {
         # state root: registry AOF, WAL, logs, backups
@@ -42,7 +45,10 @@ DEFAULTS: Dict[str, Dict[str, Any]] = {
         "max_seqs": 2048,          # sequence-slot budget (device mirrors)
         "graph_capture": True,     # hipGraph-capture the decode step
         "async_decode": True,      # one-step-lag speculative resolution
+        "prefill_stream": True,    # prefill on a side HIP stream
         "prefix_sharing": True,    # COW system-prompt KV across agents
+        "moe_ep_mode": "auto",     # auto|a2a|dense (Mixtral EP dispatch)
+        "moe_a2a_min_tokens": 128,
         "dense_quant": "",         # fp8|mxfp4 decode projections (turbo)
         "expert_fp8": False,       # e4m3 Mixtral expert GEMMs
         "expert_fp4": False,       # MXFP4 Mixtral expert GEMMs

@@ -262,6 +262,7 @@ class ModelInstance:
         # rollback slots applied locally but not yet mirrored to workers
         # (flushed at the head of the next broadcast, any kind)
         self._pending_rb: List[int] = []
+        self._phase_ms: Dict[str, float] = {}
         # worker-side speculative mirror: previous step's device sample
         self._spec_worker: Optional[Dict[str, Any]] = None
         self._spec: Optional[Dict[str, Any]] = None
@@ -274,7 +275,9 @@ class ModelInstance:
         # cross-stream readers are speculatively-launched rows that are
         # already invalidated (their outputs are discarded).
         self._prefill_stream = (torch.cuda.Stream()
-                                if self.is_gpu else None)
+                                if self.is_gpu
+                                and bool(engine_cfg.get("prefill_stream", True))
+                                else None)
         # set whenever pages were freed (reset/rollback) while a decode
         # step may still be in flight: the next side-stream prefill must
         # order itself after that decode so a speculative K/V write can't
@@ -336,10 +339,33 @@ class ModelInstance:
         return st is not None and (time.time() - st) > self.watchdog_timeout_s
 
     def _loop(self):
+        # AGENTAINER_ENGINE_PROFILE=/path.txt: cProfile the first ~800
+        # engine steps and dump cumulative stats (diagnosing host-side
+        # step cost under serving load)
+        prof_path = os.environ.get("AGENTAINER_ENGINE_PROFILE", "")
+        prof = None
+        prof_steps = 0
+        if prof_path:
+            import cProfile
+            prof = cProfile.Profile()
         while not self._stop.is_set():
             try:
                 with self._step_mutex:
+                    if prof is not None:
+                        prof.enable()
                     did = self.step()
+                    if prof is not None:
+                        prof.disable()
+                        prof_steps += 1 if did else 0
+                        if prof_steps == 800:
+                            import io
+                            import pstats
+                            s = io.StringIO()
+                            pstats.Stats(prof, stream=s).sort_stats(
+                                "cumulative").print_stats(40)
+                            with open(prof_path, "w") as f:
+                                f.write(s.getvalue())
+                            prof = None
             except Exception:
                 traceback.print_exc()
                 did = False
@@ -362,6 +388,10 @@ class ModelInstance:
         if self.async_decode:
             ran = self._step_async()
             self.steps += 1 if ran else 0
+            if ran:  # EMA of wall step time (engine-side diagnostics)
+                dt = (time.time() - self._step_started) * 1000.0
+                self._step_ms_ema = (0.05 * dt
+                                     + 0.95 * getattr(self, "_step_ms_ema", dt))
             self._step_started = None
             self.last_step_t = time.time()
             return ran
@@ -388,9 +418,17 @@ class ModelInstance:
 
     # ---------- async (speculative) decode step ----------
 
+    def _phase_mark(self, key: str, t0: float) -> float:
+        """EMA per-phase wall times (stats diagnostics); returns now."""
+        now = time.time()
+        ema = self._phase_ms.get(key, 0.0)
+        self._phase_ms[key] = 0.05 * (now - t0) * 1000.0 + 0.95 * ema
+        return now
+
     def _step_async(self) -> bool:
         kvm = self.kvm
         dev = self.device
+        t0 = time.time()
         # phase 1: LAUNCH the next decode for the current running set; old
         # rows' input tokens come straight from the previous step's device
         # sample buffer (their host values are not resolved yet)
@@ -466,17 +504,20 @@ class ModelInstance:
             self.decode_tokens += B
             self.occupancy_acc += B / max(1, self.max_decode_batch)
             launched = {"reqs": batch, "sampled": sampled, "invalid": set()}
+        t0 = self._phase_mark("p1_launch_ms", t0)
         prev = self._spec
         self._spec = launched
 
         # phase 2: RESOLVE the previous step's tokens on the host while the
         # GPU runs the step launched above
         self._resolve_spec(prev, launched)
+        t0 = self._phase_mark("p2_resolve_ms", t0)
 
         # phase 3: admission + prefill, overlapping the decode replay
         # launched in phase 1 (side stream; fully host-synced by the
         # prefill sampling before this returns)
         admitted = self._admit()
+        t0 = self._phase_mark("p3_admit_ms", t0)
         if admitted:
             if self._prefill_stream is not None:
                 if self._pages_freed:
@@ -496,6 +537,7 @@ class ModelInstance:
                     self._prefill(admitted)
                 except OutOfPages as e:
                     self._fail_prefill(admitted, str(e))
+            self._phase_mark("p3_prefill_ms", t0)
         return bool(batch) or bool(admitted) or prev is not None
 
     def _fail_prefill(self, reqs: List[GenRequest], msg: str) -> None:
@@ -685,6 +727,8 @@ class ModelInstance:
             par.broadcast_obj(cmd)
 
     def _prefill(self, reqs: List[GenRequest]):
+        t0 = time.time()
+        c0 = time.thread_time()
         plan = []
         n_extra = 0  # shared-prefix rows prepended to the plan (no GenRequest)
         final = []   # reqs whose slice completes the prompt -> sample
@@ -719,7 +763,9 @@ class ModelInstance:
             r.prefill_pos += sl
             final.append(is_final)
         self._bcast(("prefill", self.name, plan))
+        t0 = self._phase_mark("pf_plan_ms", t0)
         logits = self._prefill_exec(plan)
+        t0 = self._phase_mark("pf_exec_ms", t0)  # launch-side (no sync)
         logits = logits[n_extra:]
         f_rows = [i for i, f in enumerate(final) if f]
         f_reqs = [r for r, f in zip(reqs, final) if f]
@@ -731,13 +777,35 @@ class ModelInstance:
             return
         if len(f_rows) < len(reqs):
             logits = logits[f_rows]
-        toks = self._sample(logits, f_reqs)
+        if self.is_gpu and os.environ.get("AGENTAINER_PF_SYNC_PROBE"):
+            torch.cuda.synchronize()  # diagnose GPU-wait vs sample cost
+            t0 = self._phase_mark("pf_sync_ms", t0)
+        toks = self._sample(logits, f_reqs)  # host sync point
+        self._phase_mark("pf_sample_ms", t0)
+        ema = self._phase_ms.get("pf_cpu_ms", 0.0)
+        self._phase_ms["pf_cpu_ms"] = (0.05 * (time.thread_time() - c0) * 1000.0
+                                       + 0.95 * ema)
         now = time.time()
         with self._lock:
             for r, t in zip(f_reqs, toks):
                 r.generated.append(int(t))
                 r.first_token_t = now
                 self._finish_or_run(r, int(t))
+
+    @staticmethod
+    def _prefill_bucket(n: int) -> int:
+        """Round a prefill token count up to a small set of GEMM shapes.
+
+        hipBLASLt runs its (CPU-expensive) algorithm heuristic per NEW
+        problem shape; real-text prompts make nearly every prefill batch
+        a never-seen M, which measured as ~100 ms of pure CPU per prefill
+        under HTTP serving (tools/server_load.py engine_mode.pf_cpu_ms).
+        Bucketing M to powers of two keeps the library cache warm after
+        one occurrence of each bucket."""
+        b = 32
+        while b < n:
+            b <<= 1
+        return b
 
     def _prefill_exec(self, plan) -> torch.Tensor:
         """Runs identically on every TP rank (plan = seq ids + tokens)."""
@@ -766,21 +834,60 @@ class ModelInstance:
         if self.dev_decode:
             for sid in seq_ids:
                 self.kvm.push_dev(sid)
+        self.prefill_tokens += len(ids)
+        page_table = self.kvm.page_table(seq_ids, device="cpu")
+        seq_lens = self.kvm.seq_lens(seq_ids, device="cpu")
+        n_real = len(seq_ids)
+        # pad the token dimension to a bucketed GEMM shape: extra tokens
+        # form one fake row whose K/V land in the reserved scratch page 0
+        # and whose logits are never read (last_rows covers real rows only)
+        n_pad = (self._prefill_bucket(len(ids)) - len(ids)
+                 if self.is_gpu else 0)
+        if n_pad > 0:
+            PS = self.kvm.page_size
+            q_starts.append(len(ids))
+            q_lens.append(n_pad)
+            ids.extend([0] * n_pad)
+            positions.extend(i % PS for i in range(n_pad))
+            slots.extend(i % PS for i in range(n_pad))  # page 0 scratch
+            # the attention kernel requires seq_len >= qlen (causal window
+            # math); the pad row's "pages" are ceil(n_pad/PS) references to
+            # scratch page 0, so its K/V reads stay in-bounds
+            pad_pages = -(-n_pad // PS)
+            width = max(page_table.size(1), pad_pages)
+            if width > page_table.size(1):
+                page_table = torch.cat(
+                    [page_table,
+                     torch.zeros(page_table.size(0),
+                                 width - page_table.size(1),
+                                 dtype=torch.int32)], dim=1)
+            pad_row = torch.zeros(1, width, dtype=torch.int32)
+            page_table = torch.cat([page_table, pad_row], dim=0)
+            seq_lens = torch.cat([seq_lens, torch.tensor(
+                [n_pad], dtype=torch.int32)])
         md = AttnMetadata(
-            page_table=self.kvm.page_table(seq_ids, device=dev),
-            seq_lens=self.kvm.seq_lens(seq_ids, device=dev),
+            page_table=page_table.to(dev),
+            seq_lens=seq_lens.to(dev),
             slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
             positions=torch.tensor(positions, dtype=torch.int32, device=dev),
             is_prefill=True,
             query_starts=torch.tensor(q_starts, dtype=torch.int32, device=dev),
             query_lens=torch.tensor(q_lens, dtype=torch.int32, device=dev),
+            max_qlen=max(q_lens),
         )
         input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
-        last_rows = torch.tensor([s + l - 1 for s, l in zip(q_starts, q_lens)],
-                                 dtype=torch.long, device=dev)
-        logits = self.model(input_ids, md, self.kvm.kv_caches(), last_rows)
-        self.prefill_tokens += len(ids)
-        return logits
+        last_rows = torch.tensor(
+            [s + l - 1 for s, l in zip(q_starts[:n_real], q_lens[:n_real])],
+            dtype=torch.long, device=dev)
+        c0 = time.thread_time()
+        out = self.model(input_ids, md, self.kvm.kv_caches(), last_rows)
+        fwd_cpu = (time.thread_time() - c0) * 1000.0
+        ema = self._phase_ms.get("pf_fwd_cpu_ms", 0.0)
+        self._phase_ms["pf_fwd_cpu_ms"] = 0.05 * fwd_cpu + 0.95 * ema
+        mn = self._phase_ms.get("pf_fwd_cpu_min_ms")
+        if mn is None or fwd_cpu < mn:
+            self._phase_ms["pf_fwd_cpu_min_ms"] = fwd_cpu
+        return out
 
     # ---------- device-side decode (hipGraph) ----------
 
@@ -1553,6 +1660,14 @@ class LLMEngine:
                 "steps": inst.steps,
                 "decode_tokens": inst.decode_tokens,
                 "prefill_tokens": inst.prefill_tokens,
+                # execution-mode diagnostics: use_graph flips False on a
+                # capture failure; buckets list which batch sizes captured
+                "use_graph": inst.use_graph,
+                "async_decode": inst.async_decode,
+                "graph_buckets": sorted(inst._graphs.keys()),
+                "step_ms_ema": round(getattr(inst, "_step_ms_ema", 0.0), 3),
+                "phase_ms": {k: round(v, 3)
+                             for k, v in inst._phase_ms.items()},
                 "kv_pages_used": inst.kvm.used_pages,
                 "kv_pages_free": inst.kvm.free_pages,
                 "kv_pages_shared": len(inst.kvm._refs),

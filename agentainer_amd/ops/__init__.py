@@ -131,12 +131,13 @@ def paged_decode_attention(out, q, k_cache, v_cache, page_table, seq_lens,
 
 
 def paged_prefill_attention(out, q, k_cache, v_cache, page_table, seq_lens,
-                            query_starts, query_lens, scale: float):
+                            query_starts, query_lens, scale: float,
+                            max_qlen: int = 0):
     mod = _dispatch("paged_prefill_attention", q)
     if mod:
         mod.paged_prefill_attention(out, q, k_cache, v_cache, page_table,
                                     seq_lens, query_starts, query_lens,
-                                    float(scale))
+                                    float(scale), int(max_qlen))
     else:
         reference.paged_prefill_attention(out, q, k_cache, v_cache, page_table,
                                           seq_lens, query_starts, query_lens,

@@ -21,7 +21,7 @@ void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor page_table, torch::Tensor seq_lens,
                              torch::Tensor query_starts, torch::Tensor query_lens,
-                             double scale);
+                             double scale, long max_qlen);
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void topp_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                  torch::Tensor top_ps, torch::Tensor seeds);

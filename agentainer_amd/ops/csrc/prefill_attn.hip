@@ -306,7 +306,7 @@ void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor page_table, torch::Tensor seq_lens,
                              torch::Tensor query_starts, torch::Tensor query_lens,
-                             double scale) {
+                             double scale, long max_qlen) {
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16);
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
@@ -320,11 +320,17 @@ void paged_prefill_attention(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(ratio % 4 == 0 || ratio == n_q,
               "GQA ratio must be a multiple of 4 (waves of a workgroup share "
               "one kv head)");
-  auto q_lens_cpu = query_lens.to(torch::kCPU);
-  int max_qlen = 0;
-  auto* ql = q_lens_cpu.data_ptr<int>();
-  for (int i = 0; i < B; ++i) max_qlen = std::max(max_qlen, ql[i]);
-  if (max_qlen == 0) return;
+  if (max_qlen <= 0) {
+    // fallback: derive from the device tensor. This D2H copy BLOCKS the
+    // host until the stream drains (measured ~2 ms/call under serving
+    // load with a deep queue — 32x per prefill); hot callers pass
+    // max_qlen explicitly (AttnMetadata.max_qlen).
+    auto q_lens_cpu = query_lens.to(torch::kCPU);
+    auto* ql = q_lens_cpu.data_ptr<int>();
+    for (int i = 0; i < B; ++i)
+      max_qlen = std::max<long>(max_qlen, ql[i]);
+    if (max_qlen == 0) return;
+  }
   const int qtiles = (max_qlen + QBLK - 1) / QBLK;
   auto stream = at::hip::getCurrentHIPStream();
 #define PF_LAUNCH(CT)                                                          \
