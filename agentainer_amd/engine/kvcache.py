@@ -331,12 +331,16 @@ class KVCacheManager:
                 return
             if short > len(self._free):
                 raise OutOfPages(f"KV pool exhausted ({self.n_pages} pages)")
-            for _ in range(short):
-                page = self._free.pop()
-                s.pages.append(page)
-                if self.mirrors:
-                    self.dev_page_table[self._slot_of[seq_id],
-                                        len(s.pages) - 1] = page
+            start = len(s.pages)
+            taken = self._free[-short:][::-1]  # same order as a pop() loop
+            del self._free[-short:]
+            s.pages.extend(taken)
+            if self.mirrors:
+                # one batched mirror write — per-page scalar assigns cost
+                # a tiny H2D launch EACH (visible in the admit phase)
+                self.dev_page_table[self._slot_of[seq_id],
+                                    start:start + short] = torch.tensor(
+                    taken, dtype=torch.int32, device=self.device)
 
     def can_append(self, seq_id: str, n_tokens: int) -> bool:
         with self._lock:
