@@ -263,6 +263,8 @@ class ModelInstance:
         # (flushed at the head of the next broadcast, any kind)
         self._pending_rb: List[int] = []
         self._phase_ms: Dict[str, float] = {}
+        # deferred prefill sample (async engines): (logits, final_reqs)
+        self._pf_pending: Optional[tuple] = None
         # worker-side speculative mirror: previous step's device sample
         self._spec_worker: Optional[Dict[str, Any]] = None
         self._spec: Optional[Dict[str, Any]] = None
@@ -429,6 +431,12 @@ class ModelInstance:
         kvm = self.kvm
         dev = self.device
         t0 = time.time()
+        # phase 0: publish last step's deferred prefill sample, so those
+        # rows are in `running` for this step's snapshot (same step they
+        # would have joined under synchronous sampling)
+        had_pf = self._pf_pending is not None
+        self._resolve_prefill()
+        t0 = self._phase_mark("p0_pf_resolve_ms", t0)
         # phase 1: LAUNCH the next decode for the current running set; old
         # rows' input tokens come straight from the previous step's device
         # sample buffer (their host values are not resolved yet)
@@ -538,7 +546,7 @@ class ModelInstance:
                 except OutOfPages as e:
                     self._fail_prefill(admitted, str(e))
             self._phase_mark("p3_prefill_ms", t0)
-        return bool(batch) or bool(admitted) or prev is not None
+        return bool(batch) or bool(admitted) or prev is not None or had_pf
 
     def _fail_prefill(self, reqs: List[GenRequest], msg: str) -> None:
         """Admission reserves full KV room, so a prefill-time OutOfPages
@@ -600,9 +608,10 @@ class ModelInstance:
         self.kvm.rollback_many(rollback_seqs)
 
     def drain_async(self) -> None:
-        """Resolve the in-flight speculative step WITHOUT launching a new
-        one (called before detach/offload and at engine stop, so
-        checkpoints never capture an unresolved speculative token)."""
+        """Resolve the in-flight speculative step AND any deferred prefill
+        WITHOUT launching new work (called before detach/offload and at
+        engine stop, so checkpoints never capture an unresolved token)."""
+        self._resolve_prefill()
         prev = self._spec
         self._spec = None
         self._resolve_spec(prev, None)
@@ -780,6 +789,15 @@ class ModelInstance:
         if self.is_gpu and os.environ.get("AGENTAINER_PF_SYNC_PROBE"):
             torch.cuda.synchronize()  # diagnose GPU-wait vs sample cost
             t0 = self._phase_mark("pf_sync_ms", t0)
+        if self.async_decode:
+            # DEFER the sample: a prefilled row can only join the decode
+            # batch at the NEXT step's snapshot anyway, so resolving at
+            # the head of the next _step_async costs no latency and lets
+            # the prefill's GPU tail overlap a whole decode step instead
+            # of host-blocking here
+            self._pf_pending = (logits, f_reqs)
+            self._phase_mark("pf_sample_ms", t0)
+            return
         toks = self._sample(logits, f_reqs)  # host sync point
         self._phase_mark("pf_sample_ms", t0)
         ema = self._phase_ms.get("pf_cpu_ms", 0.0)
@@ -788,6 +806,28 @@ class ModelInstance:
         now = time.time()
         with self._lock:
             for r, t in zip(f_reqs, toks):
+                r.generated.append(int(t))
+                r.first_token_t = now
+                self._finish_or_run(r, int(t))
+
+    def _resolve_prefill(self) -> None:
+        """Sample + publish a deferred prefill (see _prefill). Runs at the
+        head of the next step and at every drain point, so no checkpoint
+        or unbind ever observes an unsampled prefill."""
+        pend = self._pf_pending
+        if pend is None:
+            return
+        self._pf_pending = None
+        logits, f_reqs = pend
+        if self._prefill_stream is not None:
+            # order the sampling kernels after the side-stream producer
+            torch.cuda.current_stream().wait_stream(self._prefill_stream)
+        toks = self._sample(logits, f_reqs)
+        now = time.time()
+        with self._lock:
+            for r, t in zip(f_reqs, toks):
+                if r.done.is_set():
+                    continue  # failed/detached while deferred
                 r.generated.append(int(t))
                 r.first_token_t = now
                 self._finish_or_run(r, int(t))
