@@ -234,3 +234,34 @@ def test_prometheus_metrics_endpoint(client):
     assert body.count("# TYPE agentainer_agents_attached") == 1
     # unauthenticated scrape rejected
     assert client.get("/metrics/prometheus").status_code == 401
+
+
+def test_threadpool_sized_for_concurrent_generations(tmp_path):
+    """anyio's default 40-token limiter silently capped the server at ~40
+    concurrent generations (each blocking chat parks a worker thread);
+    the startup hook must raise it to server.threadpool_size."""
+    import anyio
+
+    from agentainer_amd.api.server import create_app
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.echo import EchoEngine
+    from agentainer_amd.service import Runtime
+    from agentainer_amd.store import Store
+    from starlette.testclient import TestClient
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    root = str(tmp_path / "root")
+    cfg.data["store"]["path"] = root
+    s = Store(root + "/state", sync="never")
+    rt = Runtime(cfg, engine=EchoEngine(s), store=s, state_root=root)
+    app = create_app(rt)
+
+    @app.get("/__limiter")  # test-only: the limiter is per-event-loop
+    async def _lim():
+        return {"tokens": anyio.to_thread
+                .current_default_thread_limiter().total_tokens}
+
+    with TestClient(app) as client:  # runs startup events
+        tokens = client.get("/__limiter").json()["tokens"]
+        assert tokens >= 1024, tokens
+    rt.shutdown()
