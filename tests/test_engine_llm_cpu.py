@@ -347,3 +347,52 @@ def test_multi_model_cohosting(llm_runtime):
     # engine stats report both models
     st = rt.engine.stats()
     assert {"tiny-llama", "tiny-mixtral"} <= set(st["models"])
+
+
+def test_decode_batch_cap_is_fair(tmp_path):
+    """running > max_decode_batch must round-robin, not starve the tail
+    (found as a 34.6s p99 at 512 agents against the 256-row cap)."""
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.llm import GenRequest, LLMEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    cfg.data["store"]["path"] = str(tmp_path / "root")
+    cfg.data["engine"]["kv_pool_gb"] = 0.01
+    cfg.data["engine"]["sync_mode"] = True
+    cfg.data["engine"]["max_decode_batch"] = 2   # oversubscribe: 6 agents
+    s = Store(str(tmp_path / "root" / "state"), sync="never")
+    eng = LLMEngine(s, cfg, device="cpu", state_root=str(tmp_path / "root"))
+    man = Manager(s, eng, cfg)
+    reqs = []
+    for i in range(6):
+        a = man.deploy(name=f"fair{i}", model="tiny-llama",
+                       sampling={"max_tokens": 5})
+        man.start(a.id)
+        inst = eng._instances["tiny-llama"]
+        r = GenRequest(agent_id=a.id, prompt_tokens=list(range(3, 19)),
+                       max_new=5, temperature=0.0, top_p=1.0, seed=0)
+        b = inst.binding(a.id)
+        with inst._lock:
+            b.queue.put(r)
+            inst._pump_agent(b)
+        reqs.append(r)
+    # each of the 6 rows needs 4 decode steps at 2 rows/step => ~12 decode
+    # steps + prefills; WITH starvation the last rows would need the first
+    # ones to fully finish first, which still converges — so assert the
+    # stronger property: progress interleaves (no row finishes 5 tokens
+    # before every row has at least 1)
+    inst = eng._instances["tiny-llama"]
+    for _ in range(40):
+        inst.step()
+        lens = [len(r.generated) for r in reqs]
+        if max(lens) >= 5:
+            assert min(lens) >= 1, f"tail starved: {lens}"
+        if all(r.done.is_set() for r in reqs):
+            break
+    if inst.async_decode:
+        inst.drain_async()
+    assert all(r.done.is_set() and not r.error for r in reqs)
+    eng.shutdown()
+    s.close()
