@@ -101,6 +101,7 @@ def main():
         stop = threading.Event()
         rec_on = threading.Event()
         lat = []
+        ttfts = []
         lat_lock = threading.Lock()
         counts = [0] * args.agents
         prompt = " ".join(f"w{i}" for i in range(args.prompt_words))
@@ -120,9 +121,16 @@ def main():
                     if args.clear_every and turns % args.clear_every == 0:
                         c.post(base + f"/agent/{aid}/clear", json={})
                     if rec_on.is_set():
+                        tf = None
+                        try:
+                            tf = r.json().get("ttft_s")
+                        except ValueError:
+                            pass
                         with lat_lock:
                             lat.append(dt)
                             counts[i] += 1
+                            if tf is not None:
+                                ttfts.append(float(tf))
 
         def engine_stats():
             try:
@@ -159,6 +167,11 @@ def main():
             "p99_e2e_s": (round(lat[min(n - 1, int(0.99 * n))], 4)
                           if lat else None),
             "completed": n,
+            "ttft_p50_s": (round(statistics.median(ttfts), 4)
+                           if ttfts else None),
+            "ttft_p99_s": (round(sorted(ttfts)[min(len(ttfts) - 1,
+                                                   int(0.99 * len(ttfts)))], 4)
+                           if ttfts else None),
             # engine-side view of the same window: starved batch shows as
             # low decode occupancy; a starved engine thread as low steps/s
             "engine_steps_per_s": round(dsteps / elapsed, 1),
