@@ -309,6 +309,11 @@ def create_app(rt: Runtime) -> FastAPI:
         rt.agents.get(agent_id)
         return envelope(True, "", rt.metrics.get_metrics_history(agent_id, duration_s))
 
+    @app.get("/metrics/device")
+    async def device_metrics(_tok: str = Depends(require_auth)):
+        # device-level sample (HBM + xGMI) from the 10s metrics loop
+        return envelope(True, data=rt.store.get("metrics:current:device") or {})
+
     @app.get("/metrics/engine")
     async def engine_metrics(_tok: str = Depends(require_auth)):
         return envelope(True, "", rt.engine.stats())

@@ -277,12 +277,22 @@ def health(client: Client, agent_id: Optional[str]):
 
 
 @cli.command()
-@click.argument("agent_id")
+@click.argument("agent_id", required=False)
 @click.option("--history", is_flag=True)
 @click.option("--duration", default=3600.0, type=float, help="history window (s)")
+@click.option("--device", "device_", is_flag=True,
+              help="device-level metrics (HBM, xGMI link counters)")
 @pass_client
-def metrics(client: Client, agent_id: str, history: bool, duration: float):
-    """Show agent metrics (tokens/s, req/s, p50/p99, KV usage)."""
+def metrics(client: Client, agent_id, history: bool, duration: float,
+            device_: bool):
+    """Show agent metrics (tokens/s, req/s, p50/p99, KV usage), or
+    device-level HBM/xGMI metrics with --device."""
+    if device_:
+        resp = client.call("GET", "/metrics/device")
+        click.echo(json.dumps(resp.get("data"), indent=2))
+        return
+    if not agent_id:
+        raise click.UsageError("AGENT_ID required (or use --device)")
     if history:
         resp = client.call("GET", f"/agents/{agent_id}/metrics/history",
                            params={"duration_s": duration})

@@ -112,3 +112,19 @@ def test_cli_backup_export_import(run, server, tmp_path):
     out = run("backup", "import", bundle)
     assert bid in out
     assert bid in run("backup", "list")
+
+
+def test_metrics_device_cli(run):
+    """`agentainer metrics --device` hits /metrics/device (empty JSON on
+    the echo engine — populated by the 10s device sampler on GPU)."""
+    import json as _json
+
+    out = run("metrics", "--device")
+    assert isinstance(_json.loads(out), dict)
+
+
+def test_metrics_requires_agent_or_device(run):
+    try:
+        run("metrics", expect=2)  # click usage error
+    except AssertionError:
+        raise
