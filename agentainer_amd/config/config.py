@@ -14,7 +14,6 @@ from __future__ import annotations
 
 import copy
 import os
-from dataclasses import dataclass, field
 from typing import Any, Dict, Optional
 
 import yaml

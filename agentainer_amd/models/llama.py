@@ -16,8 +16,8 @@ by rank: column-parallel qkv/gate/up, row-parallel o/down with all-reduce
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional
+from dataclasses import dataclass
+from typing import Dict, List, Optional
 
 import torch
 import torch.nn.functional as F

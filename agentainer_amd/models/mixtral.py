@@ -36,7 +36,6 @@ MoE execution model, chosen for the MI355X serving regime:
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 

@@ -12,8 +12,6 @@ paged-cache layouts k_cache [P, n_kv, D/8, PS, 8] / v_cache [P, n_kv, PS, D].
 
 from __future__ import annotations
 
-import math
-from typing import Optional
 
 import torch
 

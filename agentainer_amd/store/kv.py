@@ -27,8 +27,8 @@ import json
 import os
 import threading
 import time
-from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+from dataclasses import dataclass
+from typing import Any, Callable, Dict, List, Optional, Tuple
 
 _TOMBSTONE = object()
 
