@@ -93,6 +93,13 @@ class MixtralMoE(torch.nn.Module):
     ep_mode = "auto"
     a2a_min_tokens = A2A_MIN_TOKENS
 
+    # prefill-scale batches route SPARSELY (gather per expert): dense
+    # routing computes every expert over the WHOLE batch — fine for tiny
+    # decode batches where every expert is hit anyway (and required for
+    # hipGraph capture), but at prefill it wastes (E - top_k)/E of the
+    # expert FLOPs. Crossover left at the decode/prefill boundary.
+    SPARSE_MIN_TOKENS = 65
+
     def forward(self, h: torch.Tensor, ep_group=None) -> torch.Tensor:
         T = h.size(0)
         if ep_group is not None and torch.distributed.get_world_size(ep_group) > 1:
@@ -104,6 +111,11 @@ class MixtralMoE(torch.nn.Module):
         probs = torch.softmax(logits, dim=-1)
         topv, topi = probs.topk(self.top_k, dim=-1)         # [T, k]
         topv = topv / topv.sum(dim=-1, keepdim=True)        # renormalize
+        if T >= self.SPARSE_MIN_TOKENS:
+            out = self._experts_sparse(h, topi, topv)
+            if ep_group is not None:
+                torch.distributed.all_reduce(out, group=ep_group)
+            return out
         # dense top-k mask: weight[t, e] (0 if e not in token t's top-k)
         weight = torch.zeros_like(probs)
         weight.scatter_(1, topi, topv)
@@ -136,6 +148,32 @@ class MixtralMoE(torch.nn.Module):
         if ep_group is not None:
             torch.distributed.all_reduce(out, group=ep_group)
         return out
+
+    def _experts_sparse(self, h: torch.Tensor, topi: torch.Tensor,
+                        topv: torch.Tensor) -> torch.Tensor:
+        """Gather-per-expert execution for prefill-scale batches: each
+        local expert runs only over the tokens that routed to it (top_k/E
+        of the batch on average), with a weighted index_add_ scatter back.
+        Deterministic (fp32 accumulation per token, fixed expert order),
+        so greedy replay stays exact."""
+        T, H = h.shape
+        out = torch.zeros(T, H, dtype=torch.float32, device=h.device)
+        for i in range(self.n_local):
+            e = self.e0 + i
+            hit = (topi == e)                          # [T, k]
+            rows = hit.any(dim=-1).nonzero(as_tuple=True)[0]
+            if rows.numel() == 0:
+                continue
+            w = (topv * hit.float()).sum(dim=-1)[rows]  # [n] routing weight
+            x = h.index_select(0, rows)
+            gu = ops.linear(x, self.gate_up[i], None)
+            gate, up = gu[:, :self.inter], gu[:, self.inter:]
+            act = torch.empty(x.size(0), self.inter, dtype=gu.dtype,
+                              device=gu.device)
+            ops.silu_mul(act, gate, up)
+            eo = ops.linear(act, self.down[i], None)
+            out.index_add_(0, rows, w.unsqueeze(1).float() * eo.float())
+        return out.to(h.dtype)
 
     # ---------- token-shuffle EP (prefill) ----------
 

@@ -78,3 +78,30 @@ def test_mixtral_stop_resume_kv(mixtral_rt):
     r2a = rt.agent_request(a.id, "POST", "/chat", body={"message": "two"})[1]
     r2b = rt.agent_request(ctl.id, "POST", "/chat", body={"message": "two"})[1]
     assert r2a["response"] == r2b["response"]
+
+
+def test_sparse_prefill_experts_match_dense():
+    """Prefill-scale batches route sparsely (gather per expert, 2/8 of
+    the FLOPs); the output must match the dense-routed reference."""
+    import torch
+
+    from agentainer_amd.models.mixtral import MIXTRAL_CONFIGS, MixtralMoE
+
+    torch.manual_seed(4)
+    cfg = MIXTRAL_CONFIGS["tiny-mixtral"]
+    moe = MixtralMoE(cfg)
+    g = torch.Generator().manual_seed(4)
+    for p in moe.parameters():
+        p.data.normal_(0, 0.05, generator=g)
+    h = torch.randn(100, cfg.hidden_size, dtype=torch.bfloat16)
+    sparse = moe(h.clone())                    # T=100 >= SPARSE_MIN_TOKENS
+    try:
+        moe.SPARSE_MIN_TOKENS = 10**9          # force the dense path
+        dense = moe(h.clone())
+    finally:
+        del moe.SPARSE_MIN_TOKENS              # restore class attribute
+    diff = (sparse.float() - dense.float()).abs().max().item()
+    assert diff < 0.05, diff
+    # decode-scale batches stay dense (hipGraph-capturable)
+    small = moe(h[:8].clone())
+    assert small.shape == (8, cfg.hidden_size)
