@@ -109,3 +109,20 @@ def test_bench_tp2_gloo_contract():
     assert out["scaling"] == "strong"
     assert out["value"] > 0
     assert out["config"]["p50_e2e_s"] is not None
+
+
+@pytest.mark.timeout(300)
+def test_server_load_harness_contract():
+    """tools/server_load.py (the HTTP serving measurement behind the
+    round-2 serving-path fixes) must stay runnable: echo engine, tiny
+    window, one JSON line with the req/s + latency + engine fields."""
+    r = subprocess.run(
+        [sys.executable, "tools/server_load.py", "--device", "echo",
+         "--agents", "4", "--window", "3", "--warmup", "1"],
+        capture_output=True, text=True, cwd=ROOT, timeout=280)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = _last_json_line(r.stdout)
+    assert out["tool"] == "server_load"
+    assert out["http_req_per_s"] > 0
+    assert out["p50_e2e_s"] is not None
+    assert "engine_mode" in out and "decode_occupancy" in out
