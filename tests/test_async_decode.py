@@ -197,3 +197,21 @@ def test_tp2_async_decode_token_exact(tmp_path):
     res = torch.load(result_file, weights_only=True)
     assert all(len(t) > 0 for t in res["eager"])
     assert res["async"] == res["eager"], res
+
+
+def test_async_emulation_chunked_prefill_token_exact(tmp_path):
+    """Chunked prefill (prompt > max_batch_tokens) through the
+    speculative engine: slices interleave with in-flight decode steps and
+    the deferred sample lands after the FINAL slice — tokens must match
+    the synchronous engine exactly."""
+    tmp = str(tmp_path)
+    long_prompt = [3 + (i % 400) for i in range(300)]
+    specs = [(long_prompt, 6, 0.0, 0), (PROMPT, 8, 0.0, 0)]
+    extra = {"max_batch_tokens": 96}  # 300-token prompt -> 4 slices
+    e1, m1, s1 = _engine(tmp, "sync-ch", emulate=False, extra=extra)
+    want = _gen(e1, m1, "tiny-llama", specs, max_steps=60)
+    e1.shutdown(); s1.close()
+    e2, m2, s2 = _engine(tmp, "async-ch", emulate=True, extra=extra)
+    got = _gen(e2, m2, "tiny-llama", specs, max_steps=60)
+    e2.shutdown(); s2.close()
+    assert got == want, (got, want)
