@@ -345,3 +345,45 @@ def test_dense_quant_turbo_generation(tmp_path):
             assert o1 == o2 and len(o1) == 8, (dq, o1, o2)
         finally:
             engine.shutdown()
+
+
+@pytest.mark.gpu
+def test_weights_path_deploy_gpu(tmp_path):
+    """Weights-path deploy on device: HF checkpoint dir -> TP-aware
+    loader -> real BPE tokenizer -> greedy chat on the HIP kernels
+    (threaded engine drives its own steps)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from test_weights_path import _write_checkpoint
+
+    from agentainer_amd.service import Runtime
+
+    d = str(tmp_path / "ckpt")
+    _write_checkpoint(d)
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    root = str(tmp_path / "root")
+    cfg.data["store"]["path"] = root
+    cfg.data["engine"]["kv_pool_gb"] = 2.0
+    s = Store(root + "/state", sync="never")
+    eng = LLMEngine(s, cfg, device="cuda", state_root=root)
+    rt = Runtime(cfg, engine=eng, store=s, state_root=root)
+    try:
+        a = rt.agents.deploy(name="ckpt-gpu", model=d,
+                             sampling={"max_tokens": 6})
+        rt.agents.start(a.id)
+        st, p = rt.agent_request(a.id, "POST", "/chat",
+                                 body={"message": "the agent replies"})
+        assert st == 200, p
+        assert 0 < p["tokens"] <= 6
+        # determinism: same prompt from a fresh agent matches (loader +
+        # tokenizer + kernels all deterministic)
+        b = rt.agents.deploy(name="ckpt-gpu2", model=d,
+                             sampling={"max_tokens": 6})
+        rt.agents.start(b.id)
+        st2, p2 = rt.agent_request(b.id, "POST", "/chat",
+                                   body={"message": "the agent replies"})
+        assert st2 == 200 and p2["response"] == p["response"]
+    finally:
+        rt.shutdown()
