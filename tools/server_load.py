@@ -64,7 +64,11 @@ def main():
         args.window = min(args.window, 10.0)
         args.warmup = min(args.warmup, 2.0)
 
-    root = tempfile.mkdtemp(prefix="srvload-")
+    # SERVER_LOAD_LOGDIR: put the state root (incl. server.log) somewhere
+    # inspectable — e.g. gpurun_out/ so a crashed server's log survives
+    root = (os.environ.get("SERVER_LOAD_LOGDIR")
+            or tempfile.mkdtemp(prefix="srvload-"))
+    os.makedirs(root, exist_ok=True)
     port = _free_port()
     env = dict(os.environ)
     env.update({"AGENTAINER_STORE_PATH": root,
