@@ -147,10 +147,13 @@ class KVCacheManager:
         with self._lock:
             if seq_id in self._seqs:
                 raise ValueError(f"sequence {seq_id} exists")
+            if not self._free_slots:
+                # check BEFORE registering: raising after the _seqs insert
+                # left a zombie slotless entry behind (found by the
+                # stateful fuzz in tests/test_kvcache_properties.py)
+                raise OutOfPages("no free sequence slots")
             s = Sequence(seq_id)
             self._seqs[seq_id] = s
-            if not self._free_slots:
-                raise OutOfPages("no free sequence slots")
             slot = self._free_slots.pop()
             self._slot_of[seq_id] = slot
             self._seq_of_slot[slot] = seq_id
