@@ -266,11 +266,21 @@ class Store:
     def lpush(self, key: str, *vals: Any) -> int:
         return self._mutate(["lpush", key, *vals])
 
+    def _drop_if_empty(self, key: str) -> None:
+        """Redis semantics: a collection key vanishes when its last
+        element is removed (an empty list/zset/hash/set never lingers in
+        keys() scans)."""
+        e = self._data.get(key)
+        if e is not None and e.kind != "str" and not e.value:
+            del self._data[key]
+
     def _do_lpop(self, key: str):
         lst = self._list(key)
         if not lst:
             return None
-        return lst.pop(0)
+        v = lst.pop(0)
+        self._drop_if_empty(key)
+        return v
 
     def lpop(self, key: str) -> Any:
         return self._mutate(["lpop", key])
@@ -290,6 +300,7 @@ class Store:
             return 0
         n = lst.count(value)
         self._data[key].value = [v for v in lst if v != value]
+        self._drop_if_empty(key)
         return n
 
     def lrem(self, key: str, value: Any) -> int:
@@ -300,6 +311,7 @@ class Store:
         if lst is None:
             return
         self._data[key].value = lst[start : (None if stop == -1 else stop + 1)]
+        self._drop_if_empty(key)
 
     def ltrim(self, key: str, start: int, stop: int) -> None:
         self._mutate(["ltrim", key, start, stop])
@@ -343,6 +355,7 @@ class Store:
         dead = [m for m, s in z.items() if lo <= s <= hi]
         for m in dead:
             del z[m]
+        self._drop_if_empty(key)
         return len(dead)
 
     def zremrangebyscore(self, key: str, lo: float, hi: float) -> int:
@@ -406,6 +419,7 @@ class Store:
             return
         for m in members:
             s.discard(m)
+        self._drop_if_empty(key)
 
     def srem(self, key: str, *members: str) -> None:
         self._mutate(["srem", key, *members])
