@@ -116,6 +116,12 @@ class RequestManager:
         req = self.get(agent_id, req_id)
         if req is None:
             return None
+        if req.status != PENDING:
+            # idempotent under at-least-once delivery: a duplicate success
+            # (client retry, replay race) must not re-append the completed
+            # queue or resurrect a dead-lettered record (the reference
+            # double-stored here — replay_worker.go:158)
+            return req
         req.status = COMPLETED
         req.processed_at = time.time()
         req.response = response
@@ -131,6 +137,8 @@ class RequestManager:
         req = self.get(agent_id, req_id)
         if req is None:
             return None
+        if req.status != PENDING:
+            return req  # completed/dead-lettered records never regress
         req.retry_count += 1
         req.error = error
         if req.retry_count >= req.max_retries:
