@@ -129,3 +129,53 @@ def test_backup_restores_kv_exact(tmp_path):
         assert p_r["response"] == p_c["response"]
     finally:
         rt.shutdown()
+
+
+def test_backup_roundtrip_property(tmp_path):
+    """Property-style backup/restore: for a randomized set of agents with
+    randomized histories/metrics, create -> restore must reproduce every
+    agent config and conversation exactly (as {name}-restored)."""
+    import random
+
+    from agentainer_amd.backup import BackupManager
+    from agentainer_amd.config import load_config
+    from agentainer_amd.engine.echo import EchoEngine
+    from agentainer_amd.registry import Manager
+    from agentainer_amd.store import Store
+
+    rng = random.Random(7)
+    cfg = load_config(path="/nonexistent.yaml", env={})
+    store = Store(str(tmp_path / "state"), sync="never")
+    eng = EchoEngine(store)
+    man = Manager(store, eng, cfg)
+    bm = BackupManager(store, man, str(tmp_path / "bk"))
+    want = {}
+    for i in range(rng.randint(3, 6)):
+        sp = f"sys-{rng.randint(0, 99)}" if rng.random() < 0.7 else ""
+        a = man.deploy(name=f"pb{i}", model="echo", system_prompt=sp,
+                       auto_restart=rng.random() < 0.5,
+                       sampling={"max_tokens": rng.randint(1, 64)})
+        if rng.random() < 0.8:
+            man.start(a.id)
+        hist = [{"user": f"u{j}", "assistant": f"r{j}"}
+                for j in range(rng.randint(0, 5))]
+        for h in hist:
+            store.rpush(f"agent:{a.id}:conversations", h)
+        want[a.name] = (sp, a.auto_restart, dict(a.sampling), hist)
+    b = bm.create(name="prop", description="fuzz")
+    # wipe everything, then restore
+    for a in list(man.list()):
+        if a.status == "running":
+            man.stop(a.id)
+        man.remove(a.id)
+    assert not man.list()
+    bm.restore(b["id"] if isinstance(b, dict) else b.id)
+    got = {a.name: a for a in man.list()}
+    assert set(got) == {f"{n}-restored" for n in want}, got
+    for name, (sp, auto, sampling, hist) in want.items():
+        a = got[f"{name}-restored"]
+        assert (a.system_prompt or "") == sp
+        assert a.auto_restart == auto
+        assert dict(a.sampling) == sampling
+        assert store.lrange(f"agent:{a.id}:conversations") == hist
+    store.close()
